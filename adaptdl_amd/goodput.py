@@ -1,0 +1,267 @@
+"""Goodput model: throughput x statistical efficiency, and its fitting.
+
+Model form follows the Pollux reference implementation
+(``/root/reference/adaptdl/adaptdl/goodput.py:22-259``) so fitted parameters
+and scheduling decisions are comparable:
+
+  T_accum   ~ alpha_c + beta_c * atomic_bsz
+  T_network ~ alpha_{n|r} + beta_{n|r} * max(replicas - 2, eps)
+  T_optim   ~ (T_accum^gamma + T_network^gamma)^(1/gamma)
+  T_step    = accum_steps * T_accum + T_optim
+  goodput   = (replicas * atomic_bsz * (accum_steps+1)) / T_step * efficiency
+
+The intra-node (alpha_r/beta_r) branch models RCCL ring all-reduce over the
+MI355X's point-to-point xGMI mesh (7 links/GPU); the retrogression term
+captures the per-link serialization of rings as replica count grows.
+
+Unlike the reference, :func:`fit_perf_params` uses a hand-derived analytic
+gradient (validated against finite differences in tests) instead of the
+``autograd`` package, and never swaps numpy modules.
+"""
+
+import collections
+
+import numpy as np
+import scipy.optimize
+
+PerfParams = collections.namedtuple("PerfParams", [
+    "alpha_c",  # constant compute time per (accumulation) step
+    "beta_c",   # compute time per sample of atomic batch
+    "alpha_n",  # inter-node network constant
+    "beta_n",   # inter-node retrogression per replica beyond 2
+    "alpha_r",  # intra-node (xGMI) network constant
+    "beta_r",   # intra-node retrogression per replica beyond 2
+    "gamma",    # overlap p-norm exponent in [1, 10]
+])
+
+GradParams = collections.namedtuple("GradParams", ["sqr", "var"])
+
+_EPS = 1e-8
+
+
+class GoodputFunction(object):
+    """Evaluates/optimizes goodput for a given fitted model + GNS stats."""
+
+    def __init__(self, perf_params, grad_params, init_batch_size):
+        self._perf_params = PerfParams(*perf_params)
+        self._grad_params = GradParams(*grad_params)
+        self._init_batch_size = init_batch_size
+
+    def __call__(self, num_nodes, num_replicas, atomic_bsz, accum_steps):
+        return self.evaluate(num_nodes, num_replicas, atomic_bsz, accum_steps)
+
+    def evaluate(self, num_nodes, num_replicas, atomic_bsz, accum_steps):
+        batch_size = num_replicas * atomic_bsz * (accum_steps + 1)
+        assert np.all(self._init_batch_size <= batch_size)
+        return (self.throughput(num_nodes, num_replicas, atomic_bsz,
+                                accum_steps)
+                * self.efficiency(batch_size))
+
+    def throughput(self, num_nodes, num_replicas, atomic_bsz, accum_steps):
+        accum_time = _predict_accum_time(self._perf_params, atomic_bsz)
+        network_time = _predict_network_time(self._perf_params, num_nodes,
+                                             num_replicas)
+        optim_time = np.exp(_predict_log_optim_time(
+            self._perf_params, accum_time, network_time))
+        total_time = accum_steps * accum_time + optim_time
+        batch_size = num_replicas * atomic_bsz * (accum_steps + 1)
+        return batch_size / total_time
+
+    def efficiency(self, batch_size):
+        grad_sqr = self._grad_params.sqr
+        grad_var = self._grad_params.var
+        scale = batch_size / self._init_batch_size
+        denom = grad_var / scale + grad_sqr
+        gain = np.where(denom > 0, (grad_var + grad_sqr) / denom, 1.0)
+        return gain / scale
+
+    def optimize(self, num_nodes, num_replicas, max_batch_size=None,
+                 atomic_bsz_range=None, accumulation=False):
+        """Choose (atomic_bsz, accum_steps) maximizing goodput.
+
+        Samples 50 total batch sizes in geometric space per (nodes, replicas)
+        configuration, derives per-replica (atomic_bsz, accum_steps), and
+        returns the best goodput with its configuration.  Semantics match the
+        reference ``GoodputFunction.optimize`` (goodput.py:88-148) including
+        the single-replica accumulation floor for statistics quality.
+        """
+        assert np.all(np.less_equal(1, num_nodes))
+        assert np.all(np.less_equal(num_nodes, num_replicas))
+        if max_batch_size is None:
+            max_batch_size = self._init_batch_size
+        assert self._init_batch_size <= max_batch_size
+        atomic_bsz_range = atomic_bsz_range or (None, None)
+        min_atomic_bsz = atomic_bsz_range[0] or 1
+        max_atomic_bsz = atomic_bsz_range[1] or max_batch_size
+        output_shape = np.broadcast(num_nodes, num_replicas).shape
+        output_scalar = np.isscalar(num_nodes) or np.isscalar(num_replicas)
+        num_nodes = np.broadcast_to(num_nodes, output_shape).flatten()
+        num_replicas = np.broadcast_to(num_replicas, output_shape).flatten()
+        min_batch_size = np.maximum(self._init_batch_size,
+                                    min_atomic_bsz * num_replicas)
+        batch_size = np.geomspace(min_batch_size, max_batch_size)
+        local_bsz = batch_size / num_replicas
+        eps = 1e-8
+        if accumulation:
+            accum_steps = np.ceil(local_bsz / max_atomic_bsz - eps) - 1
+            # With a single replica and a scaled-up batch, require at least
+            # one accumulation step so the GNS has >= 2 samples per step.
+            accum_steps = np.where(
+                np.logical_and(num_replicas == 1,
+                               local_bsz > self._init_batch_size + eps),
+                np.maximum(accum_steps, 1), accum_steps).astype(int)
+            atomic_bsz = np.ceil(local_bsz / (accum_steps + 1) - eps)
+            atomic_bsz = atomic_bsz.astype(int)
+        else:
+            accum_steps = np.zeros_like(local_bsz, dtype=int)
+            atomic_bsz = np.where(
+                num_replicas == 1,
+                self._init_batch_size,
+                np.ceil(local_bsz - eps)).astype(int)
+        atomic_bsz = np.clip(atomic_bsz, min_atomic_bsz, max_atomic_bsz)
+        goodput = self.evaluate(num_nodes, num_replicas, atomic_bsz,
+                                accum_steps)
+        indices = np.argmax(goodput, axis=0), np.arange(goodput.shape[1])
+        goodput = goodput[indices].reshape(output_shape)
+        atomic_bsz = atomic_bsz[indices].reshape(output_shape)
+        accum_steps = accum_steps[indices].reshape(output_shape)
+        if output_scalar:
+            goodput = goodput.item()
+            atomic_bsz = atomic_bsz.item()
+            accum_steps = accum_steps.item()
+        return goodput, atomic_bsz, accum_steps
+
+
+def _predict_accum_time(params, atomic_bsz):
+    params = PerfParams(*params)
+    return params.alpha_c + params.beta_c * atomic_bsz
+
+
+def _predict_log_optim_time(params, accum_time, network_time):
+    gamma = PerfParams(*params).gamma
+    return np.log(accum_time ** gamma + network_time ** gamma) / gamma
+
+
+def _predict_network_time(params, num_nodes, num_replicas):
+    params = PerfParams(*params)
+    conds = [num_nodes > 1, num_replicas > 1]
+    bottleneck = np.select(conds, [params.alpha_n, params.alpha_r], _EPS)
+    retrogress = np.select(conds, [params.beta_n, params.beta_r], _EPS)
+    return bottleneck + retrogress * np.maximum(num_replicas - 2, _EPS)
+
+
+def _obj_and_grad(x, num_nodes, num_replicas, atomic_bsz,
+                  accum_step_time, optim_step_time):
+    """Objective of the fit plus its analytic gradient w.r.t. the 7 params.
+
+    Objective = RMSLE(accum prediction) + RMSLE(optim prediction)
+              + 1e-3 (gamma-1)^2 + 1e-2 ((beta_n/alpha_n)^2
+                                         + (beta_r/alpha_r)^2)
+    (matches the reference ``_obj_fn``, goodput.py:215-231).
+    """
+    alpha_c, beta_c, alpha_n, beta_n, alpha_r, beta_r, gamma = x
+    m = len(atomic_bsz)
+
+    inter = num_nodes > 1
+    intra = np.logical_and(~inter, num_replicas > 1)
+    none = np.logical_and(~inter, ~intra)
+    rfac = np.maximum(num_replicas - 2, _EPS)
+
+    A = alpha_c + beta_c * atomic_bsz
+    N = np.select([inter, intra], [alpha_n + beta_n * rfac,
+                                   alpha_r + beta_r * rfac],
+                  _EPS + _EPS * rfac)
+
+    logA = np.log(A)
+    Ag = A ** gamma
+    Ng = N ** gamma
+    S = Ag + Ng
+    L = np.log(S) / gamma
+
+    # err1: RMSE of logA vs log(accum_step_time)
+    d1 = logA - np.log(accum_step_time)
+    err1 = np.sqrt(np.mean(d1 ** 2))
+    # err2: RMSE of L vs log(optim_step_time)
+    d2 = L - np.log(optim_step_time)
+    err2 = np.sqrt(np.mean(d2 ** 2))
+
+    reg1 = 1e-3 * (gamma - 1.0) ** 2
+    reg2 = 1e-2 * ((beta_n / alpha_n) ** 2 + (beta_r / alpha_r) ** 2)
+    obj = err1 + err2 + reg1 + reg2
+
+    # --- gradient ---
+    g = np.zeros(7)
+    # err1 path: d err1/dA_i = d1_i / (m * err1 * A_i)
+    if err1 > 0:
+        w1 = d1 / (m * err1 * A)
+        g[0] += np.sum(w1)                   # alpha_c
+        g[1] += np.sum(w1 * atomic_bsz)      # beta_c
+    # err2 path: d err2/dL_i = d2_i / (m * err2)
+    if err2 > 0:
+        w2 = d2 / (m * err2)
+        dL_dA = Ag / (A * S)                 # = gamma*A^(g-1)/S / gamma
+        dL_dN = Ng / (N * S)
+        g[0] += np.sum(w2 * dL_dA)
+        g[1] += np.sum(w2 * dL_dA * atomic_bsz)
+        g[2] += np.sum(w2 * dL_dN * inter)             # alpha_n
+        g[3] += np.sum(w2 * dL_dN * rfac * inter)      # beta_n
+        g[4] += np.sum(w2 * dL_dN * intra)             # alpha_r
+        g[5] += np.sum(w2 * dL_dN * rfac * intra)      # beta_r
+        # dL/dgamma = ((A^g lnA + N^g lnN)/S - L) / gamma
+        logN = np.log(N)
+        dL_dg = ((Ag * logA + Ng * logN) / S - L) / gamma
+        g[6] += np.sum(w2 * dL_dg)
+    del none
+    # reg1
+    g[6] += 2e-3 * (gamma - 1.0)
+    # reg2
+    g[2] += 1e-2 * (-2.0 * beta_n ** 2 / alpha_n ** 3)
+    g[3] += 1e-2 * (2.0 * beta_n / alpha_n ** 2)
+    g[4] += 1e-2 * (-2.0 * beta_r ** 2 / alpha_r ** 3)
+    g[5] += 1e-2 * (2.0 * beta_r / alpha_r ** 2)
+    return obj, g
+
+
+def fit_perf_params(num_nodes, num_replicas, atomic_bsz,
+                    accum_step_time, optim_step_time):
+    """Fit PerfParams to profiled (accum, optim) step times.
+
+    Uses L-BFGS-B with the analytic gradient of :func:`_obj_and_grad` and the
+    same data-dependent parameter freezing as the reference
+    (goodput.py:151-208): parameters unidentifiable from the observed
+    configurations are pinned so the model stays optimistic about unseen
+    scaling directions.
+    """
+    num_nodes = np.asarray(num_nodes, dtype=float)
+    num_replicas = np.asarray(num_replicas, dtype=float)
+    atomic_bsz = np.asarray(atomic_bsz, dtype=float)
+    accum_step_time = np.asarray(accum_step_time, dtype=float)
+    optim_step_time = np.asarray(optim_step_time, dtype=float)
+
+    params = np.array([1e-1, 1e-2] * 3 + [1.0 + 1e-3])
+    lower = np.array([1e-8, 1e-8] * 3 + [1.0])
+    upper = np.array([np.inf, np.inf] * 3 + [10.0])
+    if len(np.unique(atomic_bsz)) == 1:
+        # Single observed atomic batch size: split accum time evenly between
+        # constant and linear terms (optimistic w.r.t. scaling the bsz up).
+        params[0] = upper[0] = lower[0] = np.mean(accum_step_time) / 2
+    if not np.any(num_nodes > 1):
+        params[2] = upper[2] = lower[2]
+        params[3] = upper[3] = lower[3]
+    if not np.any(np.logical_and(num_nodes == 1, num_replicas > 1)):
+        params[4] = upper[4] = lower[4]
+        params[5] = upper[5] = lower[5]
+    if not np.any(num_replicas > 2):
+        params[3] = upper[3] = lower[3]
+        params[5] = upper[5] = lower[5]
+    bounds = scipy.optimize.Bounds(lower, upper, keep_feasible=True)
+    args = (num_nodes, num_replicas, atomic_bsz,
+            accum_step_time, optim_step_time)
+    result = scipy.optimize.minimize(_obj_and_grad, params, args=args,
+                                     jac=True, bounds=bounds)
+    params = result.x
+    if not np.any(num_nodes > 1):
+        # Prior: inter-node is at least 10% worse than intra-node.
+        params[2] = max(params[2], params[4] * 1.1)
+        params[3] = max(params[3], params[5] * 1.1)
+    return PerfParams(*params)

@@ -1,0 +1,92 @@
+import multiprocessing as mp
+import functools
+import os
+import signal
+import socket
+import tempfile
+
+import pytest
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+
+
+def pick_unused_port():
+    with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def elastic_multiprocessing(func):
+    """Run ``func`` in N forked replica processes with a full ADAPTDL_* env,
+    restarting the whole group with K replicas whenever it returns K.
+
+    This simulates rescale-via-checkpoint-restart entirely on localhost/CPU,
+    mirroring the behavior of the reference test harness
+    (/root/reference/adaptdl/adaptdl/conftest.py:25-100).
+    """
+
+    @functools.wraps(func)
+    def wrapper(*args, **kwargs):
+        num_restarts = 0
+        num_replicas = 1
+        ctx = mp.get_context("fork")
+        with tempfile.TemporaryDirectory() as tmpdir:
+            while num_replicas:
+                assert isinstance(num_replicas, int)
+                master_port = pick_unused_port()
+                queue = ctx.Queue()
+
+                def run(rank):
+                    os.environ["ADAPTDL_CHECKPOINT_PATH"] = str(tmpdir)
+                    os.environ["ADAPTDL_JOB_ID"] = "tmpjob"
+                    os.environ["ADAPTDL_MASTER_ADDR"] = "127.0.0.1"
+                    os.environ["ADAPTDL_MASTER_PORT"] = str(master_port)
+                    os.environ["ADAPTDL_REPLICA_RANK"] = str(rank)
+                    os.environ["ADAPTDL_NUM_REPLICAS"] = str(num_replicas)
+                    os.environ["ADAPTDL_NUM_NODES"] = "1"
+                    os.environ["ADAPTDL_NUM_RESTARTS"] = str(num_restarts)
+                    os.environ["MASTER_ADDR"] = "127.0.0.1"
+                    ret = None
+                    try:
+                        ret = func(*args, **kwargs)
+                    finally:
+                        queue.put((rank, ret))
+
+                procs = [ctx.Process(target=run, args=(rank,))
+                         for rank in range(num_replicas)]
+                for proc in procs:
+                    proc.start()
+                try:
+                    for i in range(num_replicas):
+                        rank, ret = queue.get(timeout=180)
+                        procs[rank].join()
+                        assert procs[rank].exitcode == 0, \
+                            "rank {} exited {}".format(
+                                rank, procs[rank].exitcode)
+                        if i == 0:
+                            num_replicas = ret
+                        assert num_replicas == ret
+                finally:
+                    for proc in procs:
+                        if proc.is_alive():
+                            os.kill(proc.pid, signal.SIGKILL)
+                        proc.join()
+                    queue.close()
+                num_restarts += 1
+
+    return wrapper
+
+
+@pytest.fixture
+def tmp_ckpt_env(tmp_path, monkeypatch):
+    """Single-replica local env with a tmp checkpoint dir."""
+    monkeypatch.setenv("ADAPTDL_CHECKPOINT_PATH", str(tmp_path))
+    monkeypatch.setenv("ADAPTDL_JOB_ID", "tmpjob")
+    monkeypatch.setenv("ADAPTDL_REPLICA_RANK", "0")
+    monkeypatch.setenv("ADAPTDL_NUM_REPLICAS", "1")
+    monkeypatch.setenv("ADAPTDL_NUM_NODES", "1")
+    monkeypatch.setenv("ADAPTDL_NUM_RESTARTS", "0")
+    yield tmp_path
