@@ -1,0 +1,74 @@
+"""Restart-safe epoch loops.
+
+Training programs written against adaptdl_amd must be *idempotent up to
+epoch granularity*: after a checkpoint-restart the program is re-executed
+from the top, and :func:`remaining_epochs_until` skips epochs that already
+finished, while dataloader/accumulator state replays position within the
+current epoch.  Semantics match the reference
+(``/root/reference/adaptdl/adaptdl/torch/epoch.py:96-178``).
+"""
+
+import logging
+import pickle
+
+import adaptdl_amd.checkpoint
+
+LOG = logging.getLogger(__name__)
+
+
+def remaining_epochs_until(epoch):
+    """Iterate over the epochs in ``[finished_epochs(), epoch)``.
+
+    After a checkpoint-restart, previously finished epochs are skipped.
+
+    Raises:
+        RuntimeError: If invoked before a previous epoch loop has ended.
+    """
+    if current_epoch() is not None:
+        raise RuntimeError("overlapping epoch loops detected")
+    if finished_epochs() < epoch:
+        LOG.info("starting at epoch %s", finished_epochs())
+    else:
+        LOG.info("skipping all epochs up to %s", epoch)
+    while finished_epochs() < epoch:
+        _epoch_state().current_epoch = finished_epochs()
+        try:
+            yield current_epoch()
+        finally:
+            # Catches breaks and exceptions escaping the epoch body too.
+            _epoch_state().finished_epochs += 1
+            _epoch_state().current_epoch = None
+
+
+def current_epoch():
+    """The current epoch, or None outside a remaining_epochs_until loop."""
+    return _epoch_state().current_epoch
+
+
+def finished_epochs():
+    """Number of epochs finished across all restarts."""
+    return _epoch_state().finished_epochs
+
+
+class _EpochState(adaptdl_amd.checkpoint.State):
+    def __init__(self):
+        super().__init__(".adaptdl-epoch")
+        self.finished_epochs = 0
+        self.current_epoch = None
+
+    def save(self, fileobj):
+        pickle.dump(self.finished_epochs, fileobj)
+
+    def load(self, fileobj):
+        self.finished_epochs = pickle.load(fileobj)
+
+
+_EPOCH_STATE = None
+
+
+def _epoch_state():
+    global _EPOCH_STATE
+    if _EPOCH_STATE is None:
+        _EPOCH_STATE = _EpochState()
+        adaptdl_amd.checkpoint.load_state(_EPOCH_STATE)
+    return _EPOCH_STATE

@@ -1,0 +1,157 @@
+"""Gradient-noise-scale numerics against hand-computed oracles.
+
+The engine's bucketed statistics must reproduce the reference definitions:
+  local_sqr  = sum over replicas+microbatches of ||g_microbatch||^2
+  total_sqr  = || mean gradient ||^2
+  grad_sqr   = (count * total - local) / (count - 1)
+  grad_var   = (local - total) * scale / (count - 1)
+(reference: adaptdl/torch/gradient_noise_scale.py:242-273)
+"""
+
+import numpy as np
+import pytest
+import torch
+
+import adaptdl_amd.collective as collective
+import adaptdl_amd.env as env
+
+from conftest import elastic_multiprocessing
+
+
+class _FakeADP:
+    """Minimal stand-in for AdaptiveDataParallel in unit tests."""
+    require_backward_grad_sync = True
+
+    def _after_sync(self):
+        pass
+
+
+def _make_model_and_gns(world=1, lr=0.1):
+    from adaptdl_amd.torch.gradient_noise_scale import GradientNoiseScale
+    torch.manual_seed(0)
+    model = torch.nn.Linear(4, 1, bias=False)
+    optim = torch.optim.SGD(model.parameters(), lr=lr)
+    adp = _FakeADP()
+    gns = GradientNoiseScale(adp, optim, num_replicas=world)
+    return model, optim, adp, gns
+
+
+def test_engine_single_replica_differenced():
+    model, optim, adp, gns = _make_model_and_gns(world=1)
+    w = list(model.parameters())[0]
+    xs = [torch.tensor([[1.0, 2.0, 0.5, -1.0]]),
+          torch.tensor([[0.5, -1.0, 2.0, 1.0]])]
+    grads = []
+    for x in xs:
+        out = model(x).sum()
+        out.backward()
+        g = w.grad.detach().clone().numpy().ravel()
+        grads.append(g)
+        gns.reset_accumulation()
+    # For y = w.x, dL/dw = x; each step's grad == x.
+    for g, x in zip(grads, xs):
+        assert np.allclose(g, x.numpy().ravel())
+    # After two steps the differenced estimator has updated:
+    g1, g2 = grads
+    local = (np.sum(g1 ** 2) + np.sum(g2 ** 2)) / 2
+    total = np.sum(((g1 + g2) / 2) ** 2)
+    grad_sqr = 2 * total - local
+    grad_var = (local - total) * 2
+    assert gns._state["biased"]
+    assert np.isclose(gns._state["sqr_avg"][0], grad_sqr, rtol=1e-6)
+    assert np.isclose(gns._state["var_avg"][0], grad_var, rtol=1e-6)
+
+
+def test_engine_accumulation_stats():
+    model, optim, adp, gns = _make_model_and_gns(world=1)
+    w = list(model.parameters())[0]
+    xs = [torch.tensor([[1.0, 2.0, 0.5, -1.0]]),
+          torch.tensor([[0.5, -1.0, 2.0, 1.0]]),
+          torch.tensor([[2.0, 0.0, -1.0, 0.5]])]
+    # Three microbatches accumulated into one step.
+    for i, x in enumerate(xs):
+        gns.engine.require_sync = (i == len(xs) - 1)
+        adp.require_backward_grad_sync = gns.engine.require_sync
+        model(x).sum().backward()
+    gs = [x.numpy().ravel() for x in xs]
+    count = 3
+    local = sum(np.sum(g ** 2) for g in gs) / count
+    mean = sum(gs) / count
+    total = np.sum(mean ** 2)
+    # Gradient must equal the mean of microbatch gradients.
+    assert np.allclose(w.grad.detach().numpy().ravel(), mean, rtol=1e-6)
+    scale = gns.accum_scale * count
+    grad_sqr = (count * total - local) / (count - 1)
+    grad_var = (local - total) * scale / (count - 1)
+    assert not gns._state["biased"]
+    assert np.isclose(gns._state["sqr_avg"][0], grad_sqr, rtol=1e-5)
+    assert np.isclose(gns._state["var_avg"][0], grad_var, rtol=1e-5)
+    assert gns.should_zero_grad
+    gns.reset_accumulation()
+    assert np.allclose(w.grad.detach().numpy(), 0.0)
+
+
+@elastic_multiprocessing
+def _run_gns_two_replicas():
+    import torch.distributed
+    from adaptdl_amd.torch.gradient_noise_scale import GradientNoiseScale
+    collective.initialize()
+    if env.num_restarts() == 0:
+        collective.teardown()
+        return 2
+    torch.distributed.init_process_group(
+        "gloo", init_method="tcp://127.0.0.1:{}".format(
+            collective.broadcast(_free_port())),
+        world_size=env.num_replicas(), rank=env.replica_rank())
+    torch.manual_seed(0)
+    model = torch.nn.Linear(4, 1, bias=False)
+    optim = torch.optim.SGD(model.parameters(), lr=0.1)
+    adp = _FakeADP()
+    gns = GradientNoiseScale(adp, optim)
+    w = list(model.parameters())[0]
+    data = {0: torch.tensor([[1.0, 2.0, 0.5, -1.0]]),
+            1: torch.tensor([[0.5, -1.0, 2.0, 1.0]])}
+    x = data[env.replica_rank()]
+    model(x).sum().backward()
+    g0 = data[0].numpy().ravel()
+    g1 = data[1].numpy().ravel()
+    mean = (g0 + g1) / 2
+    # Synchronized gradient must be the replica mean.
+    assert np.allclose(w.grad.detach().numpy().ravel(), mean, atol=1e-6)
+    count = 2
+    local = (np.sum(g0 ** 2) + np.sum(g1 ** 2)) / count
+    total = np.sum(mean ** 2)
+    grad_sqr = count * total - local
+    grad_var = (local - total) * 2
+    assert np.isclose(gns._state["sqr_avg"][0], grad_sqr, rtol=1e-5)
+    assert np.isclose(gns._state["var_avg"][0], grad_var, rtol=1e-5)
+    torch.distributed.destroy_process_group()
+    collective.teardown()
+    return 0
+
+
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def test_gns_two_replicas():
+    _run_gns_two_replicas()
+
+
+def test_gain_formula():
+    model, optim, adp, gns = _make_model_and_gns(world=1)
+    gns._state["sqr_avg"] = np.array([0.5])
+    gns._state["var_avg"] = np.array([2.0])
+    expected = (2.0 + 0.5) / (2.0 / 4 + 0.5)
+    assert np.isclose(gns.gain(4.0), expected)
+
+
+def test_nan_gradient_skipped():
+    model, optim, adp, gns = _make_model_and_gns(world=1)
+    sqr_before = gns._state["sqr_avg"].copy()
+    x = torch.tensor([[float("nan"), 1.0, 1.0, 1.0]])
+    model(x).sum().backward()
+    assert np.allclose(gns._state["sqr_avg"], sqr_before)
