@@ -35,7 +35,15 @@ def master_addr():
 
 def master_port():
     """Port used by the control-plane object collectives (rank 0 listens)."""
-    return int(os.getenv("ADAPTDL_MASTER_PORT", "0"))
+    port = os.getenv("ADAPTDL_MASTER_PORT")
+    if port is not None:
+        return int(port)
+    # Under torchrun, derive a deterministic side port from MASTER_PORT
+    # (which is used by the torch.distributed store itself).
+    tr_port = os.getenv("MASTER_PORT")
+    if tr_port is not None:
+        return int(tr_port) + 17
+    return 0
 
 
 def replica_rank():

@@ -1,0 +1,301 @@
+// CDNA4 (gfx950 / MI355X) kernels for the adaptdl_amd gradient hot path.
+//
+// These implement the fused operations of SURVEY.md §7 N1/N3: one
+// bandwidth-bound pass over each flat gradient bucket computes the scale
+// (gradient averaging) together with the fp64 sum-of-squares statistics
+// that drive the gradient noise scale, replacing the reference's
+// per-parameter `pow(2).sum(dtype=float64)` hook kernels
+// (reference: adaptdl/torch/gradient_noise_scale.py:33-39,181-182) and
+// `grad.div_(accum_count)` passes (:228).
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
+// - wavefront = 64; block = 256 threads (4 waves); grid-stride loop capped
+//   at ~2048 blocks (256 CU * 8 blocks) for memory-bound ops (Guideline 11)
+// - float4 vectorized global access (16 B/lane; Guideline 13), with
+//   scalar peel for unaligned segment starts and tails
+// - fp64 accumulation per thread -> wave __shfl_down reduce -> LDS partial
+//   per wave -> one global fp64 atomicAdd per block (Guideline 12)
+// - elementwise writes (scale / prev-copy) fused into the same pass so
+//   gradient bytes are touched exactly once more than strictly necessary
+
+#include <hip/hip_runtime.h>
+
+#define WAVE 64
+#define BLOCK 256
+#define MAX_BLOCKS 2048
+
+__device__ __forceinline__ double wave_reduce_add(double v) {
+    #pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1) {
+        v += __shfl_down(v, off, WAVE);
+    }
+    return v;
+}
+
+// Block-level reduction: per-wave shuffle, LDS partials, lane-0 atomic.
+__device__ __forceinline__ void block_reduce_atomic(double v, double* out) {
+    __shared__ double partials[BLOCK / WAVE];
+    v = wave_reduce_add(v);
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    if (lane == 0) partials[wid] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        double s = 0.0;
+        #pragma unroll
+        for (int i = 0; i < BLOCK / WAVE; ++i) s += partials[i];
+        atomicAdd(out, s);
+    }
+}
+
+// Grid-stride traversal over n floats with float4 vector body.
+// Each functor F provides: double elem(long i) applied per scalar index.
+// We specialize the main kernels directly instead of templating over
+// functors to keep the vector path explicit.
+
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum(
+        const float* __restrict__ x, long n, double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    // Head: scalar until 16-byte aligned.
+    long head = (16 - ((size_t)x & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        double v = (double)x[j];
+        acc += v * v;
+    }
+    const float4* xv = (const float4*)(x + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 v = xv[j];
+        acc += (double)v.x * v.x + (double)v.y * v.y +
+               (double)v.z * v.z + (double)v.w * v.w;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        double v = (double)x[j];
+        acc += v * v;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK) void k_scale_sqsum(
+        float* __restrict__ x, long n, float scale,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = (16 - ((size_t)x & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        float v = x[j] * scale;
+        x[j] = v;
+        acc += (double)v * v;
+    }
+    float4* xv = (float4*)(x + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 v = xv[j];
+        v.x *= scale; v.y *= scale; v.z *= scale; v.w *= scale;
+        xv[j] = v;
+        acc += (double)v.x * v.x + (double)v.y * v.y +
+               (double)v.z * v.z + (double)v.w * v.w;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        float v = x[j] * scale;
+        x[j] = v;
+        acc += (double)v * v;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+// out += sum((cur - prev)^2); prev = cur.  (accumulation microbatch stat)
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_diff_update(
+        const float* __restrict__ cur, float* __restrict__ prev, long n,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    // cur and prev share alignment (same bucket layout).
+    long head = (16 - ((size_t)cur & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        float c = cur[j];
+        double d = (double)(c - prev[j]);
+        prev[j] = c;
+        acc += d * d;
+    }
+    const float4* cv = (const float4*)(cur + head);
+    float4* pv = (float4*)(prev + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 c = cv[j];
+        float4 p = pv[j];
+        double d0 = (double)(c.x - p.x), d1 = (double)(c.y - p.y);
+        double d2 = (double)(c.z - p.z), d3 = (double)(c.w - p.w);
+        pv[j] = c;
+        acc += d0 * d0 + d1 * d1 + d2 * d2 + d3 * d3;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        float c = cur[j];
+        double d = (double)(c - prev[j]);
+        prev[j] = c;
+        acc += d * d;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+// out += sum(((cur + prev) / 2)^2)   (differenced single-sample estimator)
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_avg(
+        const float* __restrict__ cur, const float* __restrict__ prev,
+        long n, double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = (16 - ((size_t)cur & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        double a = 0.5 * ((double)cur[j] + (double)prev[j]);
+        acc += a * a;
+    }
+    const float4* cv = (const float4*)(cur + head);
+    const float4* pv = (const float4*)(prev + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 c = cv[j];
+        float4 p = pv[j];
+        double a0 = 0.5 * ((double)c.x + p.x), a1 = 0.5 * ((double)c.y + p.y);
+        double a2 = 0.5 * ((double)c.z + p.z), a3 = 0.5 * ((double)c.w + p.w);
+        acc += a0 * a0 + a1 * a1 + a2 * a2 + a3 * a3;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        double a = 0.5 * ((double)cur[j] + (double)prev[j]);
+        acc += a * a;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+// out += sum((g / pinv)^2), pinv = sqrt(v / corr) + eps (Adam precondition;
+// reference AdamGradientNoiseScale._calculate_preconditioner)
+extern "C" __global__ __launch_bounds__(BLOCK) void k_precond_sqsum(
+        const float* __restrict__ g, const float* __restrict__ v, long n,
+        float inv_corr_sqrt, float eps, double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    for (long j = i; j < n; j += stride) {
+        float pinv = sqrtf(v[j]) * inv_corr_sqrt + eps;
+        double q = (double)(g[j] / pinv);
+        acc += q * q;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+// ---- fused flat-bucket optimizers -----------------------------------
+
+// SGD with momentum on one flat bucket (param/grad/momentum share layout).
+extern "C" __global__ __launch_bounds__(BLOCK) void k_fused_sgd(
+        float* __restrict__ p, const float* __restrict__ g,
+        float* __restrict__ m, long n, float lr, float momentum,
+        float weight_decay, float dampening, int nesterov, int has_momentum) {
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    for (long j = i; j < n; j += stride) {
+        float dp = g[j];
+        float pj = p[j];
+        if (weight_decay != 0.f) dp = fmaf(weight_decay, pj, dp);
+        if (has_momentum) {
+            float mj = fmaf(momentum, m[j], (1.f - dampening) * dp);
+            m[j] = mj;
+            dp = nesterov ? fmaf(momentum, mj, dp) : mj;
+        }
+        p[j] = fmaf(-lr, dp, pj);
+    }
+}
+
+// Adam / AdamW on one flat bucket.
+extern "C" __global__ __launch_bounds__(BLOCK) void k_fused_adamw(
+        float* __restrict__ p, const float* __restrict__ g,
+        float* __restrict__ m, float* __restrict__ v, long n, float lr,
+        float beta1, float beta2, float eps, float weight_decay,
+        float bias1, float bias2, int adam_mode) {
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    const float step_size = lr / bias1;
+    const float inv_bias2_sqrt = rsqrtf(bias2);
+    for (long j = i; j < n; j += stride) {
+        float gj = g[j];
+        float pj = p[j];
+        if (adam_mode && weight_decay != 0.f) gj = fmaf(weight_decay, pj, gj);
+        else if (!adam_mode && weight_decay != 0.f)
+            pj *= (1.f - lr * weight_decay);
+        float mj = fmaf(beta1, m[j], (1.f - beta1) * gj);
+        float vj = fmaf(beta2, v[j], (1.f - beta2) * gj * gj);
+        m[j] = mj;
+        v[j] = vj;
+        float denom = sqrtf(vj) * inv_bias2_sqrt + eps;
+        p[j] = fmaf(-step_size, mj / denom, pj);
+    }
+}
+
+// ---- host-side launchers (called from bindings.cpp, which is compiled
+// by the host compiler and cannot launch kernels itself) ---------------
+
+static inline dim3 grid_for(long n) {
+    long blocks = (n + BLOCK - 1) / BLOCK;
+    if (blocks > MAX_BLOCKS) blocks = MAX_BLOCKS;
+    if (blocks < 1) blocks = 1;
+    return dim3((unsigned)blocks);
+}
+
+extern "C" void launch_sqsum(const float* x, long n, double* out,
+                             hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum, grid_for(n), dim3(BLOCK), 0, s, x, n, out);
+}
+
+extern "C" void launch_scale_sqsum(float* x, long n, float scale,
+                                   double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_scale_sqsum, grid_for(n), dim3(BLOCK), 0, s, x, n,
+                       scale, out);
+}
+
+extern "C" void launch_sqsum_diff_update(const float* cur, float* prev,
+                                         long n, double* out,
+                                         hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_diff_update, grid_for(n), dim3(BLOCK), 0, s,
+                       cur, prev, n, out);
+}
+
+extern "C" void launch_sqsum_avg(const float* cur, const float* prev, long n,
+                                 double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_avg, grid_for(n), dim3(BLOCK), 0, s, cur,
+                       prev, n, out);
+}
+
+extern "C" void launch_precond_sqsum(const float* g, const float* v, long n,
+                                     float inv_corr_sqrt, float eps,
+                                     double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_precond_sqsum, grid_for(n), dim3(BLOCK), 0, s, g, v,
+                       n, inv_corr_sqrt, eps, out);
+}
+
+extern "C" void launch_fused_sgd(float* p, const float* g, float* m, long n,
+                                 float lr, float momentum,
+                                 float weight_decay, float dampening,
+                                 int nesterov, int has_momentum,
+                                 hipStream_t s) {
+    hipLaunchKernelGGL(k_fused_sgd, grid_for(n), dim3(BLOCK), 0, s, p, g, m,
+                       n, lr, momentum, weight_decay, dampening, nesterov,
+                       has_momentum);
+}
+
+extern "C" void launch_fused_adamw(float* p, const float* g, float* m,
+                                   float* v, long n, float lr, float beta1,
+                                   float beta2, float eps,
+                                   float weight_decay, float bias1,
+                                   float bias2, int adam_mode,
+                                   hipStream_t s) {
+    hipLaunchKernelGGL(k_fused_adamw, grid_for(n), dim3(BLOCK), 0, s, p, g,
+                       m, v, n, lr, beta1, beta2, eps, weight_decay, bias1,
+                       bias2, adam_mode);
+}

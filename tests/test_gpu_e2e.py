@@ -1,0 +1,65 @@
+"""End-to-end GPU tests: ResNet-18 training step through the full stack."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_resnet18_train_steps(tmp_ckpt_env):
+    import adaptdl_amd.collective as collective
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd import ops
+    from adaptdl_amd.models import ResNet18
+
+    assert ops.has_extension()
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+    device = torch.device("cuda:0")
+    torch.manual_seed(0)
+    model = ResNet18().to(device)
+    optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    adp = adl.AdaptiveDataParallel(model, optim)
+
+    dataset = torch.utils.data.TensorDataset(torch.arange(256))
+    loader = adl.AdaptiveDataLoader(dataset, batch_size=64)
+    losses = []
+    x = torch.randn(64, 3, 32, 32, device=device)
+    y = torch.randint(0, 10, (64,), device=device)
+    for epoch in adl.remaining_epochs_until(1):
+        for _ in loader:
+            optim.zero_grad()
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                loss = torch.nn.functional.cross_entropy(adp(x), y)
+            loss.backward()
+            optim.step()
+            losses.append(loss.item())
+    assert len(losses) == 4
+    assert all(torch.isfinite(torch.tensor(v)) for v in losses)
+    # Training on a fixed batch must reduce the loss.
+    assert losses[-1] < losses[0]
+    # GNS state was updated through the fused kernels.
+    assert adp.gns.sqr_avg() >= 0.0
+    assert adp.gns.var_avg() >= 0.0
+
+
+def test_bucket_views_on_gpu():
+    """param.grad must be views into flat buckets (zero-copy design)."""
+    from adaptdl_amd.torch.gradient_noise_scale import GradientNoiseScale
+
+    class ADP:
+        require_backward_grad_sync = True
+
+        def _after_sync(self):
+            pass
+
+    model = torch.nn.Sequential(
+        torch.nn.Linear(32, 32), torch.nn.ReLU(),
+        torch.nn.Linear(32, 4)).cuda()
+    optim = torch.optim.SGD(model.parameters(), lr=0.1)
+    gns = GradientNoiseScale(ADP(), optim, num_replicas=1)
+    x = torch.randn(8, 32, device="cuda")
+    model(x).sum().backward()
+    for bucket in gns.engine.buckets:
+        for p, off, n in bucket.segments:
+            assert p.grad.data_ptr() == bucket.flat[off:off + n].data_ptr()
