@@ -121,7 +121,7 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_diff_update(
     if (head > n) head = n;
     for (long j = i; j < head; j += stride) {
         float c = cur[j];
-        double d = (double)(c - prev[j]);
+        double d = (double)c - (double)prev[j];
         prev[j] = c;
         acc += d * d;
     }
@@ -131,14 +131,14 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_diff_update(
     for (long j = i; j < nv; j += stride) {
         float4 c = cv[j];
         float4 p = pv[j];
-        double d0 = (double)(c.x - p.x), d1 = (double)(c.y - p.y);
-        double d2 = (double)(c.z - p.z), d3 = (double)(c.w - p.w);
+        double d0 = (double)c.x - (double)p.x, d1 = (double)c.y - (double)p.y;
+        double d2 = (double)c.z - (double)p.z, d3 = (double)c.w - (double)p.w;
         pv[j] = c;
         acc += d0 * d0 + d1 * d1 + d2 * d2 + d3 * d3;
     }
     for (long j = head + nv * 4 + i; j < n; j += stride) {
         float c = cur[j];
-        double d = (double)(c - prev[j]);
+        double d = (double)c - (double)prev[j];
         prev[j] = c;
         acc += d * d;
     }

@@ -81,8 +81,11 @@ def main():
     pool_y = torch.randint(0, 10, (args.pool,), generator=g).to(device)
 
     dataset = SyntheticIndices(args.dataset_size)
+    # drop_last keeps conv shapes static (a partial final batch would
+    # trigger a multi-second MIOpen kernel search for the odd shape).
     loader = adl.AdaptiveDataLoader(dataset, batch_size=args.init_batch,
-                                    collate_fn=_collate, num_workers=0)
+                                    collate_fn=_collate, num_workers=0,
+                                    drop_last=True)
     lo, hi = (int(v) for v in args.bounds.split(","))
     if args.max_batch > 0:
         loader.autoscale_batch_size(args.max_batch,
