@@ -50,11 +50,33 @@ def _require_ext(t):
     return None
 
 
+def _flat(t):
+    """Flatten to a storage-order VIEW (never a copy).
+
+    ``view(-1)`` only works for default-contiguous tensors; dense tensors
+    in a non-default stride order (channels_last params / Adam moments)
+    are flattened with as_strided over the same storage so in-place
+    updates still hit the original tensor.  Element order follows storage
+    order, which is what the flat-bucket segments use too.
+    """
+    if t.is_contiguous():
+        return t.view(-1)
+    if t.dim() in (4, 5) and (
+            t.is_contiguous(memory_format=torch.channels_last)
+            or (t.dim() == 5 and
+                t.is_contiguous(memory_format=torch.channels_last_3d))):
+        return t.detach().as_strided((t.numel(),), (1,),
+                                     t.storage_offset())
+    raise RuntimeError(
+        "cannot flatten non-dense tensor of shape {} / strides {} without "
+        "a copy".format(tuple(t.shape), tuple(t.stride())))
+
+
 def sqsum(t, out):
     """out += sum(t.double() ** 2)  (out: 0-dim float64 on same device)."""
     ext = _require_ext(t)
     if ext is not None:
-        ext.sqsum(t.view(-1), out)
+        ext.sqsum(_flat(t), out)
     else:
         out.add_(t.double().pow(2).sum())
 
@@ -63,7 +85,7 @@ def scale_and_sqsum(t, scale, out):
     """t *= scale; out += sum(t.double() ** 2), in one pass."""
     ext = _require_ext(t)
     if ext is not None:
-        ext.scale_and_sqsum(t.view(-1), float(scale), out)
+        ext.scale_and_sqsum(_flat(t), float(scale), out)
     else:
         if scale != 1.0:
             t.mul_(scale)
@@ -79,7 +101,7 @@ def sqsum_diff_update(cur, prev, out):
     """
     ext = _require_ext(cur)
     if ext is not None:
-        ext.sqsum_diff_update(cur.view(-1), prev.view(-1), out)
+        ext.sqsum_diff_update(_flat(cur), _flat(prev), out)
     else:
         out.add_((cur.double() - prev.double()).pow(2).sum())
         prev.copy_(cur)
@@ -89,7 +111,7 @@ def sqsum_avg(cur, prev, out):
     """out += sum(((cur + prev) / 2)**2) (differenced GNS estimator)."""
     ext = _require_ext(cur)
     if ext is not None:
-        ext.sqsum_avg(cur.view(-1), prev.view(-1), out)
+        ext.sqsum_avg(_flat(cur), _flat(prev), out)
     else:
         out.add_(((cur.double() + prev.double()) / 2).pow(2).sum())
 
@@ -99,8 +121,8 @@ def fused_sgd_step(param, grad, momentum_buf, lr, momentum, weight_decay,
     """Flat-bucket SGD update (single fused pass on GPU)."""
     ext = _require_ext(param)
     if ext is not None:
-        ext.fused_sgd(param.view(-1), grad.view(-1),
-                      momentum_buf.view(-1) if momentum_buf is not None
+        ext.fused_sgd(_flat(param), _flat(grad),
+                      _flat(momentum_buf) if momentum_buf is not None
                       else torch.empty(0, device=param.device,
                                        dtype=param.dtype),
                       float(lr), float(momentum), float(weight_decay),
@@ -126,8 +148,8 @@ def fused_adamw_step(param, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
     """
     ext = _require_ext(param)
     if ext is not None:
-        ext.fused_adamw(param.view(-1), grad.view(-1), exp_avg.view(-1),
-                        exp_avg_sq.view(-1), float(lr), float(beta1),
+        ext.fused_adamw(_flat(param), _flat(grad), _flat(exp_avg),
+                        _flat(exp_avg_sq), float(lr), float(beta1),
                         float(beta2), float(eps), float(weight_decay),
                         int(step), bool(adam_mode))
     else:
@@ -148,9 +170,9 @@ def precond_sqsum(grad, exp_avg_sq, beta2, eps, step, out):
     pinv = sqrt(exp_avg_sq / (1 - beta2**step)) + eps  (fp64 accumulate)."""
     ext = _require_ext(grad)
     if ext is not None:
-        ext.precond_sqsum(grad.view(-1), exp_avg_sq.view(-1), float(beta2),
+        ext.precond_sqsum(_flat(grad), _flat(exp_avg_sq), float(beta2),
                           float(eps), int(step), out)
     else:
         corr = 1.0 - beta2 ** step
-        pinv = (exp_avg_sq.double() / corr).sqrt().add_(eps)
-        out.add_((grad.double() / pinv).pow(2).sum())
+        pinv = (_flat(exp_avg_sq).double() / corr).sqrt().add_(eps)
+        out.add_((_flat(grad).double() / pinv).pow(2).sum())

@@ -37,6 +37,27 @@ LOG = logging.getLogger(__name__)
 _DEFAULT_BUCKET_CAP_MB = 32
 
 
+def _segment_view(seg, p):
+    """View a flat bucket segment with the same memory layout as ``p``.
+
+    For channels_last (and channels_last_3d) parameters the view is
+    strided to match, so autograd's in-place accumulation into ``.grad``
+    is layout-identical to the produced gradient (no NCHW<->NHWC
+    conversion kernels on the hot path).  The flat buffer then simply
+    holds the segment in the parameter's storage order — element order is
+    irrelevant to all-reduce and the sum-of-squares statistics.
+    """
+    if p.dim() == 4 and not p.is_contiguous() and \
+            p.is_contiguous(memory_format=torch.channels_last):
+        n_, c, h, w = p.shape
+        return seg.view(n_, h, w, c).permute(0, 3, 1, 2)
+    if p.dim() == 5 and not p.is_contiguous() and \
+            p.is_contiguous(memory_format=torch.channels_last_3d):
+        n_, c, d, h, w = p.shape
+        return seg.view(n_, d, h, w, c).permute(0, 4, 1, 2, 3)
+    return seg.view(p.shape)
+
+
 class Bucket(object):
     __slots__ = ["group_idx", "flat", "segments", "ready", "work", "prev"]
 
@@ -55,7 +76,7 @@ class Bucket(object):
 
     def param_views(self, flat=None):
         flat = self.flat if flat is None else flat
-        return [(p, flat[off:off + n].view(p.shape))
+        return [(p, _segment_view(flat[off:off + n], p))
                 for p, off, n in self.segments]
 
 
@@ -141,7 +162,7 @@ class GradSyncEngine(object):
         for p in params:
             n = p.numel()
             segments.append((p, offset, n))
-            p.grad = flat[offset:offset + n].view(p.shape)
+            p.grad = _segment_view(flat[offset:offset + n], p)
             offset += n
         bucket = Bucket(gidx, flat, segments)
         self.buckets.append(bucket)

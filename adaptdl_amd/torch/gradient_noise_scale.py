@@ -299,7 +299,7 @@ class AdamGradientNoiseScale(GradientNoiseScale):
             if step < self._PRECOND_MIN_STEPS or "exp_avg_sq" not in state:
                 ops.sqsum(seg, out)
             else:
-                ops.precond_sqsum(seg, state["exp_avg_sq"].view(-1), beta2,
+                ops.precond_sqsum(seg, state["exp_avg_sq"], beta2,
                                   eps, step, out)
 
     def _bucket_sqsum(self, bucket, flat, out):

@@ -55,6 +55,10 @@ def main():
     parser.add_argument("--bounds", type=str, default="32,1024")
     parser.add_argument("--dataset-size", type=int, default=50000)
     parser.add_argument("--pool", type=int, default=8192)
+    parser.add_argument("--memory-format", choices=["channels_last", "nchw"],
+                        default="channels_last",
+                        help="channels_last keeps MIOpen on its native NHWC "
+                             "igemm kernels with no transpose kernels")
     args = parser.parse_args()
     args.warmup = max(args.warmup, 5)
 
@@ -71,13 +75,18 @@ def main():
         else torch.device("cpu")
 
     torch.manual_seed(1234)  # same random init on every rank
+    channels_last = use_gpu and args.memory_format == "channels_last"
     model = ResNet18().to(device)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                             weight_decay=5e-4)
     adp = adl.AdaptiveDataParallel(model, optim)
 
     g = torch.Generator(device="cpu").manual_seed(4321 + rank)
     pool_x = torch.randn(args.pool, 3, 32, 32, generator=g).to(device)
+    if channels_last:
+        pool_x = pool_x.contiguous(memory_format=torch.channels_last)
     pool_y = torch.randint(0, 10, (args.pool,), generator=g).to(device)
 
     dataset = SyntheticIndices(args.dataset_size)
@@ -95,6 +104,8 @@ def main():
     def train_step(idx):
         idx = (idx % args.pool).to(device, non_blocking=True)
         x = pool_x[idx]
+        if channels_last:
+            x = x.contiguous(memory_format=torch.channels_last)
         y = pool_y[idx]
         optim.zero_grad()
         if use_gpu:

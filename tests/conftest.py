@@ -40,6 +40,11 @@ def elastic_multiprocessing(func):
                 queue = ctx.Queue()
 
                 def run(rank):
+                    # Fork-safety: the parent may have used torch's OpenMP
+                    # pool already; a forked child inheriting it deadlocks
+                    # on its first parallel region.  Run single-threaded.
+                    import torch
+                    torch.set_num_threads(1)
                     os.environ["ADAPTDL_CHECKPOINT_PATH"] = str(tmpdir)
                     os.environ["ADAPTDL_JOB_ID"] = "tmpjob"
                     os.environ["ADAPTDL_MASTER_ADDR"] = "127.0.0.1"
