@@ -1,0 +1,8 @@
+"""In-process scheduling for a single 8xMI355X node (and beyond).
+
+The reference runs Pollux as a Kubernetes control plane
+(/root/reference/sched/adaptdl_sched/); here the same policy runs as an
+in-process allocator thread (adaptdl_amd.sched.allocator) next to a local
+job controller (adaptdl_amd.sched.controller) that implements
+checkpoint-restart elasticity with worker processes, one per GPU.
+"""
