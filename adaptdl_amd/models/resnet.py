@@ -9,6 +9,13 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from adaptdl_amd.torch.layers import FusedBatchNormAct2d
+
+
+def _bn(planes, relu=False):
+    """BatchNorm2d with optional fused ReLU (CDNA4 kernel on GPU)."""
+    return FusedBatchNormAct2d(planes, relu=relu)
+
 
 class BasicBlock(nn.Module):
     expansion = 1
@@ -17,19 +24,19 @@ class BasicBlock(nn.Module):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride,
                                padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = _bn(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = _bn(planes)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
                 nn.Conv2d(in_planes, self.expansion * planes, 1,
                           stride=stride, bias=False),
-                nn.BatchNorm2d(self.expansion * planes))
+                _bn(self.expansion * planes))
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
         out = out + self.shortcut(x)
         return F.relu(out)
@@ -41,23 +48,23 @@ class Bottleneck(nn.Module):
     def __init__(self, in_planes, planes, stride=1):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = _bn(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1,
                                bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = _bn(planes, relu=True)
         self.conv3 = nn.Conv2d(planes, self.expansion * planes, 1,
                                bias=False)
-        self.bn3 = nn.BatchNorm2d(self.expansion * planes)
+        self.bn3 = _bn(self.expansion * planes)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
                 nn.Conv2d(in_planes, self.expansion * planes, 1,
                           stride=stride, bias=False),
-                nn.BatchNorm2d(self.expansion * planes))
+                _bn(self.expansion * planes))
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = F.relu(self.bn2(self.conv2(out)))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         out = self.bn3(self.conv3(out))
         out = out + self.shortcut(x)
         return F.relu(out)
@@ -70,7 +77,7 @@ class CifarResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(3, 64, 3, stride=1, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = _bn(64, relu=True)
         self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
         self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
         self.layer3 = self._make_layer(block, 256, num_blocks[2], 2)
@@ -86,7 +93,7 @@ class CifarResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         out = self.layer4(self.layer3(self.layer2(self.layer1(out))))
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.linear(out)
@@ -99,7 +106,7 @@ class ImageNetResNet(nn.Module):
         super().__init__()
         self.in_planes = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = _bn(64, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(block, 64, num_blocks[0], 1)
         self.layer2 = self._make_layer(block, 128, num_blocks[1], 2)
@@ -110,7 +117,7 @@ class ImageNetResNet(nn.Module):
     _make_layer = CifarResNet._make_layer
 
     def forward(self, x):
-        out = self.maxpool(F.relu(self.bn1(self.conv1(x))))
+        out = self.maxpool(self.bn1(self.conv1(x)))
         out = self.layer4(self.layer3(self.layer2(self.layer1(out))))
         out = F.adaptive_avg_pool2d(out, 1).flatten(1)
         return self.fc(out)

@@ -24,6 +24,14 @@ void launch_fused_sgd(float*, const float*, float*, long, float, float,
 void launch_fused_adamw(float*, const float*, float*, float*, long, float,
                         float, float, float, float, float, float, int,
                         hipStream_t);
+void launch_bn_fwd(const unsigned short*, unsigned short*, long, int,
+                   const float*, const float*, float*, float*, float, float,
+                   int, int, float*, float*, float*, float*, float*,
+                   hipStream_t);
+void launch_bn_bwd(const unsigned short*, const unsigned short*,
+                   unsigned short*, long, int, const float*, const float*,
+                   const float*, const float*, const float*, int, int,
+                   float*, float*, float*, float*, hipStream_t);
 }
 
 namespace {
@@ -123,6 +131,84 @@ void fused_adamw(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        stream());
 }
 
+// ---- fused NHWC BatchNorm(+ReLU) -------------------------------------
+
+void check_bn_x(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name,
+                " must be bfloat16");
+    TORCH_CHECK(t.dim() == 4, name, " must be 4-D NCHW-shaped");
+    TORCH_CHECK(t.is_contiguous(at::MemoryFormat::ChannelsLast), name,
+                " must be channels_last");
+}
+
+void check_vecf(const torch::Tensor& t, long n, const char* name) {
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kFloat32 &&
+                t.is_contiguous() && t.numel() == n,
+                name, " must be a contiguous fp32 GPU tensor of size ", n);
+}
+
+// y = [relu](bn(x)); fills sums(2C ws), save_mean/save_rstd/scale/shift(C).
+void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
+            torch::Tensor beta, torch::Tensor running_mean,
+            torch::Tensor running_var, double momentum, double eps,
+            bool train, bool relu, torch::Tensor sums,
+            torch::Tensor save_mean, torch::Tensor save_rstd,
+            torch::Tensor scale, torch::Tensor shift) {
+    check_bn_x(x, "x"); check_bn_x(y, "y");
+    const long c = x.size(1);
+    const long m = x.numel() / c;
+    TORCH_CHECK(c % 8 == 0 && c / 8 <= 256,
+                "bn_fwd requires C % 8 == 0 and C <= 2048, got C=", c);
+    check_vecf(gamma, c, "gamma"); check_vecf(beta, c, "beta");
+    check_vecf(sums, 2 * c, "sums");
+    check_vecf(save_mean, c, "save_mean");
+    check_vecf(save_rstd, c, "save_rstd");
+    check_vecf(scale, c, "scale"); check_vecf(shift, c, "shift");
+    float* rm = nullptr;
+    float* rv = nullptr;
+    if (running_mean.defined() && running_mean.numel() > 0) {
+        check_vecf(running_mean, c, "running_mean");
+        check_vecf(running_var, c, "running_var");
+        rm = running_mean.data_ptr<float>();
+        rv = running_var.data_ptr<float>();
+    } else {
+        TORCH_CHECK(train, "eval-mode bn_fwd requires running statistics");
+    }
+    launch_bn_fwd((const unsigned short*)x.data_ptr(),
+                  (unsigned short*)y.data_ptr(), m, (int)c,
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(), rm, rv,
+                  (float)momentum, (float)eps, (int)train, (int)relu,
+                  sums.data_ptr<float>(), save_mean.data_ptr<float>(),
+                  save_rstd.data_ptr<float>(), scale.data_ptr<float>(),
+                  shift.data_ptr<float>(), stream());
+}
+
+void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
+            torch::Tensor gamma, torch::Tensor save_mean,
+            torch::Tensor save_rstd, torch::Tensor scale,
+            torch::Tensor shift, bool train, bool relu, torch::Tensor sums,
+            torch::Tensor dgamma, torch::Tensor dbeta, torch::Tensor pqr) {
+    check_bn_x(x, "x"); check_bn_x(dy, "dy"); check_bn_x(dx, "dx");
+    const long c = x.size(1);
+    const long m = x.numel() / c;
+    check_vecf(gamma, c, "gamma");
+    check_vecf(save_mean, c, "save_mean");
+    check_vecf(save_rstd, c, "save_rstd");
+    check_vecf(scale, c, "scale"); check_vecf(shift, c, "shift");
+    check_vecf(sums, 2 * c, "sums");
+    check_vecf(dgamma, c, "dgamma"); check_vecf(dbeta, c, "dbeta");
+    check_vecf(pqr, 3 * c, "pqr");
+    launch_bn_bwd((const unsigned short*)x.data_ptr(),
+                  (const unsigned short*)dy.data_ptr(),
+                  (unsigned short*)dx.data_ptr(), m, (int)c,
+                  gamma.data_ptr<float>(), save_mean.data_ptr<float>(),
+                  save_rstd.data_ptr<float>(), scale.data_ptr<float>(),
+                  shift.data_ptr<float>(), (int)train, (int)relu,
+                  sums.data_ptr<float>(), dgamma.data_ptr<float>(),
+                  dbeta.data_ptr<float>(), pqr.data_ptr<float>(), stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -136,4 +222,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             "out += sum((g/pinv)^2), Adam preconditioner");
     mod.def("fused_sgd", &fused_sgd, "fused flat-bucket SGD step");
     mod.def("fused_adamw", &fused_adamw, "fused flat-bucket Adam(W) step");
+    mod.def("bn_fwd", &bn_fwd, "fused NHWC bf16 BatchNorm(+ReLU) forward");
+    mod.def("bn_bwd", &bn_bwd, "fused NHWC bf16 BatchNorm(+ReLU) backward");
 }
