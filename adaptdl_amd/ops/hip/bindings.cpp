@@ -26,12 +26,22 @@ void launch_fused_adamw(float*, const float*, float*, float*, long, float,
                         hipStream_t);
 void launch_bn_fwd(const unsigned short*, unsigned short*, long, int,
                    const float*, const float*, float*, float*, float, float,
-                   int, int, float*, float*, float*, float*, float*,
+                   int, int, float*, float*, float*, float*, float*, float*,
                    hipStream_t);
 void launch_bn_bwd(const unsigned short*, const unsigned short*,
                    unsigned short*, long, int, const float*, const float*,
                    const float*, const float*, const float*, int, int,
-                   float*, float*, float*, float*, hipStream_t);
+                   float*, float*, float*, float*, float*, hipStream_t);
+}
+
+// Mirror of the kernel-side stage-1 grid sizing (bn_kernels.hip).
+static long bn_num_rblocks(long m, long c) {
+    long rpb = 256 / (c / 8);
+    if (rpb < 1) rpb = 1;
+    long nb = (m + rpb - 1) / rpb;
+    if (nb > 1024) nb = 1024;
+    if (nb < 1) nb = 1;
+    return nb;
 }
 
 namespace {
@@ -152,7 +162,7 @@ void check_vecf(const torch::Tensor& t, long n, const char* name) {
 void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
             torch::Tensor beta, torch::Tensor running_mean,
             torch::Tensor running_var, double momentum, double eps,
-            bool train, bool relu, torch::Tensor sums,
+            bool train, bool relu, torch::Tensor ws, torch::Tensor sums,
             torch::Tensor save_mean, torch::Tensor save_rstd,
             torch::Tensor scale, torch::Tensor shift) {
     check_bn_x(x, "x"); check_bn_x(y, "y");
@@ -161,6 +171,8 @@ void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
     TORCH_CHECK(c % 8 == 0 && c / 8 <= 256,
                 "bn_fwd requires C % 8 == 0 and C <= 2048, got C=", c);
     check_vecf(gamma, c, "gamma"); check_vecf(beta, c, "beta");
+    if (train)
+        check_vecf(ws, 2 * c * bn_num_rblocks(m, c), "ws");
     check_vecf(sums, 2 * c, "sums");
     check_vecf(save_mean, c, "save_mean");
     check_vecf(save_rstd, c, "save_rstd");
@@ -179,6 +191,7 @@ void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
                   (unsigned short*)y.data_ptr(), m, (int)c,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(), rm, rv,
                   (float)momentum, (float)eps, (int)train, (int)relu,
+                  train ? ws.data_ptr<float>() : nullptr,
                   sums.data_ptr<float>(), save_mean.data_ptr<float>(),
                   save_rstd.data_ptr<float>(), scale.data_ptr<float>(),
                   shift.data_ptr<float>(), stream());
@@ -187,8 +200,9 @@ void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
 void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
             torch::Tensor gamma, torch::Tensor save_mean,
             torch::Tensor save_rstd, torch::Tensor scale,
-            torch::Tensor shift, bool train, bool relu, torch::Tensor sums,
-            torch::Tensor dgamma, torch::Tensor dbeta, torch::Tensor pqr) {
+            torch::Tensor shift, bool train, bool relu, torch::Tensor ws,
+            torch::Tensor sums, torch::Tensor dgamma, torch::Tensor dbeta,
+            torch::Tensor pqr) {
     check_bn_x(x, "x"); check_bn_x(dy, "dy"); check_bn_x(dx, "dx");
     const long c = x.size(1);
     const long m = x.numel() / c;
@@ -196,6 +210,7 @@ void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
     check_vecf(save_mean, c, "save_mean");
     check_vecf(save_rstd, c, "save_rstd");
     check_vecf(scale, c, "scale"); check_vecf(shift, c, "shift");
+    check_vecf(ws, 2 * c * bn_num_rblocks(m, c), "ws");
     check_vecf(sums, 2 * c, "sums");
     check_vecf(dgamma, c, "dgamma"); check_vecf(dbeta, c, "dbeta");
     check_vecf(pqr, 3 * c, "pqr");
@@ -205,8 +220,9 @@ void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
                   gamma.data_ptr<float>(), save_mean.data_ptr<float>(),
                   save_rstd.data_ptr<float>(), scale.data_ptr<float>(),
                   shift.data_ptr<float>(), (int)train, (int)relu,
-                  sums.data_ptr<float>(), dgamma.data_ptr<float>(),
-                  dbeta.data_ptr<float>(), pqr.data_ptr<float>(), stream());
+                  ws.data_ptr<float>(), sums.data_ptr<float>(),
+                  dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                  pqr.data_ptr<float>(), stream());
 }
 
 }  // namespace

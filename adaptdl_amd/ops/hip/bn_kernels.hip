@@ -4,28 +4,29 @@
 // rocprof on ResNet-18/CIFAR (profiles/bench_r02_channels_last_kernel_
 // stats.csv) shows MIOpen's NHWC BN at 56% of total kernel time, ~10x off
 // the HBM3E bandwidth bound.  These kernels are plain bandwidth-bound
-// passes designed for the memory-bound rules of
-// /opt/skills/guides/cdna_hip_programming.md (Guidelines 11-13, App. B):
-//   - bf16 loads/stores vectorized 8-wide (16 B/lane),
-//   - per-thread fp32 partials -> LDS tree over the block's row group ->
-//     one atomicAdd per (block, channel-octet) (Guideline 12),
-//   - grid-stride over rows, grid capped at 2048 blocks,
-//   - ReLU (and its backward mask, recomputed from x — no extra read)
-//     fused into the BN passes so no separate threshold kernels run.
+// passes designed per /opt/skills/guides/cdna_hip_programming.md
+// (Guidelines 11-13, App. B):
+//   - bf16 loads/stores vectorized 8-wide (16 B/lane); a wave's 64 lanes
+//     cover whole consecutive NHWC rows, so global access is one
+//     contiguous 1 KiB span per wave instruction,
+//   - every thread owns a FIXED channel octet (slot); per-channel
+//     parameters are loaded into registers once, outside the row loop,
+//   - reductions are two-stage: per-block partials to a workspace (NO
+//     cross-block atomics — v1 used one atomicAdd per (block, channel)
+//     and measured 432 us vs the 20 us bandwidth-bound apply pass, pure
+//     atomic serialization), then a parallel per-channel block reduce,
+//   - ReLU and its backward mask (recomputed from x — no extra stream)
+//     are folded in, so no separate threshold/clamp kernels run.
 //
 // Layout contract: x is NHWC ("channels_last") bf16 with C % 8 == 0 and
-// C/8 <= 256; per-channel parameters are fp32.  M = N*H*W rows.
-//
-// fwd:  reduce(x) -> sums[2C];  finalize -> mean/rstd/scale/shift +
-//       running-stat update;  apply: y = relu?(x*scale+shift)
-// bwd:  reduce(x, dy) -> sums[2C] (masked dy, dy*xhat);  finalize ->
-//       dgamma/dbeta + per-channel (p, q, r);  apply: dx = p*g - q*x + r
+// C/8 <= 256; per-channel parameters fp32.  M = N*H*W rows.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
 #define WAVE 64
 #define BLOCK 256
+#define MAX_RBLOCKS 1024   // stage-1 reduction grid cap (partials rows)
 #define MAX_BLOCKS 2048
 #define VEC 8
 
@@ -47,40 +48,10 @@ __device__ __forceinline__ ushort_t f2b(float f) {
     return *reinterpret_cast<ushort_t*>(&h);
 }
 
-// ---------------------------------------------------------------------
-// Forward reduction: sums[c] += sum_x, sums[C + c] += sum_x2.
-// Threads: slot = tid % slots (owns channels slot*8..+8), row group
-// rg = tid / slots; rows grid-stride with stride gridDim.x * rpb.
-// ---------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_reduce(
-        const ushort_t* __restrict__ x, long m, int c,
-        float* __restrict__ sums) {
-    const int slots = c / VEC;
-    const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
-    const int slot = threadIdx.x % slots;
-    const int rg = threadIdx.x / slots;
-    __shared__ float lds[BLOCK * VEC];
-
-    float s[VEC], s2[VEC];
-    #pragma unroll
-    for (int k = 0; k < VEC; ++k) { s[k] = 0.f; s2[k] = 0.f; }
-
-    if (rg < rpb) {
-        const ushort_t* base = x + (size_t)slot * VEC;
-        for (long row = (long)blockIdx.x * rpb + rg; row < m;
-                row += (long)gridDim.x * rpb) {
-            Vec8 v;
-            v.f4 = *reinterpret_cast<const float4*>(base + (size_t)row * c);
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k) {
-                float f = b2f(v.u[k]);
-                s[k] += f;
-                s2[k] += f * f;
-            }
-        }
-    }
-    // LDS tree over the row-group dimension (rg), separately for s and s2.
-    // Index layout: lds[tid * VEC + k], tid = rg * slots + slot.
+// Shared LDS tree over the row-group dimension for VEC per-thread values.
+// tid = rg * slots + slot; reduces rg in [0, rpb) down to rg == 0.
+__device__ __forceinline__ void rg_tree_reduce(
+        float (&s)[VEC], float* lds, int slots, int rpb, int rg) {
     int p2 = 1;
     while (p2 * 2 <= rpb) p2 *= 2;
     #pragma unroll
@@ -104,38 +75,77 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_reduce(
         }
         __syncthreads();
     }
-    if (rg == 0) {
-        #pragma unroll
-        for (int k = 0; k < VEC; ++k)
-            atomicAdd(&sums[slot * VEC + k], s[k]);
-    }
-    // Second pass of the same tree for s2 (reuse LDS).
-    __syncthreads();
+}
+
+// ---------------------------------------------------------------------
+// Stage-1 forward reduction: per-block partial (sum, sumsq) per channel.
+// ws layout: ws[c * nb + block] (c in [0, C) sums, [C, 2C) sum-squares)
+// so the stage-2 per-channel reduce reads contiguously.
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_reduce(
+        const ushort_t* __restrict__ x, long m, int c, int nb,
+        float* __restrict__ ws) {
+    const int slots = c / VEC;
+    const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
+    const int slot = threadIdx.x % slots;
+    const int rg = threadIdx.x / slots;
+    __shared__ float lds[BLOCK * VEC];
+
+    float s[VEC], s2[VEC];
     #pragma unroll
-    for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = s2[k];
-    __syncthreads();
-    if (rg < p2 && rg + p2 < rpb) {
-        #pragma unroll
-        for (int k = 0; k < VEC; ++k)
-            s2[k] += lds[(threadIdx.x + p2 * slots) * VEC + k];
-    }
-    for (int r = p2 / 2; r > 0; r >>= 1) {
-        if (rg < p2) {
+    for (int k = 0; k < VEC; ++k) { s[k] = 0.f; s2[k] = 0.f; }
+
+    if (rg < rpb) {
+        const ushort_t* base = x + (size_t)slot * VEC;
+        for (long row = (long)blockIdx.x * rpb + rg; row < m;
+                row += (long)gridDim.x * rpb) {
+            Vec8 v;
+            v.f4 = *reinterpret_cast<const float4*>(base + (size_t)row * c);
             #pragma unroll
-            for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = s2[k];
+            for (int k = 0; k < VEC; ++k) {
+                float f = b2f(v.u[k]);
+                s[k] += f;
+                s2[k] = fmaf(f, f, s2[k]);
+            }
         }
-        __syncthreads();
-        if (rg < r) {
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k)
-                s2[k] += lds[(threadIdx.x + r * slots) * VEC + k];
-        }
-        __syncthreads();
     }
+    rg_tree_reduce(s, lds, slots, rpb, rg);
     if (rg == 0) {
         #pragma unroll
         for (int k = 0; k < VEC; ++k)
-            atomicAdd(&sums[c + slot * VEC + k], s2[k]);
+            ws[(size_t)(slot * VEC + k) * nb + blockIdx.x] = s[k];
+    }
+    __syncthreads();
+    rg_tree_reduce(s2, lds, slots, rpb, rg);
+    if (rg == 0) {
+        #pragma unroll
+        for (int k = 0; k < VEC; ++k)
+            ws[(size_t)(c + slot * VEC + k) * nb + blockIdx.x] = s2[k];
+    }
+}
+
+// ---------------------------------------------------------------------
+// Stage-2: one block per channel-statistic row; sums ws[row * nb .. +nb].
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_reduce_ws(
+        const float* __restrict__ ws, int nb, float* __restrict__ out) {
+    __shared__ float lds[BLOCK];
+    const float* row = ws + (size_t)blockIdx.x * nb;
+    float s = 0.f;
+    for (int i = threadIdx.x; i < nb; i += BLOCK) s += row[i];
+    // wave shuffle reduce, then LDS across the block's 4 waves
+    #pragma unroll
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        s += __shfl_down(s, off, WAVE);
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x / WAVE;
+    if (lane == 0) lds[wid] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.f;
+        #pragma unroll
+        for (int i = 0; i < BLOCK / WAVE; ++i) t += lds[i];
+        out[blockIdx.x] = t;
     }
 }
 
@@ -188,39 +198,50 @@ extern "C" __global__ void k_bn_eval_finalize(
 }
 
 // ---------------------------------------------------------------------
-// Forward apply: y = [relu](x * scale + shift), 8 channels per lane.
+// Forward apply: y = [relu](x * scale + shift).  Row-loop structure with
+// a fixed channel octet per thread so scale/shift live in registers.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_apply(
         const ushort_t* __restrict__ x, ushort_t* __restrict__ y,
-        long nvec, int slots, const float* __restrict__ scale,
+        long m, int c, const float* __restrict__ scale,
         const float* __restrict__ shift, int relu) {
-    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
-    const long stride = (long)gridDim.x * BLOCK;
-    for (long v = i; v < nvec; v += stride) {
-        const int cb = (int)(v % slots) * VEC;
+    const int slots = c / VEC;
+    const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
+    const int slot = threadIdx.x % slots;
+    const int rg = threadIdx.x / slots;
+    if (rg >= rpb) return;
+    float sc[VEC], sh[VEC];
+    #pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+        sc[k] = scale[slot * VEC + k];
+        sh[k] = shift[slot * VEC + k];
+    }
+    const ushort_t* xb = x + (size_t)slot * VEC;
+    ushort_t* yb = y + (size_t)slot * VEC;
+    for (long row = (long)blockIdx.x * rpb + rg; row < m;
+            row += (long)gridDim.x * rpb) {
         Vec8 in, out;
-        in.f4 = *reinterpret_cast<const float4*>(x + v * VEC);
+        in.f4 = *reinterpret_cast<const float4*>(xb + (size_t)row * c);
         #pragma unroll
         for (int k = 0; k < VEC; ++k) {
-            float f = fmaf(b2f(in.u[k]), scale[cb + k], shift[cb + k]);
+            float f = fmaf(b2f(in.u[k]), sc[k], sh[k]);
             if (relu) f = f > 0.f ? f : 0.f;
             out.u[k] = f2b(f);
         }
-        *reinterpret_cast<float4*>(y + v * VEC) = out.f4;
+        *reinterpret_cast<float4*>(yb + (size_t)row * c) = out.f4;
     }
 }
 
 // ---------------------------------------------------------------------
-// Backward reduction: masked dy and dy * xhat sums per channel.
-// g = dy * (relu ? (x*scale+shift > 0) : 1); xhat = (x - mean) * rstd.
-// sums[c] += sum g; sums[C + c] += sum g * xhat.
+// Stage-1 backward reduction: per-block partials of (masked dy) and
+// (masked dy * xhat) per channel, into the same ws layout as forward.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_reduce(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-        long m, int c, const float* __restrict__ mean,
+        long m, int c, int nb, const float* __restrict__ mean,
         const float* __restrict__ rstd, const float* __restrict__ scale,
         const float* __restrict__ shift, int relu,
-        float* __restrict__ sums) {
+        float* __restrict__ ws) {
     const int slots = c / VEC;
     const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
     const int slot = threadIdx.x % slots;
@@ -251,64 +272,22 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_reduce(
                 float g = b2f(vd.u[k]);
                 if (relu && fmaf(fx, sc[k], sh[k]) <= 0.f) g = 0.f;
                 sg[k] += g;
-                sgx[k] += g * (fx - mn[k]) * rs[k];
+                sgx[k] = fmaf(g, (fx - mn[k]) * rs[k], sgx[k]);
             }
         }
     }
-    int p2 = 1;
-    while (p2 * 2 <= rpb) p2 *= 2;
-    #pragma unroll
-    for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = sg[k];
-    __syncthreads();
-    if (rg < p2 && rg + p2 < rpb) {
-        #pragma unroll
-        for (int k = 0; k < VEC; ++k)
-            sg[k] += lds[(threadIdx.x + p2 * slots) * VEC + k];
-    }
-    for (int r = p2 / 2; r > 0; r >>= 1) {
-        if (rg < p2) {
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = sg[k];
-        }
-        __syncthreads();
-        if (rg < r) {
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k)
-                sg[k] += lds[(threadIdx.x + r * slots) * VEC + k];
-        }
-        __syncthreads();
-    }
+    rg_tree_reduce(sg, lds, slots, rpb, rg);
     if (rg == 0) {
         #pragma unroll
         for (int k = 0; k < VEC; ++k)
-            atomicAdd(&sums[slot * VEC + k], sg[k]);
+            ws[(size_t)(slot * VEC + k) * nb + blockIdx.x] = sg[k];
     }
     __syncthreads();
-    #pragma unroll
-    for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = sgx[k];
-    __syncthreads();
-    if (rg < p2 && rg + p2 < rpb) {
-        #pragma unroll
-        for (int k = 0; k < VEC; ++k)
-            sgx[k] += lds[(threadIdx.x + p2 * slots) * VEC + k];
-    }
-    for (int r = p2 / 2; r > 0; r >>= 1) {
-        if (rg < p2) {
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k) lds[threadIdx.x * VEC + k] = sgx[k];
-        }
-        __syncthreads();
-        if (rg < r) {
-            #pragma unroll
-            for (int k = 0; k < VEC; ++k)
-                sgx[k] += lds[(threadIdx.x + r * slots) * VEC + k];
-        }
-        __syncthreads();
-    }
+    rg_tree_reduce(sgx, lds, slots, rpb, rg);
     if (rg == 0) {
         #pragma unroll
         for (int k = 0; k < VEC; ++k)
-            atomicAdd(&sums[c + slot * VEC + k], sgx[k]);
+            ws[(size_t)(c + slot * VEC + k) * nb + blockIdx.x] = sgx[k];
     }
 }
 
@@ -344,40 +323,52 @@ extern "C" __global__ void k_bn_bwd_finalize(
 }
 
 // ---------------------------------------------------------------------
-// Backward apply: dx = p * g - q * x + r (g = masked dy).
+// Backward apply: dx = p * g - q * x + r (g = masked dy), row-loop with
+// register-resident per-channel coefficients.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_apply(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-        ushort_t* __restrict__ dx, long nvec, int slots, int c,
+        ushort_t* __restrict__ dx, long m, int c,
         const float* __restrict__ scale, const float* __restrict__ shift,
         const float* __restrict__ pqr, int relu) {
-    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
-    const long stride = (long)gridDim.x * BLOCK;
-    for (long v = i; v < nvec; v += stride) {
-        const int cb = (int)(v % slots) * VEC;
+    const int slots = c / VEC;
+    const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
+    const int slot = threadIdx.x % slots;
+    const int rg = threadIdx.x / slots;
+    if (rg >= rpb) return;
+    float sc[VEC], sh[VEC], cp[VEC], cq[VEC], cr[VEC];
+    #pragma unroll
+    for (int k = 0; k < VEC; ++k) {
+        sc[k] = scale[slot * VEC + k];
+        sh[k] = shift[slot * VEC + k];
+        cp[k] = pqr[slot * VEC + k];
+        cq[k] = pqr[c + slot * VEC + k];
+        cr[k] = pqr[2 * c + slot * VEC + k];
+    }
+    const ushort_t* xb = x + (size_t)slot * VEC;
+    const ushort_t* db = dy + (size_t)slot * VEC;
+    ushort_t* ob = dx + (size_t)slot * VEC;
+    for (long row = (long)blockIdx.x * rpb + rg; row < m;
+            row += (long)gridDim.x * rpb) {
         Vec8 vx, vd, out;
-        vx.f4 = *reinterpret_cast<const float4*>(x + v * VEC);
-        vd.f4 = *reinterpret_cast<const float4*>(dy + v * VEC);
+        vx.f4 = *reinterpret_cast<const float4*>(xb + (size_t)row * c);
+        vd.f4 = *reinterpret_cast<const float4*>(db + (size_t)row * c);
         #pragma unroll
         for (int k = 0; k < VEC; ++k) {
             float fx = b2f(vx.u[k]);
             float g = b2f(vd.u[k]);
-            if (relu && fmaf(fx, scale[cb + k], shift[cb + k]) <= 0.f)
-                g = 0.f;
-            float p = pqr[cb + k];
-            float q = pqr[c + cb + k];
-            float r = pqr[2 * c + cb + k];
-            out.u[k] = f2b(fmaf(p, g, fmaf(-q, fx, r)));
+            if (relu && fmaf(fx, sc[k], sh[k]) <= 0.f) g = 0.f;
+            out.u[k] = f2b(fmaf(cp[k], g, fmaf(-cq[k], fx, cr[k])));
         }
-        *reinterpret_cast<float4*>(dx + v * VEC) = out.f4;
+        *reinterpret_cast<float4*>(ob + (size_t)row * c) = out.f4;
     }
 }
 
 // ---- host-side launchers ---------------------------------------------
 
-static inline unsigned bn_grid(long work_items, int per_block) {
+static inline unsigned bn_grid(long work_items, int per_block, long cap) {
     long blocks = (work_items + per_block - 1) / per_block;
-    if (blocks > MAX_BLOCKS) blocks = MAX_BLOCKS;
+    if (blocks > cap) blocks = cap;
     if (blocks < 1) blocks = 1;
     return (unsigned)blocks;
 }
@@ -386,14 +377,16 @@ extern "C" void launch_bn_fwd(
         const ushort_t* x, ushort_t* y, long m, int c,
         const float* gamma, const float* beta, float* running_mean,
         float* running_var, float momentum, float eps, int train, int relu,
-        float* sums, float* save_mean, float* save_rstd, float* scale,
-        float* shift, hipStream_t s) {
+        float* ws, float* sums, float* save_mean, float* save_rstd,
+        float* scale, float* shift, hipStream_t s) {
     const int slots = c / VEC;
+    const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
     if (train) {
-        hipMemsetAsync(sums, 0, 2 * c * sizeof(float), s);
-        const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
-        hipLaunchKernelGGL(k_bn_fwd_reduce, dim3(bn_grid(m, rpb)),
-                           dim3(BLOCK), 0, s, x, m, c, sums);
+        const unsigned nb = bn_grid(m, rpb, MAX_RBLOCKS);
+        hipLaunchKernelGGL(k_bn_fwd_reduce, dim3(nb), dim3(BLOCK), 0, s,
+                           x, m, c, (int)nb, ws);
+        hipLaunchKernelGGL(k_bn_reduce_ws, dim3(2 * c), dim3(BLOCK), 0, s,
+                           ws, (int)nb, sums);
         hipLaunchKernelGGL(k_bn_fwd_finalize,
                            dim3((c + BLOCK - 1) / BLOCK), dim3(BLOCK), 0, s,
                            sums, m, c, gamma, beta, running_mean,
@@ -405,29 +398,28 @@ extern "C" void launch_bn_fwd(
                            c, gamma, beta, running_mean, running_var, eps,
                            save_mean, save_rstd, scale, shift);
     }
-    const long nvec = m * slots;
-    hipLaunchKernelGGL(k_bn_fwd_apply, dim3(bn_grid(nvec, BLOCK)),
-                       dim3(BLOCK), 0, s, x, y, nvec, slots, scale, shift,
-                       relu);
+    hipLaunchKernelGGL(k_bn_fwd_apply, dim3(bn_grid(m, rpb, MAX_BLOCKS)),
+                       dim3(BLOCK), 0, s, x, y, m, c, scale, shift, relu);
 }
 
 extern "C" void launch_bn_bwd(
         const ushort_t* x, const ushort_t* dy, ushort_t* dx, long m, int c,
         const float* gamma, const float* save_mean, const float* save_rstd,
         const float* scale, const float* shift, int train, int relu,
-        float* sums, float* dgamma, float* dbeta, float* pqr,
+        float* ws, float* sums, float* dgamma, float* dbeta, float* pqr,
         hipStream_t s) {
     const int slots = c / VEC;
-    hipMemsetAsync(sums, 0, 2 * c * sizeof(float), s);
     const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
-    hipLaunchKernelGGL(k_bn_bwd_reduce, dim3(bn_grid(m, rpb)), dim3(BLOCK),
-                       0, s, x, dy, m, c, save_mean, save_rstd, scale,
-                       shift, relu, sums);
+    const unsigned nb = bn_grid(m, rpb, MAX_RBLOCKS);
+    hipLaunchKernelGGL(k_bn_bwd_reduce, dim3(nb), dim3(BLOCK), 0, s,
+                       x, dy, m, c, (int)nb, save_mean, save_rstd, scale,
+                       shift, relu, ws);
+    hipLaunchKernelGGL(k_bn_reduce_ws, dim3(2 * c), dim3(BLOCK), 0, s,
+                       ws, (int)nb, sums);
     hipLaunchKernelGGL(k_bn_bwd_finalize, dim3((c + BLOCK - 1) / BLOCK),
                        dim3(BLOCK), 0, s, sums, m, c, gamma, save_mean,
                        save_rstd, train, dgamma, dbeta, pqr);
-    const long nvec = m * slots;
-    hipLaunchKernelGGL(k_bn_bwd_apply, dim3(bn_grid(nvec, BLOCK)),
-                       dim3(BLOCK), 0, s, x, dy, dx, nvec, slots, c, scale,
-                       shift, pqr, relu);
+    hipLaunchKernelGGL(k_bn_bwd_apply, dim3(bn_grid(m, rpb, MAX_BLOCKS)),
+                       dim3(BLOCK), 0, s, x, dy, dx, m, c, scale, shift,
+                       pqr, relu);
 }

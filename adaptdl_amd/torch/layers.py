@@ -21,6 +21,12 @@ import torch.nn.functional as F
 from adaptdl_amd import ops
 
 
+def _num_rblocks(m, c):
+    """Mirror of bn_kernels.hip stage-1 grid sizing (partials rows)."""
+    rpb = max(256 // (c // 8), 1)
+    return max(min((m + rpb - 1) // rpb, 1024), 1)
+
+
 def _hip_bn_ok(x):
     c = x.shape[1] if x.dim() == 4 else 0
     return (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4 and
@@ -36,7 +42,10 @@ class _FusedBNFunction(torch.autograd.Function):
         ext = ops._load_extension()
         y = torch.empty_like(x)
         c = x.shape[1]
+        m = x.numel() // c
         opt = dict(dtype=torch.float32, device=x.device)
+        ws = torch.empty(2 * c * _num_rblocks(m, c), **opt) \
+            if training else torch.empty(0, **opt)
         sums = torch.empty(2 * c, **opt)
         save_mean = torch.empty(c, **opt)
         save_rstd = torch.empty(c, **opt)
@@ -47,7 +56,7 @@ class _FusedBNFunction(torch.autograd.Function):
                    running_mean if running_mean is not None else none,
                    running_var if running_var is not None else none,
                    float(momentum), float(eps), bool(training), bool(relu),
-                   sums, save_mean, save_rstd, scale, shift)
+                   ws, sums, save_mean, save_rstd, scale, shift)
         ctx.save_for_backward(x, weight, save_mean, save_rstd, scale, shift)
         ctx.bn_train = bool(training)
         ctx.bn_relu = bool(relu)
@@ -60,13 +69,15 @@ class _FusedBNFunction(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = torch.empty_like(x)
         c = x.shape[1]
+        m = x.numel() // c
         opt = dict(dtype=torch.float32, device=x.device)
+        ws = torch.empty(2 * c * _num_rblocks(m, c), **opt)
         sums = torch.empty(2 * c, **opt)
         dgamma = torch.empty(c, **opt)
         dbeta = torch.empty(c, **opt)
         pqr = torch.empty(3 * c, **opt)
         ext.bn_bwd(x, dy, dx, weight, save_mean, save_rstd, scale, shift,
-                   ctx.bn_train, ctx.bn_relu, sums, dgamma, dbeta, pqr)
+                   ctx.bn_train, ctx.bn_relu, ws, sums, dgamma, dbeta, pqr)
         return (dx, dgamma, dbeta) + (None,) * 6
 
 
