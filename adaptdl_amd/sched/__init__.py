@@ -6,3 +6,8 @@ in-process allocator thread (adaptdl_amd.sched.allocator) next to a local
 job controller (adaptdl_amd.sched.controller) that implements
 checkpoint-restart elasticity with worker processes, one per GPU.
 """
+
+from adaptdl_amd.sched.allocator import LocalAllocator  # noqa: F401,E402
+from adaptdl_amd.sched.controller import (  # noqa: F401,E402
+    JobSpec, LocalController)
+from adaptdl_amd.sched.supervisor import Supervisor  # noqa: F401,E402

@@ -31,7 +31,9 @@ _RESTART_PENALTY = 0.1
 
 
 class PolluxPolicy(object):
-    def __init__(self, seed=None):
+    def __init__(self, seed=None, pop_size=_POP_SIZE, generations=_N_GEN):
+        self._pop_size = pop_size
+        self._generations = generations
         self._prev_population = None
         self._prev_jobs = None
         self._prev_nodes = None
@@ -139,7 +141,8 @@ class PolluxPolicy(object):
             list(jobs.values()),
             list(nodes.values()) + [node_template] * len(nodes),
             base, self._rng)
-        population, values = _nsga2(problem, seeds, _POP_SIZE, _N_GEN)
+        population, values = _nsga2(problem, seeds, self._pop_size,
+                                    self._generations)
 
         self._prev_population = population.copy()
         self._prev_jobs = list(jobs)
@@ -203,7 +206,10 @@ class _AllocationProblem(object):
         self.node_res = np.array(
             [[node.resources.get(rt, 0) for rt in rtypes] for node in nodes],
             dtype=np.int64)
-        shares = self.job_res / np.sum(self.node_res, axis=0)
+        total = np.sum(self.node_res, axis=0)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            shares = np.where(total > 0,
+                              self.job_res / np.maximum(total, 1), 0.0)
         self.dominant_share = np.amax(shares, axis=1)
 
         # Per-(job, node) replica caps, after subtracting pinned usage.

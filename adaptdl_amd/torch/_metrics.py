@@ -10,6 +10,7 @@ sync engine rather than by hooks around torch DDP.
 """
 
 import collections
+import os
 import pickle
 import time
 
@@ -20,7 +21,7 @@ import adaptdl_amd.env
 from adaptdl_amd.goodput import GoodputFunction, fit_perf_params
 from adaptdl_amd.sched_hints import SCHED_HINTS, PERF_PARAMS, post_sched_hints
 
-_FIT_INTERVAL = 30.0
+_FIT_INTERVAL = float(os.getenv("ADAPTDL_FIT_INTERVAL", "30"))
 
 
 def profile_step_start(atomic_bsz):

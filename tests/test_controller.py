@@ -1,0 +1,179 @@
+"""LocalController elasticity: submit -> run -> rescale -> complete.
+
+End-to-end local-cluster counterpart of the reference's workload shell
+tests (/root/reference/tests/testworkload.sh and controller state machine
+controller.py:101-318): a real training job (full adaptdl_amd stack,
+gloo) is started as 1 worker process, rescaled to 2 mid-run via SIGTERM/
+checkpoint/exit(143)/restart, and must finish with the learned weights,
+having trained under both replica counts.
+"""
+
+import json
+import os
+import sys
+import textwrap
+import time
+
+import pytest
+
+from adaptdl_amd.sched import JobSpec, LocalController, Supervisor
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = textwrap.dedent("""
+    import json, os, sys
+    sys.path.insert(0, "@@REPO@@")
+    import torch
+    torch.set_num_threads(1)
+    import adaptdl_amd.env as env
+    import adaptdl_amd.torch as adl
+
+    adl.init_process_group("gloo")
+    torch.manual_seed(7)
+    true_w = torch.tensor([[3.0], [4.0]])
+    xs = torch.randn(160, 2)
+    ys = xs @ true_w + 0.01 * torch.randn(160, 1)
+    model = torch.nn.Linear(2, 1, bias=False)
+    with torch.no_grad():
+        model.weight.zero_()
+    optim = torch.optim.SGD(model.parameters(), lr=0.05)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    loader = adl.AdaptiveDataLoader(
+        torch.utils.data.TensorDataset(xs, ys), batch_size=16)
+
+    trace = os.path.join(env.checkpoint_path(), "trace.jsonl")
+    for epoch in adl.remaining_epochs_until(30):
+        for x, y in loader:
+            optim.zero_grad()
+            ((adp(x) - y) ** 2).mean().backward()
+            optim.step()
+        if env.replica_rank() == 0:
+            with open(trace, "a") as f:
+                rec = dict(epoch=epoch, replicas=env.num_replicas(),
+                           restarts=env.num_restarts())
+                f.write(json.dumps(rec) + "\\n")
+        # Slow the job down a little so the rescale lands mid-run.
+        import time as _t
+        _t.sleep(0.05)
+    if env.replica_rank() == 0:
+        w = model.weight.detach().reshape(-1).tolist()
+        with open(os.path.join(env.checkpoint_path(), "final.json"),
+                  "w") as f:
+            json.dump(w, f)
+""")
+
+
+@pytest.fixture
+def controller():
+    ctrl = LocalController(num_gpus=0, interval=3600)  # manual allocation
+    yield ctrl
+    ctrl.shutdown()
+
+
+def test_job_runs_rescales_completes(tmp_path, controller):
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="lr-job",
+                   job_dir=job_dir, min_replicas=1, max_replicas=2,
+                   gpus_per_replica=0)
+    controller.submit(spec)
+
+    # wait until running, then rescale to 2 replicas mid-run
+    deadline = time.time() + 60
+    while controller.status("lr-job")["state"] != "Running":
+        assert time.time() < deadline, controller.status("lr-job")
+        time.sleep(0.1)
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline
+        time.sleep(0.1)
+    controller.rescale("lr-job", 2)
+
+    state = controller.wait("lr-job", timeout=180)
+    assert state == "Succeeded", controller.status("lr-job")
+
+    trace = [json.loads(line) for line in open(trace_path)]
+    epochs = [t["epoch"] for t in trace]
+    assert sorted(set(epochs)) == list(range(30))  # every epoch ran once
+    assert {t["replicas"] for t in trace} == {1, 2}  # both configs trained
+    assert max(t["restarts"] for t in trace) >= 1
+    assert controller.status("lr-job")["restarts"] >= 1
+
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
+
+
+def test_crash_restart_limit(tmp_path, controller):
+    script = tmp_path / "crash.py"
+    script.write_text("import sys; sys.exit(7)\n")
+    job_dir = str(tmp_path / "job2")
+    spec = JobSpec([sys.executable, str(script)], name="crash-job",
+                   job_dir=job_dir, min_replicas=1, max_replicas=1,
+                   gpus_per_replica=0, restart_limit=1)
+    controller.submit(spec)
+    assert controller.wait("crash-job", timeout=60) == "Failed"
+    assert controller.status("crash-job")["restarts"] >= 1
+
+
+def test_supervisor_http_roundtrip():
+    sup = Supervisor().start()
+    try:
+        import urllib.request
+        req = urllib.request.Request(
+            sup.url + "/hints/myjob", method="PUT",
+            data=json.dumps({"initBatchSize": 64}).encode())
+        with urllib.request.urlopen(req) as resp:
+            assert resp.status == 200
+        assert sup.get_hints("myjob") == {"initBatchSize": 64}
+        sup.set_endpoints("myjob", 0, ["127.0.0.1", "127.0.0.1"])
+        with urllib.request.urlopen(sup.url + "/discover/myjob/0") as resp:
+            assert json.loads(resp.read()) == ["127.0.0.1", "127.0.0.1"]
+        with urllib.request.urlopen(sup.url + "/healthz") as resp:
+            assert resp.status == 200
+    finally:
+        sup.stop()
+
+
+def test_allocator_scales_up_from_hints(tmp_path):
+    """The reference's 60 s _optimize_all loop equivalent: worker reports
+    goodput hints -> allocator's Pollux cycle raises the replica count."""
+    from adaptdl_amd.sched import LocalAllocator
+    from adaptdl_amd.sched.policy import PolluxPolicy
+
+    ctrl = LocalController(
+        num_gpus=0, interval=4.0,
+        allocator=LocalAllocator(
+            num_gpus=0, policy=PolluxPolicy(seed=0, pop_size=30,
+                                            generations=20)))
+    try:
+        script = tmp_path / "worker.py"
+        script.write_text(WORKER.replace("@@REPO@@", REPO)
+                          .replace("remaining_epochs_until(30)",
+                                   "remaining_epochs_until(400)"))
+        seen_hints = []
+        ctrl.supervisor.register_hints_callback(
+            lambda job, hints: seen_hints.append((job, hints)))
+        job_dir = str(tmp_path / "job3")
+        os.makedirs(job_dir)
+        spec = JobSpec([sys.executable, str(script)], name="auto-job",
+                       job_dir=job_dir, min_replicas=1, max_replicas=4,
+                       gpus_per_replica=0,
+                       env={"ADAPTDL_FIT_INTERVAL": "1"})
+        ctrl.submit(spec)
+        deadline = time.time() + 150
+        saw_multi = False
+        while time.time() < deadline:
+            st = ctrl.status("auto-job")
+            if st["replicas"] > 1:
+                saw_multi = True
+                break
+            if st["state"] in ("Succeeded", "Failed"):
+                break
+            time.sleep(0.5)
+        assert seen_hints, "worker never reported hints"
+        assert saw_multi, "allocator never scaled the job past 1 replica"
+        assert ctrl.wait("auto-job", timeout=120) == "Succeeded"
+    finally:
+        ctrl.shutdown()

@@ -127,7 +127,7 @@ def test_unusable_node_requests_scaleup():
     jobs = {i: JobInfo({"gpu": 1, "cpu": 1000, "pods": 1}, speedup_fn,
                        now + 60 * i, 0, max_replicas=1)
             for i in range(3)}
-    policy = PolluxPolicy(seed=0)
+    policy = PolluxPolicy(seed=0, pop_size=40, generations=40)
     allocations, desired_nodes = policy.optimize(jobs, nodes, {}, template)
     assert desired_nodes > 3  # node 0 is cpu-starved -> ask for more nodes
     assert max(len(a) for a in allocations.values()) == 1
@@ -149,7 +149,7 @@ def test_non_preemptible_jobs_pinned():
     for i in range(4, 8):
         jobs[i] = JobInfo(job_resources, speedup_fn, now + 60 * i,
                           min_replicas=2, max_replicas=4, preemptible=False)
-    policy = PolluxPolicy(seed=0)
+    policy = PolluxPolicy(seed=0, pop_size=40, generations=40)
     prev_allocs = {}
     for cycle in range(3):
         allocations, _ = policy.optimize(jobs, nodes, prev_allocs, template)
