@@ -20,10 +20,11 @@ from adaptdl_amd.torch.epoch import (current_epoch, finished_epochs,  # noqa
                                      remaining_epochs_until)
 from adaptdl_amd.torch.data import (AdaptiveDataLoader,  # noqa
                                     AdaptiveDataLoaderHelper,
-                                    AdaptiveDataLoaderMixin,
+                                    AdaptiveDataLoaderMixin,  # noqa: F401
                                     ElasticSampler, current_dataloader)
 from adaptdl_amd.torch.parallel import AdaptiveDataParallel  # noqa
 from adaptdl_amd.torch.accumulator import Accumulator  # noqa
+from adaptdl_amd.torch.iterator import AdaptiveBPTTIterator  # noqa
 
 LOG = logging.getLogger(__name__)
 
@@ -39,6 +40,7 @@ __all__ = [
     "ElasticSampler",
     "AdaptiveDataParallel",
     "Accumulator",
+    "AdaptiveBPTTIterator",
 ]
 
 

@@ -1,0 +1,119 @@
+"""Adaptive BPTT iteration for language-model corpora.
+
+Torchtext-free counterpart of the reference's AdaptiveBPTTIterator
+(/root/reference/adaptdl/adaptdl/torch/iterator.py:18-121, which wraps
+the legacy torchtext BPTTIterator): takes a flat token tensor, reshapes
+it by the goodput-chosen batch size each pass, strides the sequence
+dimension by ``bptt_len * num_replicas`` per rank, proportionally
+recomputes the resume index after a rescale (iterator.py:51-54), and
+caps the step count at the minimum across ranks so collectives stay
+symmetric (iterator.py:94-105).
+
+Yields (text, target) pairs of shape (seq_len, batch_size) — or
+(batch_size, seq_len) with batch_first=True — like the classic torch
+word-LM pipeline.
+"""
+
+import math
+
+import torch
+
+import adaptdl_amd.env
+from adaptdl_amd.torch.data import AdaptiveDataLoaderMixin
+
+__all__ = ["AdaptiveBPTTIterator"]
+
+
+class AdaptiveBPTTIterator(AdaptiveDataLoaderMixin):
+    """BPTT iteration with adaptive batch size and elastic restarts.
+
+    Arguments:
+        data (Tensor): 1-D token-id tensor (the whole corpus).
+        batch_size (int): initial number of parallel sequences.
+        bptt_len (int): tokens per truncated-backprop segment.
+        batch_first (bool): yield (batch, seq) instead of (seq, batch).
+        pad_token (int): id used to pad the corpus to a multiple of the
+            batch size.
+        max_batch_size / local_bsz_bounds: enable autoscaling (same
+            contract as AdaptiveDataLoader.autoscale_batch_size).
+    """
+
+    def __init__(self, data, batch_size, bptt_len, batch_first=False,
+                 pad_token=0, max_batch_size=None, local_bsz_bounds=None,
+                 device=None):
+        AdaptiveDataLoaderMixin.__init__(self, batch_size)
+        if data.dim() != 1:
+            raise ValueError("data must be a flat 1-D token tensor")
+        self.data = data
+        self.bptt_len = bptt_len
+        self.batch_first = batch_first
+        self.pad_token = pad_token
+        self.device = device
+        if max_batch_size:
+            self._elastic.autoscale_batch_size(max_batch_size,
+                                               local_bsz_bounds)
+
+    @staticmethod
+    def _recompute_start(prev_curr, prev_end, curr_end):
+        """Proportionally map the resume index into the new reshaped
+        corpus length (reference iterator.py:51-54)."""
+        if prev_end == 0:
+            return prev_curr
+        return math.ceil(prev_curr * curr_end / prev_end)
+
+    def __len__(self):
+        bsz = self._elastic.current_local_bsz or self.batch_size
+        rows = math.ceil(len(self.data) / bsz)
+        return math.ceil(
+            rows / (self.bptt_len * adaptdl_amd.env.num_replicas()))
+
+    def __iter__(self):
+        num_replicas = adaptdl_amd.env.num_replicas()
+        rank = adaptdl_amd.env.replica_rank()
+        with self._elastic.context():
+            if self._elastic.skipdone():
+                return
+            atomic_bsz = self._elastic._sync_local_bsz()
+            # Reshape the corpus to the ATOMIC (per-replica) width; data
+            # parallelism comes from ranks striding disjoint bptt-sized
+            # time segments of the same reshaped matrix (reference
+            # iterator.py:60-97).
+            bsz = atomic_bsz
+            n = len(self.data)
+            pad = int(math.ceil(n / bsz) * bsz - n)
+            data = torch.cat([self.data,
+                              self.data.new_full((pad,), self.pad_token)])
+            # (bsz, rows) -> (rows, bsz): column b is a contiguous slice
+            # of the corpus, the classic word-LM "batchify".
+            data = data.view(bsz, -1).t().contiguous()
+            if self.device is not None:
+                data = data.to(self.device)
+            end = data.size(0)
+
+            self._elastic.current_index = self._recompute_start(
+                self._elastic.current_index, self._elastic.end_index, end)
+            self._elastic.end_index = end
+
+            start = self._elastic.current_index + self.bptt_len * rank
+            step = self.bptt_len * num_replicas
+            highest_start = self._elastic.current_index + \
+                self.bptt_len * (num_replicas - 1)
+            # Cap iterations at the count of the most-starved rank so
+            # every replica enters profile()/collectives equally often.
+            min_steps = max(math.ceil((end - 1 - highest_start) / step), 0)
+
+            iterations = 0
+            for i in range(start, end, step):
+                iterations += 1
+                if iterations > min_steps:
+                    break
+                with self._elastic.profile(self.training and i > 0):
+                    seq_len = min(self.bptt_len, end - i - 1)
+                    assert seq_len > 0
+                    text = data[i:i + seq_len]
+                    target = data[i + 1:i + 1 + seq_len]
+                    if self.batch_first:
+                        text = text.t().contiguous()
+                        target = target.t().contiguous()
+                    yield text, target
+                    self._elastic.current_index += step

