@@ -1,0 +1,101 @@
+"""CLI tests: daemon admin API round-trip and foreground run.
+
+Counterpart coverage for the reference's adaptdl CLI surface
+(/root/reference/cli/bin/adaptdl: submit/ls/logs) in local-node form.
+"""
+
+import json
+import os
+import subprocess
+import sys
+import time
+import urllib.request
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _req(url, method="GET", body=None):
+    data = json.dumps(body).encode() if body is not None else None
+    req = urllib.request.Request(url, data=data, method=method)
+    with urllib.request.urlopen(req, timeout=20) as resp:
+        payload = resp.read()
+    return json.loads(payload) if payload else None
+
+
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def test_daemon_submit_ls_logs(tmp_path):
+    port = _free_port()
+    env = dict(os.environ, PYTHONPATH=REPO)
+    daemon = subprocess.Popen(
+        [sys.executable, "-m", "adaptdl_amd.cli", "daemon",
+         "--bind", "127.0.0.1:{}".format(port),
+         "--state-dir", str(tmp_path / "state"),
+         "--num-gpus", "0", "--interval", "3600"],
+        env=env, cwd=str(tmp_path),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    url = "http://127.0.0.1:{}".format(port)
+    try:
+        deadline = time.time() + 30
+        while True:
+            try:
+                _req(url + "/jobs")
+                break
+            except Exception:
+                assert time.time() < deadline
+                time.sleep(0.2)
+
+        script = tmp_path / "hello.py"
+        script.write_text("print('hello from worker')\n")
+        out = _req(url + "/jobs", "POST", {
+            "argv": [sys.executable, str(script)], "name": "hello",
+            "min_replicas": 1, "max_replicas": 1, "gpus_per_replica": 0})
+        assert out["name"] == "hello"
+
+        deadline = time.time() + 60
+        while True:
+            st = _req(url + "/jobs/hello")
+            if st["state"] in ("Succeeded", "Failed"):
+                break
+            assert time.time() < deadline, st
+            time.sleep(0.3)
+        assert st["state"] == "Succeeded"
+
+        jobs = _req(url + "/jobs")
+        assert "hello" in jobs
+
+        logs = _req(url + "/jobs/hello/logs")
+        assert any("hello from worker" in v for v in logs.values())
+    finally:
+        try:
+            _req(url + "/shutdown", "POST", {})
+        except Exception:
+            pass
+        try:
+            daemon.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            daemon.kill()
+
+
+def test_cli_run_foreground(tmp_path):
+    script = tmp_path / "ok.py"
+    script.write_text("print('ran fine')\n")
+    env = dict(os.environ, PYTHONPATH=REPO)
+    p = subprocess.run(
+        [sys.executable, "-m", "adaptdl_amd.cli", "run",
+         "--name", "fg", "--job-dir", str(tmp_path / "fg"),
+         "--num-gpus", "0", "--gpus-per-replica", "0",
+         "--min-replicas", "1", "--max-replicas", "1", "--",
+         sys.executable, str(script)],
+        env=env, cwd=str(tmp_path), capture_output=True, text=True,
+        timeout=90)
+    assert p.returncode == 0, p.stdout + p.stderr
+    assert "Succeeded" in p.stdout
+    assert "ran fine" in p.stdout
