@@ -27,7 +27,8 @@ class BasicBlock(nn.Module):
         self.bn1 = _bn(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1,
                                bias=False)
-        self.bn2 = _bn(planes)
+        # bn2 fuses the residual add + final ReLU of the block.
+        self.bn2 = _bn(planes, relu=True)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
@@ -37,9 +38,7 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        out = out + self.shortcut(x)
-        return F.relu(out)
+        return self.bn2(self.conv2(out), residual=self.shortcut(x))
 
 
 class Bottleneck(nn.Module):
@@ -54,7 +53,8 @@ class Bottleneck(nn.Module):
         self.bn2 = _bn(planes, relu=True)
         self.conv3 = nn.Conv2d(planes, self.expansion * planes, 1,
                                bias=False)
-        self.bn3 = _bn(self.expansion * planes)
+        # bn3 fuses the residual add + final ReLU of the block.
+        self.bn3 = _bn(self.expansion * planes, relu=True)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
@@ -65,9 +65,7 @@ class Bottleneck(nn.Module):
     def forward(self, x):
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        out = out + self.shortcut(x)
-        return F.relu(out)
+        return self.bn3(self.conv3(out), residual=self.shortcut(x))
 
 
 class CifarResNet(nn.Module):

@@ -24,14 +24,15 @@ void launch_fused_sgd(float*, const float*, float*, long, float, float,
 void launch_fused_adamw(float*, const float*, float*, float*, long, float,
                         float, float, float, float, float, float, int,
                         hipStream_t);
-void launch_bn_fwd(const unsigned short*, unsigned short*, long, int,
-                   const float*, const float*, float*, float*, float, float,
-                   int, int, float*, float*, float*, float*, float*, float*,
-                   hipStream_t);
-void launch_bn_bwd(const unsigned short*, const unsigned short*,
+void launch_bn_fwd(const unsigned short*, const unsigned short*,
                    unsigned short*, long, int, const float*, const float*,
-                   const float*, const float*, const float*, int, int,
-                   float*, float*, float*, float*, float*, hipStream_t);
+                   float*, float*, float, float, int, int, float*, float*,
+                   float*, float*, float*, float*, hipStream_t);
+void launch_bn_bwd(const unsigned short*, const unsigned short*,
+                   const unsigned short*, unsigned short*, unsigned short*,
+                   long, int, const float*, const float*, const float*,
+                   const float*, const float*, int, int, float*, float*,
+                   float*, float*, float*, hipStream_t);
 }
 
 // Mirror of the kernel-side stage-1 grid sizing (bn_kernels.hip).
@@ -159,13 +160,20 @@ void check_vecf(const torch::Tensor& t, long n, const char* name) {
 }
 
 // y = [relu](bn(x)); fills sums(2C ws), save_mean/save_rstd/scale/shift(C).
-void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
-            torch::Tensor beta, torch::Tensor running_mean,
-            torch::Tensor running_var, double momentum, double eps,
-            bool train, bool relu, torch::Tensor ws, torch::Tensor sums,
+void bn_fwd(torch::Tensor x, torch::Tensor z, torch::Tensor y,
+            torch::Tensor gamma, torch::Tensor beta,
+            torch::Tensor running_mean, torch::Tensor running_var,
+            double momentum, double eps, bool train, bool relu,
+            torch::Tensor ws, torch::Tensor sums,
             torch::Tensor save_mean, torch::Tensor save_rstd,
             torch::Tensor scale, torch::Tensor shift) {
     check_bn_x(x, "x"); check_bn_x(y, "y");
+    const unsigned short* zp = nullptr;
+    if (z.defined() && z.numel() > 0) {
+        check_bn_x(z, "z");
+        TORCH_CHECK(z.sizes() == x.sizes(), "residual shape mismatch");
+        zp = (const unsigned short*)z.data_ptr();
+    }
     const long c = x.size(1);
     const long m = x.numel() / c;
     TORCH_CHECK(c % 8 == 0 && c / 8 <= 256,
@@ -187,7 +195,7 @@ void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
     } else {
         TORCH_CHECK(train, "eval-mode bn_fwd requires running statistics");
     }
-    launch_bn_fwd((const unsigned short*)x.data_ptr(),
+    launch_bn_fwd((const unsigned short*)x.data_ptr(), zp,
                   (unsigned short*)y.data_ptr(), m, (int)c,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(), rm, rv,
                   (float)momentum, (float)eps, (int)train, (int)relu,
@@ -197,13 +205,22 @@ void bn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma,
                   shift.data_ptr<float>(), stream());
 }
 
-void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
+void bn_bwd(torch::Tensor x, torch::Tensor z, torch::Tensor dy,
+            torch::Tensor dx, torch::Tensor dz,
             torch::Tensor gamma, torch::Tensor save_mean,
             torch::Tensor save_rstd, torch::Tensor scale,
             torch::Tensor shift, bool train, bool relu, torch::Tensor ws,
             torch::Tensor sums, torch::Tensor dgamma, torch::Tensor dbeta,
             torch::Tensor pqr) {
     check_bn_x(x, "x"); check_bn_x(dy, "dy"); check_bn_x(dx, "dx");
+    const unsigned short* zp = nullptr;
+    unsigned short* dzp = nullptr;
+    if (z.defined() && z.numel() > 0) {
+        check_bn_x(z, "z"); check_bn_x(dz, "dz");
+        TORCH_CHECK(z.sizes() == x.sizes(), "residual shape mismatch");
+        zp = (const unsigned short*)z.data_ptr();
+        dzp = (unsigned short*)dz.data_ptr();
+    }
     const long c = x.size(1);
     const long m = x.numel() / c;
     check_vecf(gamma, c, "gamma");
@@ -214,9 +231,9 @@ void bn_bwd(torch::Tensor x, torch::Tensor dy, torch::Tensor dx,
     check_vecf(sums, 2 * c, "sums");
     check_vecf(dgamma, c, "dgamma"); check_vecf(dbeta, c, "dbeta");
     check_vecf(pqr, 3 * c, "pqr");
-    launch_bn_bwd((const unsigned short*)x.data_ptr(),
+    launch_bn_bwd((const unsigned short*)x.data_ptr(), zp,
                   (const unsigned short*)dy.data_ptr(),
-                  (unsigned short*)dx.data_ptr(), m, (int)c,
+                  (unsigned short*)dx.data_ptr(), dzp, m, (int)c,
                   gamma.data_ptr<float>(), save_mean.data_ptr<float>(),
                   save_rstd.data_ptr<float>(), scale.data_ptr<float>(),
                   shift.data_ptr<float>(), (int)train, (int)relu,

@@ -202,8 +202,9 @@ extern "C" __global__ void k_bn_eval_finalize(
 // a fixed channel octet per thread so scale/shift live in registers.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_apply(
-        const ushort_t* __restrict__ x, ushort_t* __restrict__ y,
-        long m, int c, const float* __restrict__ scale,
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ z,
+        ushort_t* __restrict__ y, long m, int c,
+        const float* __restrict__ scale,
         const float* __restrict__ shift, int relu) {
     const int slots = c / VEC;
     const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
@@ -217,14 +218,18 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_apply(
         sh[k] = shift[slot * VEC + k];
     }
     const ushort_t* xb = x + (size_t)slot * VEC;
+    const ushort_t* zb = z ? z + (size_t)slot * VEC : nullptr;
     ushort_t* yb = y + (size_t)slot * VEC;
     for (long row = (long)blockIdx.x * rpb + rg; row < m;
             row += (long)gridDim.x * rpb) {
-        Vec8 in, out;
+        Vec8 in, vz, out;
         in.f4 = *reinterpret_cast<const float4*>(xb + (size_t)row * c);
+        if (zb)
+            vz.f4 = *reinterpret_cast<const float4*>(zb + (size_t)row * c);
         #pragma unroll
         for (int k = 0; k < VEC; ++k) {
             float f = fmaf(b2f(in.u[k]), sc[k], sh[k]);
+            if (zb) f += b2f(vz.u[k]);
             if (relu) f = f > 0.f ? f : 0.f;
             out.u[k] = f2b(f);
         }
@@ -237,7 +242,8 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_fwd_apply(
 // (masked dy * xhat) per channel, into the same ws layout as forward.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_reduce(
-        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ z,
+        const ushort_t* __restrict__ dy,
         long m, int c, int nb, const float* __restrict__ mean,
         const float* __restrict__ rstd, const float* __restrict__ scale,
         const float* __restrict__ shift, int relu,
@@ -260,17 +266,25 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_reduce(
     }
     if (rg < rpb) {
         const ushort_t* xb = x + (size_t)slot * VEC;
+        const ushort_t* zb = z ? z + (size_t)slot * VEC : nullptr;
         const ushort_t* db = dy + (size_t)slot * VEC;
         for (long row = (long)blockIdx.x * rpb + rg; row < m;
                 row += (long)gridDim.x * rpb) {
-            Vec8 vx, vd;
+            Vec8 vx, vz, vd;
             vx.f4 = *reinterpret_cast<const float4*>(xb + (size_t)row * c);
+            if (zb)
+                vz.f4 = *reinterpret_cast<const float4*>(
+                    zb + (size_t)row * c);
             vd.f4 = *reinterpret_cast<const float4*>(db + (size_t)row * c);
             #pragma unroll
             for (int k = 0; k < VEC; ++k) {
                 float fx = b2f(vx.u[k]);
                 float g = b2f(vd.u[k]);
-                if (relu && fmaf(fx, sc[k], sh[k]) <= 0.f) g = 0.f;
+                if (relu) {
+                    float pre = fmaf(fx, sc[k], sh[k]);
+                    if (zb) pre += b2f(vz.u[k]);
+                    if (pre <= 0.f) g = 0.f;
+                }
                 sg[k] += g;
                 sgx[k] = fmaf(g, (fx - mn[k]) * rs[k], sgx[k]);
             }
@@ -327,9 +341,11 @@ extern "C" __global__ void k_bn_bwd_finalize(
 // register-resident per-channel coefficients.
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_apply(
-        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-        ushort_t* __restrict__ dx, long m, int c,
-        const float* __restrict__ scale, const float* __restrict__ shift,
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ z,
+        const ushort_t* __restrict__ dy,
+        ushort_t* __restrict__ dx, ushort_t* __restrict__ dz, long m,
+        int c, const float* __restrict__ scale,
+        const float* __restrict__ shift,
         const float* __restrict__ pqr, int relu) {
     const int slots = c / VEC;
     const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
@@ -346,21 +362,32 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_bn_bwd_apply(
         cr[k] = pqr[2 * c + slot * VEC + k];
     }
     const ushort_t* xb = x + (size_t)slot * VEC;
+    const ushort_t* zb = z ? z + (size_t)slot * VEC : nullptr;
     const ushort_t* db = dy + (size_t)slot * VEC;
     ushort_t* ob = dx + (size_t)slot * VEC;
+    ushort_t* zo = dz ? dz + (size_t)slot * VEC : nullptr;
     for (long row = (long)blockIdx.x * rpb + rg; row < m;
             row += (long)gridDim.x * rpb) {
-        Vec8 vx, vd, out;
+        Vec8 vx, vz, vd, out, outz;
         vx.f4 = *reinterpret_cast<const float4*>(xb + (size_t)row * c);
+        if (zb)
+            vz.f4 = *reinterpret_cast<const float4*>(zb + (size_t)row * c);
         vd.f4 = *reinterpret_cast<const float4*>(db + (size_t)row * c);
         #pragma unroll
         for (int k = 0; k < VEC; ++k) {
             float fx = b2f(vx.u[k]);
             float g = b2f(vd.u[k]);
-            if (relu && fmaf(fx, sc[k], sh[k]) <= 0.f) g = 0.f;
+            if (relu) {
+                float pre = fmaf(fx, sc[k], sh[k]);
+                if (zb) pre += b2f(vz.u[k]);
+                if (pre <= 0.f) g = 0.f;
+            }
+            if (zo) outz.u[k] = f2b(g);
             out.u[k] = f2b(fmaf(cp[k], g, fmaf(-cq[k], fx, cr[k])));
         }
         *reinterpret_cast<float4*>(ob + (size_t)row * c) = out.f4;
+        if (zo)
+            *reinterpret_cast<float4*>(zo + (size_t)row * c) = outz.f4;
     }
 }
 
@@ -374,7 +401,7 @@ static inline unsigned bn_grid(long work_items, int per_block, long cap) {
 }
 
 extern "C" void launch_bn_fwd(
-        const ushort_t* x, ushort_t* y, long m, int c,
+        const ushort_t* x, const ushort_t* z, ushort_t* y, long m, int c,
         const float* gamma, const float* beta, float* running_mean,
         float* running_var, float momentum, float eps, int train, int relu,
         float* ws, float* sums, float* save_mean, float* save_rstd,
@@ -399,11 +426,13 @@ extern "C" void launch_bn_fwd(
                            save_mean, save_rstd, scale, shift);
     }
     hipLaunchKernelGGL(k_bn_fwd_apply, dim3(bn_grid(m, rpb, MAX_BLOCKS)),
-                       dim3(BLOCK), 0, s, x, y, m, c, scale, shift, relu);
+                       dim3(BLOCK), 0, s, x, z, y, m, c, scale, shift,
+                       relu);
 }
 
 extern "C" void launch_bn_bwd(
-        const ushort_t* x, const ushort_t* dy, ushort_t* dx, long m, int c,
+        const ushort_t* x, const ushort_t* z, const ushort_t* dy,
+        ushort_t* dx, ushort_t* dz, long m, int c,
         const float* gamma, const float* save_mean, const float* save_rstd,
         const float* scale, const float* shift, int train, int relu,
         float* ws, float* sums, float* dgamma, float* dbeta, float* pqr,
@@ -412,14 +441,14 @@ extern "C" void launch_bn_bwd(
     const int rpb = BLOCK / slots > 0 ? BLOCK / slots : 1;
     const unsigned nb = bn_grid(m, rpb, MAX_RBLOCKS);
     hipLaunchKernelGGL(k_bn_bwd_reduce, dim3(nb), dim3(BLOCK), 0, s,
-                       x, dy, m, c, (int)nb, save_mean, save_rstd, scale,
-                       shift, relu, ws);
+                       x, z, dy, m, c, (int)nb, save_mean, save_rstd,
+                       scale, shift, relu, ws);
     hipLaunchKernelGGL(k_bn_reduce_ws, dim3(2 * c), dim3(BLOCK), 0, s,
                        ws, (int)nb, sums);
     hipLaunchKernelGGL(k_bn_bwd_finalize, dim3((c + BLOCK - 1) / BLOCK),
                        dim3(BLOCK), 0, s, sums, m, c, gamma, save_mean,
                        save_rstd, train, dgamma, dbeta, pqr);
     hipLaunchKernelGGL(k_bn_bwd_apply, dim3(bn_grid(m, rpb, MAX_BLOCKS)),
-                       dim3(BLOCK), 0, s, x, dy, dx, m, c, scale, shift,
-                       pqr, relu);
+                       dim3(BLOCK), 0, s, x, z, dy, dx, dz, m, c, scale,
+                       shift, pqr, relu);
 }

@@ -37,8 +37,8 @@ def _hip_bn_ok(x):
 
 class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps, relu):
+    def forward(ctx, x, z, weight, bias, running_mean, running_var,
+                training, momentum, eps, relu):
         ext = ops._load_extension()
         y = torch.empty_like(x)
         c = x.shape[1]
@@ -52,22 +52,28 @@ class _FusedBNFunction(torch.autograd.Function):
         scale = torch.empty(c, **opt)
         shift = torch.empty(c, **opt)
         none = torch.empty(0, **opt)
-        ext.bn_fwd(x, y, weight, bias,
+        znone = torch.empty(0, dtype=x.dtype, device=x.device)
+        ext.bn_fwd(x, z if z is not None else znone, y, weight, bias,
                    running_mean if running_mean is not None else none,
                    running_var if running_var is not None else none,
                    float(momentum), float(eps), bool(training), bool(relu),
                    ws, sums, save_mean, save_rstd, scale, shift)
-        ctx.save_for_backward(x, weight, save_mean, save_rstd, scale, shift)
+        ctx.save_for_backward(x, z if z is not None else znone, weight,
+                              save_mean, save_rstd, scale, shift)
         ctx.bn_train = bool(training)
         ctx.bn_relu = bool(relu)
+        ctx.bn_has_z = z is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = ops._load_extension()
-        x, weight, save_mean, save_rstd, scale, shift = ctx.saved_tensors
+        x, z, weight, save_mean, save_rstd, scale, shift = \
+            ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = torch.empty_like(x)
+        dz = torch.empty_like(x) if ctx.bn_has_z else \
+            torch.empty(0, dtype=x.dtype, device=x.device)
         c = x.shape[1]
         m = x.numel() // c
         opt = dict(dtype=torch.float32, device=x.device)
@@ -76,9 +82,11 @@ class _FusedBNFunction(torch.autograd.Function):
         dgamma = torch.empty(c, **opt)
         dbeta = torch.empty(c, **opt)
         pqr = torch.empty(3 * c, **opt)
-        ext.bn_bwd(x, dy, dx, weight, save_mean, save_rstd, scale, shift,
-                   ctx.bn_train, ctx.bn_relu, ws, sums, dgamma, dbeta, pqr)
-        return (dx, dgamma, dbeta) + (None,) * 6
+        ext.bn_bwd(x, z, dy, dx, dz, weight, save_mean, save_rstd, scale,
+                   shift, ctx.bn_train, ctx.bn_relu, ws, sums, dgamma,
+                   dbeta, pqr)
+        return (dx, dz if ctx.bn_has_z else None, dgamma, dbeta) + \
+            (None,) * 6
 
 
 class FusedBatchNormAct2d(nn.BatchNorm2d):
@@ -95,16 +103,21 @@ class FusedBatchNormAct2d(nn.BatchNorm2d):
                          track_running_stats=track_running_stats)
         self.relu = relu
 
-    def forward(self, x):
+    def forward(self, x, residual=None):
         use_batch_stats = self.training or not self.track_running_stats
-        if _hip_bn_ok(x) and self.affine:
+        if _hip_bn_ok(x) and self.affine and \
+                (residual is None or
+                 (residual.dtype == x.dtype and residual.is_contiguous(
+                     memory_format=torch.channels_last))):
             if self.training and self.track_running_stats and \
                     self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
             momentum = self.momentum if self.momentum is not None else 0.0
             return _FusedBNFunction.apply(
-                x, self.weight, self.bias, self.running_mean,
+                x, residual, self.weight, self.bias, self.running_mean,
                 self.running_var, use_batch_stats, momentum, self.eps,
                 self.relu)
         y = super().forward(x)
+        if residual is not None:
+            y = y + residual
         return F.relu(y) if self.relu else y

@@ -121,3 +121,58 @@ def test_gpu_fused_bn_used_in_resnet():
     finally:
         L._FusedBNFunction.apply = orig
     assert len(calls) == 20  # every BN layer of ResNet-18 took the HIP path
+
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_cpu_fallback_residual(relu):
+    torch.manual_seed(5)
+    bn = FusedBatchNormAct2d(8, relu=relu)
+    ref = torch.nn.BatchNorm2d(8)
+    ref.load_state_dict(bn.state_dict())
+    x = torch.randn(4, 8, 6, 6, requires_grad=True)
+    z = torch.randn(4, 8, 6, 6, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    z2 = z.detach().clone().requires_grad_(True)
+    y = bn(x, residual=z)
+    yr = ref(x2) + z2
+    if relu:
+        yr = F.relu(yr)
+    assert torch.allclose(y, yr, atol=1e-6)
+    y.sum().backward()
+    yr.sum().backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+    assert torch.allclose(z.grad, z2.grad, atol=1e-6)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("relu", [False, True])
+def test_gpu_fused_bn_residual(relu):
+    torch.manual_seed(6)
+    c = 64
+    dev = torch.device("cuda")
+    bn = FusedBatchNormAct2d(c, relu=relu).to(dev)
+    ref = torch.nn.BatchNorm2d(c).to(dev)
+    ref.load_state_dict(bn.state_dict())
+
+    x = (torch.randn(8, c, 8, 8, device=dev)).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    z = (torch.randn(8, c, 8, 8, device=dev)).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x32 = x.detach().float().requires_grad_(True)
+    z32 = z.detach().float().requires_grad_(True)
+
+    y = bn(x, residual=z)
+    yr = ref(x32) + z32
+    if relu:
+        yr = F.relu(yr)
+    assert torch.allclose(y.float(), yr, atol=5e-2, rtol=5e-2)
+
+    dy = torch.randn_like(yr)
+    y.backward(dy.to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last))
+    yr.backward(dy)
+    assert torch.allclose(x.grad.float(), x32.grad, atol=7e-2, rtol=7e-2)
+    assert torch.allclose(z.grad.float(), z32.grad, atol=7e-2, rtol=7e-2)
+    assert torch.allclose(bn.weight.grad, ref.weight.grad,
+                          atol=2e-1, rtol=2e-2)
+    assert torch.allclose(bn.bias.grad, ref.bias.grad, atol=2e-1, rtol=2e-2)
