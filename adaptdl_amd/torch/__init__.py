@@ -8,6 +8,7 @@ rebuilt MI355X-first (RCCL over xGMI data plane, fused HIP statistics).
 """
 
 import logging
+import os
 import socket
 
 import torch.distributed
@@ -77,10 +78,18 @@ def init_process_group(backend=None, init_method=None, world_size=None,
         return
 
     if init_method is None:
-        # Broadcast a fresh rendezvous port from rank 0 over the control
-        # plane (a new port every (re)start avoids TIME_WAIT collisions).
-        port = adaptdl_amd.collective.broadcast(_pick_free_port())
-        init_method = "tcp://{}:{}".format(master_addr, port)
+        if os.getenv("TORCHELASTIC_RUN_ID") is not None:
+            # Under torchrun the agent process hosts the c10d store and
+            # every worker is a store CLIENT (TORCHELASTIC_USE_AGENT_STORE)
+            # — a fresh tcp:// port would have no server.  Use the
+            # torchrun-provided env:// rendezvous directly.
+            init_method = "env://"
+        else:
+            # Broadcast a fresh rendezvous port from rank 0 over the
+            # control plane (a new port every (re)start avoids TIME_WAIT
+            # collisions).
+            port = adaptdl_amd.collective.broadcast(_pick_free_port())
+            init_method = "tcp://{}:{}".format(master_addr, port)
     if backend == "nccl":
         torch.cuda.set_device(adaptdl_amd.env.local_rank()
                               % max(torch.cuda.device_count(), 1))
