@@ -9,7 +9,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from adaptdl_amd.torch.layers import FusedBatchNormAct2d
+from adaptdl_amd.torch.layers import (FusedBatchNormAct2d,
+                                      FusedConv2d)
 
 
 def _bn(planes, relu=False):
@@ -22,11 +23,11 @@ class BasicBlock(nn.Module):
 
     def __init__(self, in_planes, planes, stride=1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride,
-                               padding=1, bias=False)
+        self.conv1 = FusedConv2d(in_planes, planes, 3, stride=stride,
+                                 padding=1, bias=False)
         self.bn1 = _bn(planes, relu=True)
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1,
-                               bias=False)
+        self.conv2 = FusedConv2d(planes, planes, 3, stride=1, padding=1,
+                                 bias=False)
         # bn2 fuses the residual add + final ReLU of the block.
         self.bn2 = _bn(planes, relu=True)
         self.shortcut = nn.Sequential()
@@ -48,8 +49,8 @@ class Bottleneck(nn.Module):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
         self.bn1 = _bn(planes, relu=True)
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1,
-                               bias=False)
+        self.conv2 = FusedConv2d(planes, planes, 3, stride=stride,
+                                 padding=1, bias=False)
         self.bn2 = _bn(planes, relu=True)
         self.conv3 = nn.Conv2d(planes, self.expansion * planes, 1,
                                bias=False)

@@ -33,6 +33,11 @@ void launch_bn_bwd(const unsigned short*, const unsigned short*,
                    long, int, const float*, const float*, const float*,
                    const float*, const float*, int, int, float*, float*,
                    float*, float*, float*, hipStream_t);
+int conv3x3_wrw_supported(int, int, int, int);
+int conv3x3_wrw_nsplit(int, int, int, int, int);
+void launch_conv3x3_wrw(const unsigned short*, const unsigned short*,
+                        float*, float*, int, int, int, int, int,
+                        hipStream_t);
 }
 
 // Mirror of the kernel-side stage-1 grid sizing (bn_kernels.hip).
@@ -242,6 +247,38 @@ void bn_bwd(torch::Tensor x, torch::Tensor z, torch::Tensor dy,
                   pqr.data_ptr<float>(), stream());
 }
 
+// ---- custom MFMA conv 3x3 wrw ---------------------------------------
+
+bool conv_wrw_ok(long N, long H, long W, long C, long K) {
+    return conv3x3_wrw_supported((int)H, (int)W, (int)C, (int)K) != 0;
+}
+
+long conv_wrw_nsplit(long N, long H, long W, long C, long K) {
+    return conv3x3_wrw_nsplit((int)N, (int)H, (int)W, (int)C, (int)K);
+}
+
+// dw = wrw(x, dy): x (N,C,H,W) channels_last bf16, dy (N,K,H,W)
+// channels_last bf16, dw (K,C,3,3) channels_last fp32, ws workspace.
+void conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor ws,
+              torch::Tensor dw) {
+    check_bn_x(x, "x"); check_bn_x(dy, "dy");
+    const long N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+    const long K = dy.size(1);
+    TORCH_CHECK(dy.size(0) == N && dy.size(2) == H && dy.size(3) == W,
+                "dy shape mismatch");
+    TORCH_CHECK(conv_wrw_ok(N, H, W, C, K), "unsupported wrw shape");
+    TORCH_CHECK(dw.is_cuda() && dw.scalar_type() == torch::kFloat32 &&
+                dw.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                dw.size(0) == K && dw.size(1) == C && dw.size(2) == 3 &&
+                dw.size(3) == 3, "dw must be (K,C,3,3) channels_last f32");
+    const long nsplit = conv_wrw_nsplit(N, H, W, C, K);
+    check_vecf(ws, nsplit * K * 9 * C, "ws");
+    launch_conv3x3_wrw((const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)dy.data_ptr(),
+                       ws.data_ptr<float>(), dw.data_ptr<float>(),
+                       (int)N, (int)H, (int)W, (int)C, (int)K, stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -257,4 +294,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("fused_adamw", &fused_adamw, "fused flat-bucket Adam(W) step");
     mod.def("bn_fwd", &bn_fwd, "fused NHWC bf16 BatchNorm(+ReLU) forward");
     mod.def("bn_bwd", &bn_bwd, "fused NHWC bf16 BatchNorm(+ReLU) backward");
+    mod.def("conv_wrw_ok", &conv_wrw_ok, "3x3 wrw fast-path predicate");
+    mod.def("conv_wrw_nsplit", &conv_wrw_nsplit, "wrw workspace splits");
+    mod.def("conv_wrw", &conv_wrw, "MFMA 3x3 s1 NHWC weight gradient");
 }

@@ -1,0 +1,282 @@
+// CDNA4 (gfx950) MFMA conv kernels: 3x3 stride-1 NHWC weight gradient.
+//
+// MIOpen's igemm wrw kernels are the single worst hot spot of the
+// flagship ResNet workload (profiles/bench_r03: 256 us where the MFMA
+// floor is ~33 us, plus SubTensorOp zero/cast fills).  This kernel
+// computes dW[k][tau][c] = sum_{n,h,w} dy[n,h,w,k] * x[n,h+dh,w+dw,c]
+// as an implicit GEMM with v_mfma_f32_16x16x32_bf16:
+//
+//   - a "chunk" is P consecutive h-lines of one image (P*W pixels, a
+//     multiple of 32 = the MFMA K dimension),
+//   - dy and x chunk tiles are staged TRANSPOSED in LDS
+//     (dy_t[k][pixel], x_t[c][line-padded pixel]) so both MFMA operands
+//     read contiguous 16-byte fragments (ds_read_b128),
+//   - x lines carry left/right zero pads and halo lines, so all nine
+//     (dh, dw) taps come from two ALIGNED b128 reads per (dh, c-frag)
+//     window plus compile-time byte-rotations for dw = 0, +1
+//     (w-offsets are multiples of 8 by construction),
+//   - each wave owns a 32k x 32c output tile: 2x2 fragments x 9 taps =
+//     36 fp32 accumulators fragments; a 4-wave block owns 64k x 64c,
+//   - grid = (K/64) * (C/64) * SPLIT blocks; each split accumulates its
+//     share of chunks and writes one fp32 partial slab; a second kernel
+//     reduces the slabs (no cross-block atomics - see bn_kernels.hip
+//     for the measured atomic-serialization lesson).
+//
+// MFMA operand maps verified on hardware by tools/mfma_probe.hip:
+//   A[m][k]: lane l elem e -> m = l%16, k = (l/16)*8+e
+//   B[k][n]: lane l elem e -> k = (l/16)*8+e, n = l%16
+//   D[m][n]: lane l reg  r -> m = (l>>4)*4+r, n = l&15
+//
+// Constraints: 3x3, stride 1, pad 1, dilation 1, groups 1, NHWC bf16,
+// C % 64 == 0, K % 64 == 0, W in {8, 16, 32}, H % P == 0.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+
+typedef unsigned short ushort_t;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+#define DY_STRIDE 136        // shorts per dy_t row (>=128, 16B-aligned,
+                             // 68 dwords -> conflict-free group reads)
+#define XT_STRIDE 248        // shorts per x_t row (>= (P+2)*LS, aligned)
+#define CPMAX 128
+
+union V16 {
+    u32x4 u4;
+    unsigned int u[4];
+    short s[8];
+};
+
+extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+        float* __restrict__ ws, int N, int H, int W, int C, int K,
+        int P, int nsplit) {
+    __shared__ short dy_t[64 * DY_STRIDE];
+    __shared__ short x_t[64 * XT_STRIDE];
+
+    const int LS = W + 8;            // padded line stride (mult of 8)
+    const int CP = P * W;            // chunk pixels (mult of 32)
+    const int kchunks = CP / 32;
+    const int nc = C / 64;
+    const int nsplit_t = nsplit;
+    // block id -> (ktile, ctile, split)
+    const int tile = blockIdx.x / nsplit_t;
+    const int split = blockIdx.x % nsplit_t;
+    const int kt = tile / nc;
+    const int ct = tile % nc;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wk = wid >> 1;         // wave k-subtile (0..1)
+    const int wc = wid & 1;          // wave c-subtile (0..1)
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+
+    // 36 accumulator fragments: [mf][nf][tau]
+    f32x4 acc[2][2][9];
+    #pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+            #pragma unroll
+            for (int tau = 0; tau < 9; ++tau)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[mf][nf][tau][r] = 0.f;
+
+    const int lines_per_img = H / P;
+    const long chunks_total = (long)N * lines_per_img;
+
+    for (long q = split; q < chunks_total; q += nsplit_t) {
+        const int n = (int)(q / lines_per_img);
+        const int h0 = (int)(q % lines_per_img) * P;
+
+        __syncthreads();  // previous iteration's reads done
+        // ---- zero x_t (pads + possibly-skipped halo lines) ----
+        for (int i = t; i < 64 * XT_STRIDE / 2; i += 256)
+            reinterpret_cast<unsigned int*>(x_t)[i] = 0u;
+        __syncthreads();
+
+        // ---- stage dy chunk transposed: dy_t[k][p] ----
+        {
+            const int count16 = CP * 8;   // 16B loads: 8 k's of one pixel
+            for (int i = t; i < count16; i += 256) {
+                const int p = i >> 3;
+                const int kg = (i & 7) * 8;
+                const int h = h0 + p / W;
+                const int w = p % W;
+                const ushort_t* g = dy +
+                    (((size_t)n * H + h) * W + w) * K + (size_t)kt * 64
+                    + kg;
+                V16 v;
+                v.u4 = *reinterpret_cast<const u32x4*>(g);
+                #pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    dy_t[(kg + j) * DY_STRIDE + p] = v.s[j];
+            }
+        }
+        // ---- stage x window transposed with pads: x_t[c][line, w] ----
+        {
+            const int count16 = (P + 2) * W * 8;
+            for (int i = t; i < count16; i += 256) {
+                const int j = i / (W * 8);          // window line 0..P+1
+                const int rem = i % (W * 8);
+                const int w = rem >> 3;
+                const int cg = (rem & 7) * 8;
+                const int h = h0 - 1 + j;
+                if (h < 0 || h >= H) continue;      // stays zero
+                const ushort_t* g = x +
+                    (((size_t)n * H + h) * W + w) * C + (size_t)ct * 64
+                    + cg;
+                V16 v;
+                v.u4 = *reinterpret_cast<const u32x4*>(g);
+                #pragma unroll
+                for (int jj = 0; jj < 8; ++jj)
+                    x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w] = v.s[jj];
+            }
+        }
+        __syncthreads();
+
+        // ---- MFMA over the chunk's 32-pixel sub-chunks ----
+        for (int kc = 0; kc < kchunks; ++kc) {
+            const int p0 = kc * 32 + slot8;      // this lane's pixels
+            const int li = p0 / W;               // chunk line index
+            const int w0 = p0 % W;               // multiple of 8
+            // A fragments (dy): 16B each
+            bf16x8 afrag[2];
+            #pragma unroll
+            for (int mf = 0; mf < 2; ++mf) {
+                const short* a = &dy_t[(wk * 32 + mf * 16 + row16)
+                                       * DY_STRIDE + kc * 32 + slot8];
+                afrag[mf] = *reinterpret_cast<const bf16x8*>(a);
+            }
+            #pragma unroll
+            for (int dh = 0; dh < 3; ++dh) {
+                // 32-byte aligned window [w0-1-... ] per c-fragment:
+                // window pixels (li+dh)*LS + 4 + w0 - 1 - 3  ... the
+                // load base is 16B-aligned because (4 + w0 - 4) = w0.
+                #pragma unroll
+                for (int nf = 0; nf < 2; ++nf) {
+                    const short* b = &x_t[(wc * 32 + nf * 16 + row16)
+                                          * XT_STRIDE
+                                          + (li + dh) * LS + w0];
+                    // pixels [w-4 .. w+12) of the padded line (pad=4)
+                    V16 lo, hi;
+                    lo.u4 = *reinterpret_cast<const u32x4*>(b);
+                    hi.u4 = *reinterpret_cast<const u32x4*>(b + 8);
+                    unsigned int win[8] = {lo.u[0], lo.u[1], lo.u[2],
+                                           lo.u[3], hi.u[0], hi.u[1],
+                                           hi.u[2], hi.u[3]};
+                    // shifted vectors: tap pixel = w + dw, base w-1
+                    // window starts at pad+w0-4+... byte of pixel
+                    // (w0 + dw - 1 + 4 - w0) = 3+dw shorts into window
+                    #pragma unroll
+                    for (int dw = 0; dw < 3; ++dw) {
+                        u32x4 frag;
+                        const int sh = (3 + dw) * 2;  // byte offset 6,8,10
+                        const int d0 = sh >> 2;       // 1, 2, 2
+                        const int rem = sh & 3;       // 2, 0, 2
+                        if (rem == 0) {
+                            frag[0] = win[d0]; frag[1] = win[d0 + 1];
+                            frag[2] = win[d0 + 2]; frag[3] = win[d0 + 3];
+                        } else {
+                            frag[0] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 1], win[d0], rem);
+                            frag[1] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 2], win[d0 + 1], rem);
+                            frag[2] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 3], win[d0 + 2], rem);
+                            frag[3] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 4], win[d0 + 3], rem);
+                        }
+                        const bf16x8 bfrag =
+                            *reinterpret_cast<const bf16x8*>(&frag);
+                        const int tau = dh * 3 + dw;
+                        #pragma unroll
+                        for (int mf = 0; mf < 2; ++mf)
+                            acc[mf][nf][tau] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag[mf], bfrag, acc[mf][nf][tau],
+                                    0, 0, 0);
+                    }
+                }
+            }
+        }
+    }
+
+    // ---- epilogue: write the fp32 partial slab ----
+    // D mapping: lane l reg r -> m = (l>>4)*4 + r, n = l&15.
+    float* slab = ws + (size_t)split * K * 9 * C;
+    #pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+            #pragma unroll
+            for (int tau = 0; tau < 9; ++tau) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int klocal = wk * 32 + mf * 16
+                        + ((lane >> 4) * 4 + r);
+                    const int clocal = wc * 32 + nf * 16 + (lane & 15);
+                    slab[((size_t)(kt * 64 + klocal) * 9 + tau) * C
+                         + ct * 64 + clocal] = acc[mf][nf][tau][r];
+                }
+            }
+        }
+    }
+}
+
+// dW[i] = sum_s ws[s * n + i]; coalesced over i.
+extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
+        const float* __restrict__ ws, float* __restrict__ dw, long n,
+        int nsplit) {
+    long i = (long)blockIdx.x * 256 + threadIdx.x;
+    const long stride = (long)gridDim.x * 256;
+    for (; i < n; i += stride) {
+        float s = 0.f;
+        for (int sp = 0; sp < nsplit; ++sp)
+            s += ws[(size_t)sp * n + i];
+        dw[i] = s;
+    }
+}
+
+// ---- host-side launchers ---------------------------------------------
+
+extern "C" int conv3x3_wrw_supported(int H, int W, int C, int K) {
+    if (C % 64 || K % 64) return 0;
+    if (W != 8 && W != 16 && W != 32) return 0;
+    int P = (W == 32) ? 4 : 8;
+    if (H % P) return 0;
+    if (H < 2) return 0;
+    return 1;
+}
+
+extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
+    int P = (W == 32) ? 4 : 8;
+    long chunks = (long)N * (H / P);
+    int tiles = (K / 64) * (C / 64);
+    long target = 2048 / tiles;
+    if (target < 1) target = 1;
+    if (target > chunks) target = chunks;
+    return (int)target;
+}
+
+extern "C" void launch_conv3x3_wrw(
+        const ushort_t* x, const ushort_t* dy, float* ws, float* dw,
+        int N, int H, int W, int C, int K, hipStream_t s) {
+    const int P = (W == 32) ? 4 : 8;
+    const int nsplit = conv3x3_wrw_nsplit(N, H, W, C, K);
+    const int tiles = (K / 64) * (C / 64);
+    hipLaunchKernelGGL(k_conv3x3_wrw, dim3(tiles * nsplit), dim3(256), 0,
+                       s, x, dy, ws, N, H, W, C, K, P, nsplit);
+    const long n = (long)K * 9 * C;
+    long blocks = (n + 255) / 256;
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(k_wrw_reduce, dim3((unsigned)blocks), dim3(256), 0,
+                       s, ws, dw, n, nsplit);
+}

@@ -121,3 +121,72 @@ class FusedBatchNormAct2d(nn.BatchNorm2d):
         if residual is not None:
             y = y + residual
         return F.relu(y) if self.relu else y
+
+
+class _FusedConvFunction(torch.autograd.Function):
+    """conv2d with the weight gradient computed by the MFMA wrw kernel.
+
+    Forward runs the library conv (MIOpen); backward computes
+    grad_input via aten.convolution_backward and grad_weight via
+    ops/hip/conv_kernels.hip (3x3 stride-1 NHWC bf16, fp32 output).
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding, dilation, groups):
+        wb = weight.detach().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        y = torch.nn.functional.conv2d(x, wb, None, stride, padding,
+                                       dilation, groups)
+        ctx.save_for_backward(x, wb)
+        ctx.conv_args = (stride, padding, dilation, groups)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops._load_extension()
+        x, wb = ctx.saved_tensors
+        stride, padding, dilation, groups = ctx.conv_args
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.ops.aten.convolution_backward(
+                dy, x, wb, None, stride, padding, dilation, False,
+                [0, 0], groups, [True, False, False])[0]
+        n, c, h, w = x.shape
+        k = dy.shape[1]
+        nsplit = ext.conv_wrw_nsplit(n, h, w, c, k)
+        ws = torch.empty(nsplit * k * 9 * c, dtype=torch.float32,
+                         device=x.device)
+        dw = torch.empty(k, c, 3, 3, dtype=torch.float32,
+                         device=x.device) \
+            .contiguous(memory_format=torch.channels_last)
+        ext.conv_wrw(x, dy, ws, dw)
+        return dx, dw, None, None, None, None
+
+
+class FusedConv2d(nn.Conv2d):
+    """Conv2d whose 3x3/s1 weight gradient runs the MFMA wrw kernel."""
+
+    def _wrw_ok(self, x):
+        if not (x.is_cuda and x.dtype == torch.bfloat16 and
+                x.dim() == 4 and ops.has_extension() and
+                self.bias is None and self.groups == 1 and
+                self.kernel_size == (3, 3) and self.stride == (1, 1) and
+                self.padding == (1, 1) and self.dilation == (1, 1) and
+                x.is_contiguous(memory_format=torch.channels_last)):
+            return False
+        ext = ops._load_extension()
+        n, c, h, w = x.shape
+        return bool(ext.conv_wrw_ok(n, h, w, c, self.out_channels))
+
+    def forward(self, x):
+        if torch.is_autocast_enabled() and x.is_cuda and \
+                x.dtype != torch.bfloat16 and \
+                torch.get_autocast_dtype("cuda") == torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        if self._wrw_ok(x) and (self.weight.requires_grad or
+                                x.requires_grad):
+            return _FusedConvFunction.apply(x, self.weight, self.stride,
+                                            self.padding, self.dilation,
+                                            self.groups)
+        return super().forward(x)

@@ -23,6 +23,7 @@ setup(
                 "adaptdl_amd/ops/hip/bindings.cpp",
                 "adaptdl_amd/ops/hip/gns_kernels.hip",
                 "adaptdl_amd/ops/hip/bn_kernels.hip",
+                "adaptdl_amd/ops/hip/conv_kernels.hip",
             ],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
