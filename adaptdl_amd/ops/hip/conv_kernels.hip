@@ -51,7 +51,7 @@ union V16 {
     short s[8];
 };
 
-extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
+extern "C" __global__ __launch_bounds__(256, 2) void k_conv3x3_wrw(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
         float* __restrict__ ws, int N, int H, int W, int C, int K,
         int P, int nsplit) {
@@ -92,15 +92,23 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
     const int lines_per_img = H / P;
     const long chunks_total = (long)N * lines_per_img;
 
+    // One-time zeroing of the left/right pad columns (data writes never
+    // touch them; out-of-range halo lines are written as zeros in the
+    // staging loop itself).
+    for (int i = t; i < (P + 2) * 8 * 64; i += 256) {
+        const int cc = i / ((P + 2) * 8);
+        const int rem = i % ((P + 2) * 8);
+        const int j = rem / 8;
+        const int pp = rem % 8;
+        const int q = pp < 4 ? pp : (4 + W + (pp - 4));
+        x_t[cc * XT_STRIDE + j * LS + q] = 0;
+    }
+
     for (long q = split; q < chunks_total; q += nsplit_t) {
         const int n = (int)(q / lines_per_img);
         const int h0 = (int)(q % lines_per_img) * P;
 
         __syncthreads();  // previous iteration's reads done
-        // ---- zero x_t (pads + possibly-skipped halo lines) ----
-        for (int i = t; i < 64 * XT_STRIDE / 2; i += 256)
-            reinterpret_cast<unsigned int*>(x_t)[i] = 0u;
-        __syncthreads();
 
         // ---- stage dy chunk transposed: dy_t[k][p] ----
         {
@@ -129,12 +137,15 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
                 const int w = rem >> 3;
                 const int cg = (rem & 7) * 8;
                 const int h = h0 - 1 + j;
-                if (h < 0 || h >= H) continue;      // stays zero
-                const ushort_t* g = x +
-                    (((size_t)n * H + h) * W + w) * C + (size_t)ct * 64
-                    + cg;
                 V16 v;
-                v.u4 = *reinterpret_cast<const u32x4*>(g);
+                if (h < 0 || h >= H) {
+                    v.u4 = u32x4{0u, 0u, 0u, 0u};
+                } else {
+                    const ushort_t* g = x +
+                        (((size_t)n * H + h) * W + w) * C
+                        + (size_t)ct * 64 + cg;
+                    v.u4 = *reinterpret_cast<const u32x4*>(g);
+                }
                 #pragma unroll
                 for (int jj = 0; jj < 8; ++jj)
                     x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w] = v.s[jj];
@@ -231,17 +242,41 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
     }
 }
 
-// dW[i] = sum_s ws[s * n + i]; coalesced over i.
+// dW[i] = sum_s ws[s * n + i].  Hierarchical: each block sums a
+// <=64-split group over a 1024-float i-chunk (f32x4 per thread), then
+// atomically adds into dw (zeroed by the launcher when several groups).
 extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
         const float* __restrict__ ws, float* __restrict__ dw, long n,
-        int nsplit) {
-    long i = (long)blockIdx.x * 256 + threadIdx.x;
-    const long stride = (long)gridDim.x * 256;
-    for (; i < n; i += stride) {
-        float s = 0.f;
-        for (int sp = 0; sp < nsplit; ++sp)
-            s += ws[(size_t)sp * n + i];
-        dw[i] = s;
+        int nsplit, int ngroups, long nchunks) {
+    const long work = nchunks * ngroups;
+    for (long b = blockIdx.x; b < work; b += gridDim.x) {
+        const long ic = b % nchunks;
+        const int sg = (int)(b / nchunks);
+        const long i0 = ic * 1024 + (long)threadIdx.x * 4;
+        if (i0 >= n) continue;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        const int s_end = min(nsplit, (sg + 1) * 64);
+        const bool tail = i0 + 4 > n;
+        for (int sp = sg * 64; sp < s_end; ++sp) {
+            const float* src = ws + (size_t)sp * n + i0;
+            if (!tail) {
+                f32x4 v = *reinterpret_cast<const f32x4*>(src);
+                acc[0] += v[0]; acc[1] += v[1];
+                acc[2] += v[2]; acc[3] += v[3];
+            } else {
+                for (long j = 0; i0 + j < n; ++j) acc[j] += src[j];
+            }
+        }
+        if (ngroups == 1) {
+            if (!tail) {
+                *reinterpret_cast<f32x4*>(dw + i0) = acc;
+            } else {
+                for (long j = 0; i0 + j < n; ++j) dw[i0 + j] = acc[j];
+            }
+        } else {
+            for (long j = 0; j < 4 && i0 + j < n; ++j)
+                atomicAdd(&dw[i0 + j], acc[j]);
+        }
     }
 }
 
@@ -260,7 +295,7 @@ extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
     int P = (W == 32) ? 4 : 8;
     long chunks = (long)N * (H / P);
     int tiles = (K / 64) * (C / 64);
-    long target = 2048 / tiles;
+    long target = 640 / tiles;
     if (target < 1) target = 1;
     if (target > chunks) target = chunks;
     return (int)target;
@@ -275,8 +310,12 @@ extern "C" void launch_conv3x3_wrw(
     hipLaunchKernelGGL(k_conv3x3_wrw, dim3(tiles * nsplit), dim3(256), 0,
                        s, x, dy, ws, N, H, W, C, K, P, nsplit);
     const long n = (long)K * 9 * C;
-    long blocks = (n + 255) / 256;
-    if (blocks > 2048) blocks = 2048;
+    const int ngroups = (nsplit + 63) / 64;
+    const long nchunks = (n + 1023) / 1024;
+    if (ngroups > 1)
+        hipMemsetAsync(dw, 0, n * sizeof(float), s);
+    long blocks = nchunks * ngroups;
+    if (blocks > 4096) blocks = 4096;
     hipLaunchKernelGGL(k_wrw_reduce, dim3((unsigned)blocks), dim3(256), 0,
-                       s, ws, dw, n, nsplit);
+                       s, ws, dw, n, nsplit, ngroups, nchunks);
 }
