@@ -51,7 +51,7 @@ union V16 {
     short s[8];
 };
 
-extern "C" __global__ __launch_bounds__(256, 2) void k_conv3x3_wrw(
+extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
         float* __restrict__ ws, int N, int H, int W, int C, int K,
         int P, int nsplit) {
@@ -110,45 +110,85 @@ extern "C" __global__ __launch_bounds__(256, 2) void k_conv3x3_wrw(
 
         __syncthreads();  // previous iteration's reads done
 
-        // ---- stage dy chunk transposed: dy_t[k][p] ----
+        // ---- stage dy and x transposed.  Global loads first (one HBM
+        // latency per operand batch), then LDS writes as PACKED b32
+        // pixel-pairs: 2-byte scatter writes measured 44M bank-conflict
+        // cycles per dispatch (SQ_LDS_BANK_CONFLICT); pairing pixels
+        // (p, p+1) into one dword write quarters the conflict cost. ----
         {
-            const int count16 = CP * 8;   // 16B loads: 8 k's of one pixel
-            for (int i = t; i < count16; i += 256) {
-                const int p = i >> 3;
-                const int kg = (i & 7) * 8;
-                const int h = h0 + p / W;
-                const int w = p % W;
-                const ushort_t* g = dy +
-                    (((size_t)n * H + h) * W + w) * K + (size_t)kt * 64
-                    + kg;
-                V16 v;
-                v.u4 = *reinterpret_cast<const u32x4*>(g);
-                #pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    dy_t[(kg + j) * DY_STRIDE + p] = v.s[j];
-            }
-        }
-        // ---- stage x window transposed with pads: x_t[c][line, w] ----
-        {
-            const int count16 = (P + 2) * W * 8;
-            for (int i = t; i < count16; i += 256) {
-                const int j = i / (W * 8);          // window line 0..P+1
-                const int rem = i % (W * 8);
-                const int w = rem >> 3;
-                const int cg = (rem & 7) * 8;
-                const int h = h0 - 1 + j;
-                V16 v;
-                if (h < 0 || h >= H) {
-                    v.u4 = u32x4{0u, 0u, 0u, 0u};
-                } else {
-                    const ushort_t* g = x +
-                        (((size_t)n * H + h) * W + w) * C
-                        + (size_t)ct * 64 + cg;
-                    v.u4 = *reinterpret_cast<const u32x4*>(g);
+            const int dyn8 = CP * 4;             // pair-loads: <=512
+            V16 vdy[2][2];
+            #pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int i = t + it * 256;
+                if (i < dyn8) {
+                    const int p = (i >> 3) * 2;
+                    const ushort_t* g = dy +
+                        (((size_t)n * H + (h0 + p / W)) * W + p % W) * K
+                        + (size_t)kt * 64 + (i & 7) * 8;
+                    vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+                    vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
                 }
-                #pragma unroll
-                for (int jj = 0; jj < 8; ++jj)
-                    x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w] = v.s[jj];
+            }
+            #pragma unroll
+            for (int it = 0; it < 2; ++it) {
+                const int i = t + it * 256;
+                if (i < dyn8) {
+                    const int p = (i >> 3) * 2;
+                    const int kg = (i & 7) * 8;
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const unsigned int packed =
+                            (unsigned int)(unsigned short)vdy[it][0].s[j]
+                            | ((unsigned int)(unsigned short)
+                               vdy[it][1].s[j] << 16);
+                        *reinterpret_cast<unsigned int*>(
+                            &dy_t[(kg + j) * DY_STRIDE + p]) = packed;
+                    }
+                }
+            }
+            const int xn8 = (P + 2) * W * 4;     // <= 768
+            V16 vx[3][2];
+            #pragma unroll
+            for (int it = 0; it < 3; ++it) {
+                const int i = t + it * 256;
+                if (i < xn8) {
+                    const int j = i / (W * 4);
+                    const int rem = i % (W * 4);
+                    const int w = (rem >> 3) * 2;
+                    const int h = h0 - 1 + j;
+                    if (h < 0 || h >= H) {
+                        vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
+                        vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
+                    } else {
+                        const ushort_t* g = x +
+                            (((size_t)n * H + h) * W + w) * C
+                            + (size_t)ct * 64 + (rem & 7) * 8;
+                        vx[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+                        vx[it][1].u4 =
+                            *reinterpret_cast<const u32x4*>(g + C);
+                    }
+                }
+            }
+            #pragma unroll
+            for (int it = 0; it < 3; ++it) {
+                const int i = t + it * 256;
+                if (i < xn8) {
+                    const int j = i / (W * 4);
+                    const int rem = i % (W * 4);
+                    const int w = (rem >> 3) * 2;
+                    const int cg = (rem & 7) * 8;
+                    #pragma unroll
+                    for (int jj = 0; jj < 8; ++jj) {
+                        const unsigned int packed =
+                            (unsigned int)(unsigned short)vx[it][0].s[jj]
+                            | ((unsigned int)(unsigned short)
+                               vx[it][1].s[jj] << 16);
+                        *reinterpret_cast<unsigned int*>(
+                            &x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w])
+                            = packed;
+                    }
+                }
             }
         }
         __syncthreads();
@@ -295,7 +335,7 @@ extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
     int P = (W == 32) ? 4 : 8;
     long chunks = (long)N * (H / P);
     int tiles = (K / 64) * (C / 64);
-    long target = 640 / tiles;
+    long target = 512 / tiles;
     if (target < 1) target = 1;
     if (target > chunks) target = chunks;
     return (int)target;
