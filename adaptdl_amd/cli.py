@@ -147,7 +147,8 @@ def cmd_daemon(args):
     state_dir = os.path.abspath(args.state_dir)
     os.makedirs(state_dir, exist_ok=True)
     controller = LocalController(num_gpus=args.num_gpus,
-                                 interval=args.interval)
+                                 interval=args.interval,
+                                 metrics_port=args.metrics_port)
     host, port = args.bind.rsplit(":", 1)
     admin = _AdminServer(controller, host, int(port))
     admin.state_dir = state_dir
@@ -261,6 +262,8 @@ def main(argv=None):
     p.add_argument("--state-dir", default=".adaptdl")
     p.add_argument("--num-gpus", type=int, default=None)
     p.add_argument("--interval", type=float, default=30.0)
+    p.add_argument("--metrics-port", type=int, default=None,
+                   help="expose Prometheus metrics on this port")
 
     p = sub.add_parser("run", help="run a job in the foreground")
     add_job_opts(p)

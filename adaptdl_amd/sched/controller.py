@@ -30,6 +30,31 @@ from adaptdl_amd.sched.supervisor import Supervisor
 
 LOG = logging.getLogger(__name__)
 
+try:  # optional Prometheus metrics (reference controller.py:35-41)
+    from prometheus_client import Counter, Gauge, start_http_server
+
+    METRICS = {
+        "submitted": Counter("adaptdl_jobs_submitted",
+                             "Jobs submitted to the local controller"),
+        "succeeded": Counter("adaptdl_jobs_succeeded", "Jobs succeeded"),
+        "failed": Counter("adaptdl_jobs_failed", "Jobs failed"),
+        "preemptions": Counter("adaptdl_job_preemptions",
+                               "Graceful checkpoint-restart preemptions"),
+        "replicas": Gauge("adaptdl_running_replicas",
+                          "Currently running replica processes"),
+    }
+except ImportError:  # pragma: no cover - prometheus_client is optional
+    METRICS = None
+
+    def start_http_server(port):
+        raise RuntimeError("prometheus_client is not installed")
+
+
+def _metric(name, amount=1):
+    if METRICS is not None:
+        METRICS[name].inc(amount)
+
+
 GRACEFUL_EXIT = 143  # SIGTERM-driven checkpoint exit (reference parity)
 
 PENDING = "Pending"
@@ -99,7 +124,9 @@ class LocalController(object):
     """Submit and elastically run training jobs on this node."""
 
     def __init__(self, num_gpus=None, allocator=None, interval=30.0,
-                 poll_interval=0.2):
+                 poll_interval=0.2, metrics_port=None):
+        if metrics_port is not None:
+            start_http_server(metrics_port)
         if num_gpus is None:
             num_gpus = int(os.getenv("ADAPTDL_NUM_GPUS", "8"))
         self.num_gpus = num_gpus
@@ -124,6 +151,7 @@ class LocalController(object):
             os.makedirs(os.path.join(spec.job_dir, "logs"), exist_ok=True)
             job = _Job(spec, time.time())
             self._jobs[spec.name] = job
+        _metric("submitted")
         self._wake.set()
         return spec.name
 
@@ -319,6 +347,8 @@ class LocalController(object):
             job.procs.append(proc)
         self.supervisor.set_endpoints(spec.name, job.num_restarts,
                                       ["127.0.0.1"] * n)
+        if METRICS is not None:
+            METRICS["replicas"].inc(n)
         job.state = RUNNING
         LOG.info("job %s group %d started with %d replicas (gpus=%s)",
                  spec.name, job.num_restarts, n, gpus)
@@ -355,12 +385,15 @@ class LocalController(object):
                 p._adaptdl_log.close()
             except Exception:
                 pass
+        if METRICS is not None:
+            METRICS["replicas"].dec(len(job.procs))
         job.procs = []
         job.gpus = []
         self.supervisor.clear_job(job.spec.name)
         if all(c == 0 for c in codes):
             job.state = SUCCEEDED
             job.completion = time.time()
+            _metric("succeeded")
             LOG.info("job %s succeeded", job.spec.name)
         elif all(c in (0, GRACEFUL_EXIT) for c in codes) or \
                 expect_preemption:
@@ -369,7 +402,8 @@ class LocalController(object):
             if job.target_allocation is not None:
                 job.allocation = list(job.target_allocation)
                 job.target_allocation = None
-            job.state = PENDING if job.allocation else PENDING
+            job.state = PENDING
+            _metric("preemptions")
             LOG.info("job %s preempted (codes=%s); restart %d with %d "
                      "replicas", job.spec.name, codes, job.num_restarts,
                      len(job.allocation))
@@ -378,6 +412,7 @@ class LocalController(object):
             if job.failures > job.spec.restart_limit:
                 job.state = FAILED
                 job.completion = time.time()
+                _metric("failed")
                 LOG.warning("job %s failed (codes=%s)", job.spec.name,
                             codes)
             else:
