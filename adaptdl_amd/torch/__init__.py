@@ -26,6 +26,8 @@ from adaptdl_amd.torch.data import (AdaptiveDataLoader,  # noqa
 from adaptdl_amd.torch.parallel import AdaptiveDataParallel  # noqa
 from adaptdl_amd.torch.accumulator import Accumulator  # noqa
 from adaptdl_amd.torch.iterator import AdaptiveBPTTIterator  # noqa
+from adaptdl_amd.torch.optim import (FusedSGD, FusedAdam,  # noqa
+                                     FusedAdamW)
 
 LOG = logging.getLogger(__name__)
 
@@ -42,6 +44,9 @@ __all__ = [
     "AdaptiveDataParallel",
     "Accumulator",
     "AdaptiveBPTTIterator",
+    "FusedSGD",
+    "FusedAdam",
+    "FusedAdamW",
 ]
 
 

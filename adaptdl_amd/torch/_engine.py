@@ -59,7 +59,8 @@ def _segment_view(seg, p):
 
 
 class Bucket(object):
-    __slots__ = ["group_idx", "flat", "segments", "ready", "work", "prev"]
+    __slots__ = ["group_idx", "flat", "segments", "ready", "work", "prev",
+                 "param_flat", "sgd_momentum", "adam_state"]
 
     def __init__(self, group_idx, flat, segments):
         self.group_idx = group_idx
@@ -68,6 +69,9 @@ class Bucket(object):
         self.ready = 0
         self.work = None          # in-flight all-reduce handle
         self.prev = None          # snapshot buffer (accumulation / GNS diff)
+        self.param_flat = None    # fused-optimizer flat parameter buffer
+        self.sgd_momentum = None  # fused SGD momentum (FusedSGD)
+        self.adam_state = None    # fused Adam moments (FusedAdam/W)
 
     def ensure_prev(self):
         if self.prev is None:

@@ -72,6 +72,10 @@ class AdaptiveDataParallel(torch.nn.Module):
             self.gns = GradientNoiseScale(self, optimizer,
                                           mp_scaler=mp_scaler)
         self.scaling_rule.initialize(self, optimizer, patch_optimizer=True)
+        if hasattr(optimizer, "attach_engine"):
+            # Fused flat-bucket optimizer: flatten parameters into the
+            # engine's bucket layout (one fused kernel per bucket).
+            optimizer.attach_engine(self.gns.engine)
         self.require_backward_grad_sync = True
 
         self._sync_module_states()

@@ -79,8 +79,8 @@ def main():
     model = ResNet18().to(device)
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
-    optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
-                            weight_decay=5e-4)
+    optim = adl.FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=5e-4)
     adp = adl.AdaptiveDataParallel(model, optim)
 
     g = torch.Generator(device="cpu").manual_seed(4321 + rank)

@@ -61,7 +61,7 @@ def main():
               else BertConfig.base())
     torch.manual_seed(21)
     model = BertForMaskedLM(config).to(device)
-    optim = torch.optim.Adam(model.parameters(), lr=args.lr)
+    optim = adl.FusedAdam(model.parameters(), lr=args.lr)
     adp = adl.AdaptiveDataParallel(model, optim)  # -> AdamScale
 
     loader = adl.AdaptiveDataLoader(

@@ -51,8 +51,8 @@ def main():
     model = ResNet18().to(device)
     if use_gpu:
         model = model.to(memory_format=torch.channels_last)
-    optim = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
-                            weight_decay=5e-4)
+    optim = adl.FusedSGD(model.parameters(), lr=args.lr,
+                         momentum=0.9, weight_decay=5e-4)
     sched = torch.optim.lr_scheduler.CosineAnnealingLR(optim, args.epochs)
     adp = adl.AdaptiveDataParallel(model, optim, sched)
 
