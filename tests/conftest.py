@@ -45,6 +45,27 @@ def elastic_multiprocessing(func):
                     # on its first parallel region.  Run single-threaded.
                     import torch
                     torch.set_num_threads(1)
+                    # The child inherits any module-level singletons other
+                    # tests created in the pytest process; reset them so
+                    # State registration/loading behaves like a fresh
+                    # replica process.
+                    import sys as _sys
+                    if "adaptdl_amd.checkpoint" in _sys.modules:
+                        _sys.modules["adaptdl_amd.checkpoint"] \
+                            ._REGISTRY.clear()
+                    if "adaptdl_amd.torch._metrics" in _sys.modules:
+                        _sys.modules["adaptdl_amd.torch._metrics"] \
+                            ._METRICS_STATE = None
+                    if "adaptdl_amd.torch.epoch" in _sys.modules:
+                        _sys.modules["adaptdl_amd.torch.epoch"] \
+                            ._EPOCH_STATE = None
+                    if "adaptdl_amd.torch.data" in _sys.modules:
+                        data_mod = _sys.modules["adaptdl_amd.torch.data"]
+                        data_mod.AdaptiveDataLoaderHelper._current = None
+                        data_mod.AdaptiveDataLoaderHelper._training = None
+                        data_mod.AdaptiveDataLoaderHelper._position.clear()
+                        data_mod._AdaptiveDataLoaderState \
+                            .init_count.clear()
                     os.environ["ADAPTDL_CHECKPOINT_PATH"] = str(tmpdir)
                     os.environ["ADAPTDL_JOB_ID"] = "tmpjob"
                     os.environ["ADAPTDL_MASTER_ADDR"] = "127.0.0.1"
