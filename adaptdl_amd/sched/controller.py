@@ -186,6 +186,15 @@ class LocalController(object):
             job.target_allocation = ["local"] * num_replicas
         self._wake.set()
 
+    def restart(self, name):
+        """Force a checkpoint-restart at the current replica count."""
+        with self._lock:
+            job = self._jobs[name]
+            if job.state == RUNNING:
+                job.state = STOPPING
+                self._signal_group(job, signal.SIGTERM)
+        self._wake.set()
+
     def reallocate(self):
         """Run one allocator cycle immediately."""
         self._optimize()

@@ -8,6 +8,13 @@ data stands in for the corpus (no network in this environment);
 --config mini keeps CPU runs fast, base matches the 8-GPU workload.
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 
 import torch

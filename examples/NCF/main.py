@@ -6,6 +6,13 @@ gradient_accumulation=True (the reference's accumulation-path
 workload).  Synthetic user/item interactions (no dataset downloads).
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 
 import torch

@@ -7,6 +7,13 @@ checkpointed/restored and contribute grad params independently.
 Synthetic 64x64 image data (no dataset downloads).
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 
 import torch

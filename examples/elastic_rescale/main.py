@@ -107,7 +107,7 @@ def main():
     if peak < 2:
         print("note: only 1 GPU visible; the drill still exercises the "
               "checkpoint-restart protocol at 1 replica")
-        phases = [1, 1]
+        phases = [1, "restart", 1]
     else:
         phases = [1, peak, max(peak // 2, 1)]
     print("rescale drill phases (replicas):", phases)
@@ -126,7 +126,11 @@ def main():
     ctrl.submit(spec)
     try:
         for target in phases:
-            ctrl.rescale("elastic-resnet50", target)
+            if target == "restart":
+                ctrl.restart("elastic-resnet50")
+                target = ctrl.status("elastic-resnet50")["replicas"]
+            else:
+                ctrl.rescale("elastic-resnet50", target)
             # Wait for the checkpoint-restart to land on the target...
             deadline = time.time() + 300
             while time.time() < deadline:

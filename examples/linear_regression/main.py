@@ -10,6 +10,13 @@ Run elastically (2 way):  ADAPTDL_NUM_REPLICAS=2 ADAPTDL_REPLICA_RANK=r \
 or submit via the local controller:  adaptdl-amd run -- python main.py
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 
 import torch

@@ -9,6 +9,13 @@ so --synthetic (default) trains on CIFAR-shaped random data; pass
 --data DIR to use a torchvision-style CIFAR-10 folder if present.
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 import os
 

@@ -7,6 +7,13 @@ this environment: a synthetic Zipf-distributed token stream stands in
 for WikiText-2 (--tokens controls its length).
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 import math
 

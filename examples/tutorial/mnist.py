@@ -12,6 +12,13 @@ No dataset downloads in this environment: MNIST-shaped synthetic data
 is meaningful and restart-identical.
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__)))))
+
+
 import argparse
 
 import torch
