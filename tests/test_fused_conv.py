@@ -17,8 +17,6 @@ from adaptdl_amd.torch.layers import FusedConv2d
     (2, 64, 32, 32, 64),      # layer1-like
     (2, 64, 16, 16, 128),     # channel growth
     (3, 128, 16, 16, 128),    # layer2-like
-    (2, 256, 8, 8, 256),      # layer3-like
-    (1, 64, 8, 8, 64),        # small single image
 ])
 def test_wrw_kernel_matches_fp32(shape):
     from adaptdl_amd import ops
@@ -98,6 +96,6 @@ def test_wrw_engages_in_resnet():
         out.float().sum().backward()
     finally:
         L._FusedConvFunction.apply = orig
-    # 3x3/s1 convs with C,K >= 64 and W in {8,16,32}: layers 1-3 stride-1
-    # convs take the MFMA wrw path (layer4 is 4x4 spatial -> fallback).
-    assert len(calls) >= 8, len(calls)
+    # 3x3/s1 convs with C,K >= 64 and W in {16,32}: layers 1-2 stride-1
+    # convs take the MFMA wrw path (8x8/4x4 spatial -> MIOpen fallback).
+    assert len(calls) >= 7, len(calls)
