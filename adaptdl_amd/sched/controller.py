@@ -90,6 +90,15 @@ class JobSpec(object):
     def __init__(self, argv, name, job_dir, min_replicas=0, max_replicas=8,
                  gpus_per_replica=1, env=None, workdir=None,
                  restart_limit=3, preemptible=True):
+        # Admission validation (reference validator.py:70-101 enforces
+        # these via a k8s webhook; locally they are constructor checks).
+        if not argv:
+            raise ValueError("argv must be a non-empty command line")
+        if max_replicas <= 0 or max_replicas < min_replicas:
+            raise ValueError("maxReplicas must be positive and >= "
+                             "minReplicas")
+        if gpus_per_replica < 0:
+            raise ValueError("gpus_per_replica must be >= 0")
         self.argv = list(argv)
         self.name = name
         self.job_dir = job_dir

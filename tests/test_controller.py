@@ -177,3 +177,15 @@ def test_allocator_scales_up_from_hints(tmp_path):
         assert ctrl.wait("auto-job", timeout=120) == "Succeeded"
     finally:
         ctrl.shutdown()
+
+
+def test_jobspec_admission_validation(tmp_path):
+    """Reference validator.py parity: invalid specs rejected at submit."""
+    with pytest.raises(ValueError):
+        JobSpec([], name="bad", job_dir=str(tmp_path))
+    with pytest.raises(ValueError):
+        JobSpec(["x"], name="bad", job_dir=str(tmp_path),
+                min_replicas=4, max_replicas=2)
+    with pytest.raises(ValueError):
+        JobSpec(["x"], name="bad", job_dir=str(tmp_path),
+                max_replicas=0)
