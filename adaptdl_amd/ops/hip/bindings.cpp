@@ -34,6 +34,10 @@ void launch_bn_bwd(const unsigned short*, const unsigned short*,
                    const float*, const float*, int, int, float*, float*,
                    float*, float*, float*, hipStream_t);
 int conv3x3_wrw_supported(int, int, int, int);
+int conv3x3_mm_supported(int, int, int, int);
+void launch_conv3x3_mm(const unsigned short*, const unsigned short*,
+                       unsigned short*, int, int, int, int, int,
+                       hipStream_t);
 int conv3x3_wrw_nsplit(int, int, int, int, int);
 void launch_conv3x3_wrw(const unsigned short*, const unsigned short*,
                         float*, float*, int, int, int, int, int,
@@ -279,6 +283,32 @@ void conv_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor ws,
                        (int)N, (int)H, (int)W, (int)C, (int)K, stream());
 }
 
+bool conv_mm_ok(long N, long H, long W, long C, long K) {
+    return conv3x3_mm_supported((int)H, (int)W, (int)C, (int)K) != 0;
+}
+
+// y = conv3x3_s1p1(x, w): x (N,C,H,W) channels_last bf16; w EITHER the
+// (K,C,3,3) channels_last weight (forward) or the host-prepared
+// (C,K,3,3)-channels_last flipped transform (backward-data with C and K
+// swapped roles by the caller).
+void conv_mm(torch::Tensor x, torch::Tensor w, torch::Tensor y) {
+    check_bn_x(x, "x"); check_bn_x(y, "y");
+    const long N = x.size(0), C = x.size(1), H = x.size(2),
+        Wd = x.size(3);
+    const long K = y.size(1);
+    TORCH_CHECK(y.size(0) == N && y.size(2) == H && y.size(3) == Wd,
+                "y shape mismatch");
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
+                w.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                w.size(0) == K && w.size(1) == C && w.size(2) == 3 &&
+                w.size(3) == 3, "w must be (K,C,3,3) channels_last bf16");
+    TORCH_CHECK(conv_mm_ok(N, H, Wd, C, K), "unsupported conv_mm shape");
+    launch_conv3x3_mm((const unsigned short*)x.data_ptr(),
+                      (const unsigned short*)w.data_ptr(),
+                      (unsigned short*)y.data_ptr(),
+                      (int)N, (int)H, (int)Wd, (int)C, (int)K, stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -297,4 +327,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("conv_wrw_ok", &conv_wrw_ok, "3x3 wrw fast-path predicate");
     mod.def("conv_wrw_nsplit", &conv_wrw_nsplit, "wrw workspace splits");
     mod.def("conv_wrw", &conv_wrw, "MFMA 3x3 s1 NHWC weight gradient");
+    mod.def("conv_mm_ok", &conv_mm_ok, "3x3 fwd/bwd-data predicate");
+    mod.def("conv_mm", &conv_mm, "MFMA 3x3 s1 NHWC conv (fwd/bwd-data)");
 }

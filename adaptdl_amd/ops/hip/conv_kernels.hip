@@ -35,6 +35,11 @@
 
 #define WAVE 64
 
+__device__ __forceinline__ unsigned short f2b(float f) {
+    __hip_bfloat16 h = __float2bfloat16(f);  // RNE
+    return *reinterpret_cast<unsigned short*>(&h);
+}
+
 typedef unsigned short ushort_t;
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -361,4 +366,205 @@ extern "C" void launch_conv3x3_wrw(
     if (blocks > 4096) blocks = 4096;
     hipLaunchKernelGGL(k_wrw_reduce, dim3((unsigned)blocks), dim3(256), 0,
                        s, ws, dw, n, nsplit, ngroups, nchunks);
+}
+
+// =====================================================================
+// MFMA 3x3 stride-1 NHWC convolution: forward AND backward-data.
+//
+//   y[p][ko] = sum_{tau, ci} x[p + off(tau)][ci] * w[ko][tau][ci]
+//
+// Backward-data is the same contraction with x := dy, w := the
+// flipped/transposed weight ([C][tau'][K], prepared host-side), so one
+// kernel serves both directions.  Unlike the wrw kernel, both operands
+// stage in their NATURAL layouts (zero transpose cost):
+//   - x_s: pixel-major padded window [q][C+pad]; b128 copies, the MFMA
+//     A-fragment (16 pixels x 32 c-slots) reads 16B rows directly, with
+//     the 9 taps being pure row/column offsets into the padded window,
+//   - w_lds: [ko][tau][ci+pad], LDS-resident for the whole block
+//     (staged once, reused across the block's chunk loop),
+//   - D tile = y pixels x ko, written straight out (no split-K, no
+//     workspace: the contraction (9*C <= 1152) fits one block).
+// Dynamic LDS (up to ~128 KB) sized host-side.
+// Constraints: 3x3 s1 p1, NHWC bf16, W in {16, 32}, H % P == 0,
+// C % 32 == 0, C <= 128, K % KT == 0 (KT = 64 for C<=64 else 32).
+// =====================================================================
+
+extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
+        ushort_t* __restrict__ y, int N, int H, int W, int C, int K,
+        int P, int KT, int spread) {
+    extern __shared__ short lds[];
+    const int CS = C + 8;            // padded channel stride (16B mult)
+    const int LP = W + 2;            // padded line width (pixels)
+    const int QR = (P + 2) * LP;     // x window rows
+    short* w_lds = lds;              // [KT * 9][CS]
+    short* x_s = lds + KT * 9 * CS;  // [QR][CS]
+
+    const int CP = P * W;            // chunk pixels (= 128)
+    const int nk = K / KT;
+    const int kt = blockIdx.x / spread;
+    const int sp = blockIdx.x % spread;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wp = wid >> 1;         // pixel half (0..1): 64 pixels
+    const int wk = wid & 1;          // ko half (0..1): KT/2 channels
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+    const int KH = KT / 2;           // per-wave ko span (32 or 16)
+    const int NF = KH / 16;          // ko fragments per wave (2 or 1)
+
+    // ---- stage the weight tile once per block ----
+    {
+        const int pieces = KT * 9 * (C / 8);
+        const ushort_t* wg = w + (size_t)kt * KT * 9 * C;
+        for (int i = t; i < pieces; i += 256) {
+            const int row = i / (C / 8);        // ko * 9 + tau
+            const int cg = (i % (C / 8)) * 8;
+            V16 v;
+            v.u4 = *reinterpret_cast<const u32x4*>(
+                wg + (size_t)row * C + cg);
+            *reinterpret_cast<u32x4*>(&w_lds[row * CS + cg]) = v.u4;
+        }
+    }
+    // ---- zero the x pad columns once (cols 0 and W+1 of every line) --
+    for (int i = t; i < (P + 2) * (C / 8) * 2; i += 256) {
+        const int j = i / ((C / 8) * 2);
+        const int rem = i % ((C / 8) * 2);
+        const int col = (rem & 1) ? (W + 1) : 0;
+        const int cg = (rem >> 1) * 8;
+        u32x4 z = {0u, 0u, 0u, 0u};
+        *reinterpret_cast<u32x4*>(
+            &x_s[(j * LP + col) * CS + cg]) = z;
+    }
+    __syncthreads();
+
+    const int lines_per_img = H / P;
+    const long chunks_total = (long)N * lines_per_img;
+
+    for (long q = sp; q < chunks_total; q += spread) {
+        const int n = (int)(q / lines_per_img);
+        const int h0 = (int)(q % lines_per_img) * P;
+
+        __syncthreads();
+        // ---- stage the x window (pixel-major, natural layout) ----
+        {
+            const int pieces = (P + 2) * W * (C / 8);
+            for (int i = t; i < pieces; i += 256) {
+                const int j = i / (W * (C / 8));    // window line
+                const int rem = i % (W * (C / 8));
+                const int ww = rem / (C / 8);
+                const int cg = (rem % (C / 8)) * 8;
+                const int h = h0 - 1 + j;
+                V16 v;
+                if (h < 0 || h >= H) {
+                    v.u4 = u32x4{0u, 0u, 0u, 0u};
+                } else {
+                    v.u4 = *reinterpret_cast<const u32x4*>(
+                        x + (((size_t)n * H + h) * W + ww) * C + cg);
+                }
+                *reinterpret_cast<u32x4*>(
+                    &x_s[(j * LP + 1 + ww) * CS + cg]) = v.u4;
+            }
+        }
+        __syncthreads();
+
+        // ---- contraction: 9 taps x C/32 channel chunks ----
+        f32x4 acc[4][2];
+        #pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[mf][nf][r] = 0.f;
+
+        for (int tau = 0; tau < 9; ++tau) {
+            const int dh = tau / 3;
+            const int dw = tau % 3;
+            for (int cc = 0; cc < C / 32; ++cc) {
+                bf16x8 bfrag[2];
+                #pragma unroll
+                for (int nf = 0; nf < 2; ++nf) {
+                    if (nf < NF) {
+                        const int ko = wk * KH + nf * 16 + row16;
+                        bfrag[nf] = *reinterpret_cast<const bf16x8*>(
+                            &w_lds[(ko * 9 + tau) * CS + cc * 32
+                                   + slot8]);
+                    }
+                }
+                #pragma unroll
+                for (int mf = 0; mf < 4; ++mf) {
+                    const int p = wp * 64 + mf * 16 + row16;
+                    const int li = p / W;
+                    const int ww = p % W;
+                    const int qrow = (li + dh) * LP + ww + dw;
+                    const bf16x8 afrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &x_s[qrow * CS + cc * 32 + slot8]);
+                    #pragma unroll
+                    for (int nf = 0; nf < 2; ++nf)
+                        if (nf < NF)
+                            acc[mf][nf] =
+                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                    afrag, bfrag[nf], acc[mf][nf],
+                                    0, 0, 0);
+                }
+            }
+        }
+
+        // ---- write the y tile ----
+        const size_t ybase = ((size_t)n * H + h0) * W;  // chunk pixel 0
+        #pragma unroll
+        for (int mf = 0; mf < 4; ++mf) {
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf) {
+                if (nf >= NF) continue;
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int p = wp * 64 + mf * 16 + (lane >> 4) * 4
+                        + r;
+                    const int ko = kt * KT + wk * KH + nf * 16
+                        + (lane & 15);
+                    y[(ybase + p) * K + ko] = f2b(acc[mf][nf][r]);
+                }
+            }
+        }
+    }
+}
+
+extern "C" int conv3x3_mm_supported(int H, int W, int C, int K) {
+    if (C % 32 || C > 128) return 0;
+    if (W != 16 && W != 32) return 0;
+    int P = (W == 32) ? 4 : 8;
+    if (H % P) return 0;
+    int KT = (C <= 64) ? 64 : 32;
+    if (K % KT) return 0;
+    return 1;
+}
+
+extern "C" void launch_conv3x3_mm(
+        const ushort_t* x, const ushort_t* w, ushort_t* y,
+        int N, int H, int W, int C, int K, hipStream_t s) {
+    const int P = (W == 32) ? 4 : 8;
+    const int KT = (C <= 64) ? 64 : 32;
+    const int CS = C + 8;
+    const int QR = (P + 2) * (W + 2);
+    const size_t lds_bytes = ((size_t)KT * 9 + QR) * CS * 2;
+    static int attr_set = 0;
+    if (!attr_set) {
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_mm),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = 1;
+    }
+    const long chunks = (long)N * (H / P);
+    const int nk = K / KT;
+    long spread = 2048 / nk;
+    if (spread > chunks) spread = chunks;
+    if (spread < 1) spread = 1;
+    hipLaunchKernelGGL(k_conv3x3_mm, dim3((unsigned)(nk * spread)),
+                       dim3(256), lds_bytes, s, x, w, y, N, H, W, C, K,
+                       P, KT, (int)spread);
 }

@@ -133,10 +133,18 @@ class _FusedConvFunction(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, stride, padding, dilation, groups):
+        ext = ops._load_extension()
         wb = weight.detach().to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
-        y = torch.nn.functional.conv2d(x, wb, None, stride, padding,
-                                       dilation, groups)
+        n, c, h, w = x.shape
+        k = wb.shape[0]
+        if ext.conv_mm_ok(n, h, w, c, k):
+            y = torch.empty(n, k, h, w, dtype=x.dtype, device=x.device) \
+                .contiguous(memory_format=torch.channels_last)
+            ext.conv_mm(x, wb, y)
+        else:
+            y = torch.nn.functional.conv2d(x, wb, None, stride, padding,
+                                           dilation, groups)
         ctx.save_for_backward(x, wb)
         ctx.conv_args = (stride, padding, dilation, groups)
         return y
@@ -149,9 +157,18 @@ class _FusedConvFunction(torch.autograd.Function):
         dy = dy.contiguous(memory_format=torch.channels_last)
         dx = None
         if ctx.needs_input_grad[0]:
-            dx = torch.ops.aten.convolution_backward(
-                dy, x, wb, None, stride, padding, dilation, False,
-                [0, 0], groups, [True, False, False])[0]
+            n, c, h, w = x.shape
+            k = wb.shape[0]
+            if ext.conv_mm_ok(n, h, w, k, c):
+                # dx = conv(dy, W flipped with C/K roles swapped)
+                wt = wb.flip(2, 3).permute(1, 0, 2, 3) \
+                    .contiguous(memory_format=torch.channels_last)
+                dx = torch.empty_like(x)
+                ext.conv_mm(dy, wt, dx)
+            else:
+                dx = torch.ops.aten.convolution_backward(
+                    dy, x, wb, None, stride, padding, dilation, False,
+                    [0, 0], groups, [True, False, False])[0]
         n, c, h, w = x.shape
         k = dy.shape[1]
         nsplit = ext.conv_wrw_nsplit(n, h, w, c, k)

@@ -99,3 +99,52 @@ def test_wrw_engages_in_resnet():
     # 3x3/s1 convs with C,K >= 64 and W in {16,32}: layers 1-2 stride-1
     # convs take the MFMA wrw path (8x8/4x4 spatial -> MIOpen fallback).
     assert len(calls) >= 7, len(calls)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 32, 32, 64),
+    (2, 64, 16, 16, 128),
+    (3, 128, 16, 16, 128),
+])
+def test_conv_mm_forward_matches_fp32(shape):
+    from adaptdl_amd import ops
+    ext = ops._load_extension()
+    torch.manual_seed(8)
+    n, c, h, w, k = shape
+    dev = torch.device("cuda")
+    x = (torch.randn(n, c, h, w, device=dev) * 0.5).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    wt = (torch.randn(k, c, 3, 3, device=dev) * 0.1).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    assert ext.conv_mm_ok(n, h, w, c, k)
+    y = torch.empty(n, k, h, w, dtype=torch.bfloat16, device=dev) \
+        .contiguous(memory_format=torch.channels_last)
+    ext.conv_mm(x, wt, y)
+    ref = F.conv2d(x.float(), wt.float(), padding=1)
+    assert torch.allclose(y.float(), ref, atol=0.15, rtol=5e-2), \
+        (y.float() - ref).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_conv_mm_full_autograd_roundtrip():
+    """FusedConv2d with the custom fwd + bwd-data + wrw kernels."""
+    torch.manual_seed(9)
+    dev = torch.device("cuda")
+    conv = FusedConv2d(64, 64, 3, padding=1, bias=False).to(dev)
+    ref = torch.nn.Conv2d(64, 64, 3, padding=1, bias=False).to(dev)
+    with torch.no_grad():
+        ref.weight.copy_(conv.weight)
+    x = torch.randn(4, 64, 32, 32, device=dev).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x32 = x.detach().float().requires_grad_(True)
+    y = conv(x)
+    yr = ref(x32)
+    assert torch.allclose(y.float(), yr, atol=0.2, rtol=5e-2)
+    dy = torch.randn_like(yr)
+    y.backward(dy.to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last))
+    yr.backward(dy)
+    assert torch.allclose(x.grad.float(), x32.grad, atol=0.3, rtol=5e-2)
+    assert torch.allclose(conv.weight.grad, ref.weight.grad,
+                          atol=0.3, rtol=5e-2)
