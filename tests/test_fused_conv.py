@@ -64,10 +64,9 @@ def test_fused_conv2d_module_full_backward():
     yr = ref(x32)
     assert torch.allclose(y.float(), yr, atol=0.2, rtol=5e-2)
 
-    dy = torch.randn_like(yr)
-    y.backward(dy.to(torch.bfloat16).contiguous(
-        memory_format=torch.channels_last))
-    yr.backward(dy)
+    dy = torch.randn_like(yr).to(torch.bfloat16)
+    y.backward(dy.contiguous(memory_format=torch.channels_last))
+    yr.backward(dy.float())
     assert conv.weight.grad is not None
     assert conv.weight.grad.dtype == torch.float32
     assert torch.allclose(conv.weight.grad, ref.weight.grad,
@@ -127,8 +126,10 @@ def test_conv_mm_forward_matches_fp32(shape):
 
 
 @pytest.mark.gpu
-def test_conv_mm_full_autograd_roundtrip():
-    """FusedConv2d with the custom fwd + bwd-data + wrw kernels."""
+def test_conv_mm_full_autograd_roundtrip(monkeypatch):
+    """FusedConv2d with the custom fwd + bwd-data + wrw kernels
+    (experimental path, enabled explicitly)."""
+    monkeypatch.setenv("ADAPTDL_EXPERIMENTAL_CONV_MM", "1")
     torch.manual_seed(9)
     dev = torch.device("cuda")
     conv = FusedConv2d(64, 64, 3, padding=1, bias=False).to(dev)
@@ -141,10 +142,11 @@ def test_conv_mm_full_autograd_roundtrip():
     y = conv(x)
     yr = ref(x32)
     assert torch.allclose(y.float(), yr, atol=0.2, rtol=5e-2)
-    dy = torch.randn_like(yr)
-    y.backward(dy.to(torch.bfloat16).contiguous(
-        memory_format=torch.channels_last))
-    yr.backward(dy)
+    # use the SAME bf16-rounded dy on both sides (weight gradients sum
+    # ~4k terms; an asymmetric fp32-vs-bf16 dy alone shifts them ~1.7%)
+    dy = torch.randn_like(yr).to(torch.bfloat16)
+    y.backward(dy.contiguous(memory_format=torch.channels_last))
+    yr.backward(dy.float())
     assert torch.allclose(x.grad.float(), x32.grad, atol=0.3, rtol=5e-2)
     assert torch.allclose(conv.weight.grad, ref.weight.grad,
                           atol=0.3, rtol=5e-2)
