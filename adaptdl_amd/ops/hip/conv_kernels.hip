@@ -56,19 +56,205 @@ union V16 {
     short s[8];
 };
 
-extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
+// Software-pipelined (T14): two LDS buffer sets; the global loads for
+// chunk i+1 are issued BEFORE chunk i's MFMA phase and land in LDS
+// after it, so HBM latency hides under compute.  v3 measured wall time
+// ~5x the per-SIMD busy cycles at 1 block/CU -- pure staging latency.
+
+struct WrwRegs {
+    V16 vdy[1][2];
+    V16 vx[2][2];
+};
+
+__device__ __forceinline__ void wrw_issue(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+        int n, int h0, int H, int W, int C, int K, int kt, int ct,
+        int CP, int P, int t, WrwRegs& r) {
+#ifdef WRW_PROBE_NOLOAD
+    (void)x; (void)dy; (void)n; (void)h0;
+    return;
+#endif
+    const int dyn8 = CP * 4;
+    #pragma unroll
+    for (int it = 0; it < 1; ++it) {
+        const int i = t + it * 512;
+        if (i < dyn8) {
+            const int p = (i >> 3) * 2;
+            const ushort_t* g = dy +
+                (((size_t)n * H + (h0 + p / W)) * W + p % W) * K
+                + (size_t)kt * 64 + (i & 7) * 8;
+            r.vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+            r.vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
+        }
+    }
+    const int xn8 = (P + 2) * W * 4;
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const int i = t + it * 512;
+        if (i < xn8) {
+            const int j = i / (W * 4);
+            const int rem = i % (W * 4);
+            const int w = (rem >> 3) * 2;
+            const int h = h0 - 1 + j;
+            if (h < 0 || h >= H) {
+                r.vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
+                r.vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                const ushort_t* g = x +
+                    (((size_t)n * H + h) * W + w) * C
+                    + (size_t)ct * 64 + (rem & 7) * 8;
+                r.vx[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+                r.vx[it][1].u4 = *reinterpret_cast<const u32x4*>(g + C);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void wrw_write(
+        short* __restrict__ dy_t, short* __restrict__ x_t,
+        int W, int CP, int P, int LS, int t, WrwRegs& r) {
+    const int dyn8 = CP * 4;
+    #pragma unroll
+    for (int it = 0; it < 1; ++it) {
+        const int i = t + it * 512;
+        if (i < dyn8) {
+            const int p = (i >> 3) * 2;
+            const int kg = (i & 7) * 8;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const unsigned int packed =
+                    (unsigned int)(unsigned short)r.vdy[it][0].s[j]
+                    | ((unsigned int)(unsigned short)
+                       r.vdy[it][1].s[j] << 16);
+                *reinterpret_cast<unsigned int*>(
+                    &dy_t[(kg + j) * DY_STRIDE + p]) = packed;
+            }
+        }
+    }
+    const int xn8 = (P + 2) * W * 4;
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const int i = t + it * 512;
+        if (i < xn8) {
+            const int j = i / (W * 4);
+            const int rem = i % (W * 4);
+            const int w = (rem >> 3) * 2;
+            const int cg = (rem & 7) * 8;
+            #pragma unroll
+            for (int jj = 0; jj < 8; ++jj) {
+                const unsigned int packed =
+                    (unsigned int)(unsigned short)r.vx[it][0].s[jj]
+                    | ((unsigned int)(unsigned short)
+                       r.vx[it][1].s[jj] << 16);
+                *reinterpret_cast<unsigned int*>(
+                    &x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w])
+                    = packed;
+            }
+        }
+    }
+}
+
+#define WRW_BUF (64 * DY_STRIDE + 64 * XT_STRIDE)
+
+// Compile-time tau subsets per wave pair: WT2=0 -> taus 0..4 (groups
+// (dh0, dw0..2), (dh1, dw0..1)); WT2=1 -> taus 5..8 ((dh1, dw2),
+// (dh2, dw0..2)).  Everything constant-folds so acc stays in registers.
+template <int WT2>
+struct WrwTaus {
+    static constexpr int NG = 2;
+    static constexpr int DH[2] = {WT2 ? 1 : 0, WT2 ? 2 : 1};
+    static constexpr int LO[2] = {WT2 ? 2 : 0, 0};
+    static constexpr int HI[2] = {3, WT2 ? 3 : 2};
+};
+
+template <int WT2>
+__device__ __forceinline__ void wrw_mfma_phase(
+        const short* __restrict__ dy_t, const short* __restrict__ x_t,
+        int kchunks, int W, int LS, int wk, int wc, int row16, int slot8,
+        f32x4 (&acc)[2][2][5]) {
+    for (int kc = 0; kc < kchunks; ++kc) {
+        const int p0 = kc * 32 + slot8;
+        const int li = p0 / W;
+        const int w0 = p0 % W;
+        bf16x8 afrag[2];
+        #pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+            const short* a = &dy_t[(wk * 32 + mf * 16 + row16)
+                                   * DY_STRIDE + kc * 32 + slot8];
+            afrag[mf] = *reinterpret_cast<const bf16x8*>(a);
+        }
+        int ti = 0;
+        #pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            constexpr int dh0 = WT2 ? 1 : 0;
+            constexpr int dh1 = WT2 ? 2 : 1;
+            const int dh = g ? dh1 : dh0;
+            constexpr int lo0 = WT2 ? 2 : 0;
+            constexpr int hi1 = WT2 ? 3 : 2;
+            const int lo = g ? 0 : lo0;
+            const int hi = g ? hi1 : 3;
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf) {
+                const short* b = &x_t[(wc * 32 + nf * 16 + row16)
+                                      * XT_STRIDE + (li + dh) * LS + w0];
+                V16 vlo, vhi;
+                vlo.u4 = *reinterpret_cast<const u32x4*>(b);
+                vhi.u4 = *reinterpret_cast<const u32x4*>(b + 8);
+                unsigned int win[8] = {vlo.u[0], vlo.u[1], vlo.u[2],
+                                       vlo.u[3], vhi.u[0], vhi.u[1],
+                                       vhi.u[2], vhi.u[3]};
+                int tj = ti;
+                #pragma unroll
+                for (int dw = 0; dw < 3; ++dw) {
+                    if (dw < lo || dw >= hi) continue;
+                    u32x4 frag;
+                    const int sh = (3 + dw) * 2;
+                    const int d0 = sh >> 2;
+                    const int rem = sh & 3;
+                    if (rem == 0) {
+                        frag[0] = win[d0]; frag[1] = win[d0 + 1];
+                        frag[2] = win[d0 + 2]; frag[3] = win[d0 + 3];
+                    } else {
+                        frag[0] = __builtin_amdgcn_alignbyte(
+                            win[d0 + 1], win[d0], rem);
+                        frag[1] = __builtin_amdgcn_alignbyte(
+                            win[d0 + 2], win[d0 + 1], rem);
+                        frag[2] = __builtin_amdgcn_alignbyte(
+                            win[d0 + 3], win[d0 + 2], rem);
+                        frag[3] = __builtin_amdgcn_alignbyte(
+                            win[d0 + 4], win[d0 + 3], rem);
+                    }
+                    const bf16x8 bfrag =
+                        *reinterpret_cast<const bf16x8*>(&frag);
+                    #pragma unroll
+                    for (int mf = 0; mf < 2; ++mf)
+                        acc[mf][nf][tj] =
+                            __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                afrag[mf], bfrag, acc[mf][nf][tj],
+                                0, 0, 0);
+                    ++tj;
+                }
+            }
+            ti += (g ? hi1 : 3) - (g ? 0 : lo0);
+        }
+    }
+}
+
+// 8 waves: (wk, wc) tile quadrant x 2-wave tau split -- two waves per
+// SIMD interleave the alignbyte/window VALU work under each other's
+// MFMAs (phase probe: MFMA+VALU phase was ~150 us of the 218 us call
+// at 1 wave/SIMD; staging only ~35 us).
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
         float* __restrict__ ws, int N, int H, int W, int C, int K,
         int P, int nsplit) {
-    __shared__ short dy_t[64 * DY_STRIDE];
-    __shared__ short x_t[64 * XT_STRIDE];
+    __shared__ short lds[2 * WRW_BUF];
 
     const int LS = W + 8;            // padded line stride (mult of 8)
     const int CP = P * W;            // chunk pixels (mult of 32)
     const int kchunks = CP / 32;
     const int nc = C / 64;
     const int nsplit_t = nsplit;
-    // block id -> (ktile, ctile, split)
     const int tile = blockIdx.x / nsplit_t;
     const int split = blockIdx.x % nsplit_t;
     const int kt = tile / nc;
@@ -77,210 +263,102 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_wrw(
     const int t = threadIdx.x;
     const int lane = t & 63;
     const int wid = t >> 6;
-    const int wk = wid >> 1;         // wave k-subtile (0..1)
-    const int wc = wid & 1;          // wave c-subtile (0..1)
+    const int wk = wid >> 2;
+    const int wc = (wid >> 1) & 1;
+    const int wt2 = wid & 1;         // tau half: 0 -> taus 0..4, 1 -> 5..8
     const int row16 = lane & 15;
     const int slot8 = (lane >> 4) * 8;
 
-    // 36 accumulator fragments: [mf][nf][tau]
-    f32x4 acc[2][2][9];
+    f32x4 acc[2][2][5];
     #pragma unroll
     for (int mf = 0; mf < 2; ++mf)
         #pragma unroll
         for (int nf = 0; nf < 2; ++nf)
             #pragma unroll
-            for (int tau = 0; tau < 9; ++tau)
+            for (int ti = 0; ti < 5; ++ti)
                 #pragma unroll
                 for (int r = 0; r < 4; ++r)
-                    acc[mf][nf][tau][r] = 0.f;
+                    acc[mf][nf][ti][r] = 0.f;
 
     const int lines_per_img = H / P;
     const long chunks_total = (long)N * lines_per_img;
 
-    // One-time zeroing of the left/right pad columns (data writes never
-    // touch them; out-of-range halo lines are written as zeros in the
-    // staging loop itself).
-    for (int i = t; i < (P + 2) * 8 * 64; i += 256) {
-        const int cc = i / ((P + 2) * 8);
-        const int rem = i % ((P + 2) * 8);
-        const int j = rem / 8;
-        const int pp = rem % 8;
-        const int q = pp < 4 ? pp : (4 + W + (pp - 4));
-        x_t[cc * XT_STRIDE + j * LS + q] = 0;
+    // Zero the pad columns of BOTH x_t buffers once.
+    for (int b = 0; b < 2; ++b) {
+        short* x_t = lds + b * WRW_BUF + 64 * DY_STRIDE;
+        for (int i = t; i < (P + 2) * 8 * 64; i += 512) {
+            const int cc = i / ((P + 2) * 8);
+            const int rem = i % ((P + 2) * 8);
+            const int j = rem / 8;
+            const int pp = rem % 8;
+            const int q = pp < 4 ? pp : (4 + W + (pp - 4));
+            x_t[cc * XT_STRIDE + j * LS + q] = 0;
+        }
     }
 
-    for (long q = split; q < chunks_total; q += nsplit_t) {
-        const int n = (int)(q / lines_per_img);
-        const int h0 = (int)(q % lines_per_img) * P;
+    WrwRegs regs;
+    #define WRW_NH(qq) \
+        const int n_ = (int)((qq) / lines_per_img); \
+        const int h0_ = (int)((qq) % lines_per_img) * P;
 
-        __syncthreads();  // previous iteration's reads done
+    // Prologue: stage chunk 0 into buffer 0, issue chunk 1's loads.
+    if (split < chunks_total) {
+        WRW_NH(split)
+        wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t, regs);
+        wrw_write(lds, lds + 64 * DY_STRIDE, W, CP, P, LS, t, regs);
+    }
+    __syncthreads();
+    if (split + nsplit_t < chunks_total) {
+        WRW_NH(split + nsplit_t)
+        wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t, regs);
+    }
 
-        // ---- stage dy and x transposed.  Global loads first (one HBM
-        // latency per operand batch), then LDS writes as PACKED b32
-        // pixel-pairs: 2-byte scatter writes measured 44M bank-conflict
-        // cycles per dispatch (SQ_LDS_BANK_CONFLICT); pairing pixels
-        // (p, p+1) into one dword write quarters the conflict cost. ----
-        {
-            const int dyn8 = CP * 4;             // pair-loads: <=512
-            V16 vdy[2][2];
-            #pragma unroll
-            for (int it = 0; it < 2; ++it) {
-                const int i = t + it * 256;
-                if (i < dyn8) {
-                    const int p = (i >> 3) * 2;
-                    const ushort_t* g = dy +
-                        (((size_t)n * H + (h0 + p / W)) * W + p % W) * K
-                        + (size_t)kt * 64 + (i & 7) * 8;
-                    vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
-                    vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
-                }
-            }
-            #pragma unroll
-            for (int it = 0; it < 2; ++it) {
-                const int i = t + it * 256;
-                if (i < dyn8) {
-                    const int p = (i >> 3) * 2;
-                    const int kg = (i & 7) * 8;
-                    #pragma unroll
-                    for (int j = 0; j < 8; ++j) {
-                        const unsigned int packed =
-                            (unsigned int)(unsigned short)vdy[it][0].s[j]
-                            | ((unsigned int)(unsigned short)
-                               vdy[it][1].s[j] << 16);
-                        *reinterpret_cast<unsigned int*>(
-                            &dy_t[(kg + j) * DY_STRIDE + p]) = packed;
-                    }
-                }
-            }
-            const int xn8 = (P + 2) * W * 4;     // <= 768
-            V16 vx[3][2];
-            #pragma unroll
-            for (int it = 0; it < 3; ++it) {
-                const int i = t + it * 256;
-                if (i < xn8) {
-                    const int j = i / (W * 4);
-                    const int rem = i % (W * 4);
-                    const int w = (rem >> 3) * 2;
-                    const int h = h0 - 1 + j;
-                    if (h < 0 || h >= H) {
-                        vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
-                        vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
-                    } else {
-                        const ushort_t* g = x +
-                            (((size_t)n * H + h) * W + w) * C
-                            + (size_t)ct * 64 + (rem & 7) * 8;
-                        vx[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
-                        vx[it][1].u4 =
-                            *reinterpret_cast<const u32x4*>(g + C);
-                    }
-                }
-            }
-            #pragma unroll
-            for (int it = 0; it < 3; ++it) {
-                const int i = t + it * 256;
-                if (i < xn8) {
-                    const int j = i / (W * 4);
-                    const int rem = i % (W * 4);
-                    const int w = (rem >> 3) * 2;
-                    const int cg = (rem & 7) * 8;
-                    #pragma unroll
-                    for (int jj = 0; jj < 8; ++jj) {
-                        const unsigned int packed =
-                            (unsigned int)(unsigned short)vx[it][0].s[jj]
-                            | ((unsigned int)(unsigned short)
-                               vx[it][1].s[jj] << 16);
-                        *reinterpret_cast<unsigned int*>(
-                            &x_t[(cg + jj) * XT_STRIDE + j * LS + 4 + w])
-                            = packed;
-                    }
-                }
+    int cur = 0;
+    for (long q = split; q < chunks_total; q += nsplit_t, cur ^= 1) {
+        const short* dy_t = lds + cur * WRW_BUF;
+        const short* x_t = dy_t + 64 * DY_STRIDE;
+
+        // ---- MFMA phase (chunk q, buffer cur) ----
+        if (wt2 == 0)
+            wrw_mfma_phase<0>(dy_t, x_t, kchunks, W, LS, wk, wc, row16,
+                              slot8, acc);
+        else
+            wrw_mfma_phase<1>(dy_t, x_t, kchunks, W, LS, wk, wc, row16,
+                              slot8, acc);
+
+        // ---- stage chunk q+1 into the other buffer; issue q+2 ----
+        if (q + nsplit_t < chunks_total) {
+            short* ndy = lds + (cur ^ 1) * WRW_BUF;
+            wrw_write(ndy, ndy + 64 * DY_STRIDE, W, CP, P, LS, t, regs);
+            if (q + 2 * nsplit_t < chunks_total) {
+                WRW_NH(q + 2 * nsplit_t)
+                wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t,
+                          regs);
             }
         }
         __syncthreads();
-
-        // ---- MFMA over the chunk's 32-pixel sub-chunks ----
-        for (int kc = 0; kc < kchunks; ++kc) {
-            const int p0 = kc * 32 + slot8;      // this lane's pixels
-            const int li = p0 / W;               // chunk line index
-            const int w0 = p0 % W;               // multiple of 8
-            // A fragments (dy): 16B each
-            bf16x8 afrag[2];
-            #pragma unroll
-            for (int mf = 0; mf < 2; ++mf) {
-                const short* a = &dy_t[(wk * 32 + mf * 16 + row16)
-                                       * DY_STRIDE + kc * 32 + slot8];
-                afrag[mf] = *reinterpret_cast<const bf16x8*>(a);
-            }
-            #pragma unroll
-            for (int dh = 0; dh < 3; ++dh) {
-                // 32-byte aligned window [w0-1-... ] per c-fragment:
-                // window pixels (li+dh)*LS + 4 + w0 - 1 - 3  ... the
-                // load base is 16B-aligned because (4 + w0 - 4) = w0.
-                #pragma unroll
-                for (int nf = 0; nf < 2; ++nf) {
-                    const short* b = &x_t[(wc * 32 + nf * 16 + row16)
-                                          * XT_STRIDE
-                                          + (li + dh) * LS + w0];
-                    // pixels [w-4 .. w+12) of the padded line (pad=4)
-                    V16 lo, hi;
-                    lo.u4 = *reinterpret_cast<const u32x4*>(b);
-                    hi.u4 = *reinterpret_cast<const u32x4*>(b + 8);
-                    unsigned int win[8] = {lo.u[0], lo.u[1], lo.u[2],
-                                           lo.u[3], hi.u[0], hi.u[1],
-                                           hi.u[2], hi.u[3]};
-                    // shifted vectors: tap pixel = w + dw, base w-1
-                    // window starts at pad+w0-4+... byte of pixel
-                    // (w0 + dw - 1 + 4 - w0) = 3+dw shorts into window
-                    #pragma unroll
-                    for (int dw = 0; dw < 3; ++dw) {
-                        u32x4 frag;
-                        const int sh = (3 + dw) * 2;  // byte offset 6,8,10
-                        const int d0 = sh >> 2;       // 1, 2, 2
-                        const int rem = sh & 3;       // 2, 0, 2
-                        if (rem == 0) {
-                            frag[0] = win[d0]; frag[1] = win[d0 + 1];
-                            frag[2] = win[d0 + 2]; frag[3] = win[d0 + 3];
-                        } else {
-                            frag[0] = __builtin_amdgcn_alignbyte(
-                                win[d0 + 1], win[d0], rem);
-                            frag[1] = __builtin_amdgcn_alignbyte(
-                                win[d0 + 2], win[d0 + 1], rem);
-                            frag[2] = __builtin_amdgcn_alignbyte(
-                                win[d0 + 3], win[d0 + 2], rem);
-                            frag[3] = __builtin_amdgcn_alignbyte(
-                                win[d0 + 4], win[d0 + 3], rem);
-                        }
-                        const bf16x8 bfrag =
-                            *reinterpret_cast<const bf16x8*>(&frag);
-                        const int tau = dh * 3 + dw;
-                        #pragma unroll
-                        for (int mf = 0; mf < 2; ++mf)
-                            acc[mf][nf][tau] =
-                                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    afrag[mf], bfrag, acc[mf][nf][tau],
-                                    0, 0, 0);
-                    }
-                }
-            }
-        }
     }
+    #undef WRW_NH
 
-    // ---- epilogue: write the fp32 partial slab ----
-    // D mapping: lane l reg r -> m = (l>>4)*4 + r, n = l&15.
+    // ---- epilogue: write the fp32 partial slab (this wave's taus) ---
     float* slab = ws + (size_t)split * K * 9 * C;
     #pragma unroll
-    for (int mf = 0; mf < 2; ++mf) {
+    for (int ti = 0; ti < 5; ++ti) {
+        if (wt2 && ti >= 4)
+            continue;
+        const int tau = (wt2 ? 5 : 0) + ti;
         #pragma unroll
-        for (int nf = 0; nf < 2; ++nf) {
+        for (int mf = 0; mf < 2; ++mf) {
             #pragma unroll
-            for (int tau = 0; tau < 9; ++tau) {
+            for (int nf = 0; nf < 2; ++nf) {
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int klocal = wk * 32 + mf * 16
                         + ((lane >> 4) * 4 + r);
-                    const int clocal = wc * 32 + nf * 16 + (lane & 15);
+                    const int clocal = wc * 32 + nf * 16
+                        + (lane & 15);
                     slab[((size_t)(kt * 64 + klocal) * 9 + tau) * C
-                         + ct * 64 + clocal] = acc[mf][nf][tau][r];
+                         + ct * 64 + clocal] = acc[mf][nf][ti][r];
                 }
             }
         }
@@ -329,10 +407,7 @@ extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
 
 extern "C" int conv3x3_wrw_supported(int H, int W, int C, int K) {
     if (C % 64 || K % 64) return 0;
-    // W == 8 (8x8 feature maps) measured slower than MIOpen (199 vs
-    // 167 us): half-empty 64-pixel chunks; needs image-pairing. Gated
-    // off until then.
-    if (W != 16 && W != 32) return 0;
+    if (W != 8 && W != 16 && W != 32) return 0;
     int P = (W == 32) ? 4 : 8;
     if (H % P) return 0;
     if (H < 2) return 0;
@@ -355,7 +430,7 @@ extern "C" void launch_conv3x3_wrw(
     const int P = (W == 32) ? 4 : 8;
     const int nsplit = conv3x3_wrw_nsplit(N, H, W, C, K);
     const int tiles = (K / 64) * (C / 64);
-    hipLaunchKernelGGL(k_conv3x3_wrw, dim3(tiles * nsplit), dim3(256), 0,
+    hipLaunchKernelGGL(k_conv3x3_wrw, dim3(tiles * nsplit), dim3(512), 0,
                        s, x, dy, ws, N, H, W, C, K, P, nsplit);
     const long n = (long)K * 9 * C;
     const int ngroups = (nsplit + 63) / 64;

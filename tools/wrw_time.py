@@ -3,7 +3,7 @@ sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(_
 from adaptdl_amd import ops
 ext = ops._load_extension()
 dev = torch.device("cuda")
-shapes = [(1024, 64, 32, 32, 64), (1024, 128, 16, 16, 128)]
+shapes = [(1024, 64, 32, 32, 64), (1024, 128, 16, 16, 128), (1024, 256, 8, 8, 256)]
 for n, c, h, w, k in shapes:
     x = torch.randn(n, c, h, w, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
     dy = torch.randn(n, k, h, w, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
