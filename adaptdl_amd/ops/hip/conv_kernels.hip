@@ -464,7 +464,7 @@ extern "C" void launch_conv3x3_wrw(
 // C % 32 == 0, C <= 128, K % KT == 0 (KT = 64 for C<=64 else 32).
 // =====================================================================
 
-extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
         ushort_t* __restrict__ y, int N, int H, int W, int C, int K,
         int P, int KT, int spread) {
@@ -483,7 +483,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
     const int t = threadIdx.x;
     const int lane = t & 63;
     const int wid = t >> 6;
-    const int wp = wid >> 1;         // pixel half (0..1): 64 pixels
+    const int wp = wid >> 1;         // pixel quarter (0..3): 32 pixels
     const int wk = wid & 1;          // ko half (0..1): KT/2 channels
     const int row16 = lane & 15;
     const int slot8 = (lane >> 4) * 8;
@@ -494,7 +494,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
     {
         const int pieces = KT * 9 * (C / 8);
         const ushort_t* wg = w + (size_t)kt * KT * 9 * C;
-        for (int i = t; i < pieces; i += 256) {
+        for (int i = t; i < pieces; i += 512) {
             const int row = i / (C / 8);        // ko * 9 + tau
             const int cg = (i % (C / 8)) * 8;
             V16 v;
@@ -504,7 +504,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
         }
     }
     // ---- zero the x pad columns once (cols 0 and W+1 of every line) --
-    for (int i = t; i < (P + 2) * (C / 8) * 2; i += 256) {
+    for (int i = t; i < (P + 2) * (C / 8) * 2; i += 512) {
         const int j = i / ((C / 8) * 2);
         const int rem = i % ((C / 8) * 2);
         const int col = (rem & 1) ? (W + 1) : 0;
@@ -526,7 +526,7 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
         // ---- stage the x window (pixel-major, natural layout) ----
         {
             const int pieces = (P + 2) * W * (C / 8);
-            for (int i = t; i < pieces; i += 256) {
+            for (int i = t; i < pieces; i += 512) {
                 const int j = i / (W * (C / 8));    // window line
                 const int rem = i % (W * (C / 8));
                 const int ww = rem / (C / 8);
@@ -546,9 +546,9 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
         __syncthreads();
 
         // ---- contraction: 9 taps x C/32 channel chunks ----
-        f32x4 acc[4][2];
+        f32x4 acc[2][2];
         #pragma unroll
-        for (int mf = 0; mf < 4; ++mf)
+        for (int mf = 0; mf < 2; ++mf)
             #pragma unroll
             for (int nf = 0; nf < 2; ++nf)
                 #pragma unroll
@@ -570,8 +570,8 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
                     }
                 }
                 #pragma unroll
-                for (int mf = 0; mf < 4; ++mf) {
-                    const int p = wp * 64 + mf * 16 + row16;
+                for (int mf = 0; mf < 2; ++mf) {
+                    const int p = wp * 32 + mf * 16 + row16;
                     const int li = p / W;
                     const int ww = p % W;
                     const int qrow = (li + dh) * LP + ww + dw;
@@ -592,13 +592,13 @@ extern "C" __global__ __launch_bounds__(256, 1) void k_conv3x3_mm(
         // ---- write the y tile ----
         const size_t ybase = ((size_t)n * H + h0) * W;  // chunk pixel 0
         #pragma unroll
-        for (int mf = 0; mf < 4; ++mf) {
+        for (int mf = 0; mf < 2; ++mf) {
             #pragma unroll
             for (int nf = 0; nf < 2; ++nf) {
                 if (nf >= NF) continue;
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
-                    const int p = wp * 64 + mf * 16 + (lane >> 4) * 4
+                    const int p = wp * 32 + mf * 16 + (lane >> 4) * 4
                         + r;
                     const int ko = kt * KT + wk * KH + nf * 16
                         + (lane & 15);
@@ -640,6 +640,6 @@ extern "C" void launch_conv3x3_mm(
     if (spread > chunks) spread = chunks;
     if (spread < 1) spread = 1;
     hipLaunchKernelGGL(k_conv3x3_mm, dim3((unsigned)(nk * spread)),
-                       dim3(256), lds_bytes, s, x, w, y, N, H, W, C, K,
+                       dim3(512), lds_bytes, s, x, w, y, N, H, W, C, K,
                        P, KT, (int)spread);
 }
