@@ -418,7 +418,7 @@ extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
     int P = (W == 32) ? 4 : 8;
     long chunks = (long)N * (H / P);
     int tiles = (K / 64) * (C / 64);
-    long target = 512 / tiles;
+    long target = 256 / tiles;  // one residency round (1 block/CU at 98 KB LDS)
     if (target < 1) target = 1;
     if (target > chunks) target = chunks;
     return (int)target;
@@ -636,7 +636,9 @@ extern "C" void launch_conv3x3_mm(
     }
     const long chunks = (long)N * (H / P);
     const int nk = K / KT;
-    long spread = 2048 / nk;
+    // ~112-127 KB dynamic LDS -> 1 block/CU: size the grid to exactly
+    // one residency round so the weight tile is staged once per CU.
+    long spread = 256 / nk;
     if (spread > chunks) spread = chunks;
     if (spread < 1) spread = 1;
     hipLaunchKernelGGL(k_conv3x3_mm, dim3((unsigned)(nk * spread)),

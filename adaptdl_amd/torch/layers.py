@@ -140,11 +140,12 @@ class _FusedConvFunction(torch.autograd.Function):
             .contiguous(memory_format=torch.channels_last)
         n, c, h, w = x.shape
         k = wb.shape[0]
-        # The custom fwd kernel is numerically verified but currently
-        # slower than MIOpen's forward igemm (tools/conv_time.py: 355 vs
-        # 221 us on layer1) -- opt-in only until it wins.
-        if os.getenv("ADAPTDL_EXPERIMENTAL_CONV_MM") == "1" and \
-                ext.conv_mm_ok(n, h, w, c, k):
+        # The custom fwd/bwd-data kernel is numerics-verified and beats
+        # MIOpen's UNTUNED find choice on C=64 (213 vs 229 us), but
+        # loses to its tuned in-context pick (154 us) -- end-to-end it
+        # cost ~2.5 ms/step, so it stays opt-in (bench_r06 vs r05).
+        if os.getenv("ADAPTDL_EXPERIMENTAL_CONV_MM") == "1" \
+                and ext.conv_mm_ok(n, h, w, c, k):
             y = torch.empty(n, k, h, w, dtype=x.dtype, device=x.device) \
                 .contiguous(memory_format=torch.channels_last)
             ext.conv_mm(x, wb, y)
@@ -165,8 +166,10 @@ class _FusedConvFunction(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             n, c, h, w = x.shape
             k = wb.shape[0]
-            if os.getenv("ADAPTDL_EXPERIMENTAL_CONV_MM") == "1" and \
-                    ext.conv_mm_ok(n, h, w, k, c):
+            # MIOpen's backward-data igemm (137 us on layer1) still
+            # beats conv_mm (213 us) -- keep it unless forced.
+            if os.getenv("ADAPTDL_EXPERIMENTAL_CONV_MM") == "1" \
+                    and ext.conv_mm_ok(n, h, w, k, c):
                 # dx = conv(dy, W flipped with C/K roles swapped)
                 wt = wb.flip(2, 3).permute(1, 0, 2, 3) \
                     .contiguous(memory_format=torch.channels_last)
