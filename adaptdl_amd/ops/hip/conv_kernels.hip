@@ -464,16 +464,62 @@ extern "C" void launch_conv3x3_wrw(
 // C % 32 == 0, C <= 128, K % KT == 0 (KT = 64 for C<=64 else 32).
 // =====================================================================
 
+struct MmRegs {
+    V16 v[5];
+};
+
+__device__ __forceinline__ void mm_issue(
+        const ushort_t* __restrict__ x, int n, int h0, int H, int W,
+        int C, int ct64, int P, int t, MmRegs& r) {
+    const int pieces = (P + 2) * W * (C / 8);
+    #pragma unroll
+    for (int it = 0; it < 5; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (W * (C / 8));
+            const int rem = i % (W * (C / 8));
+            const int ww = rem / (C / 8);
+            const int cg = (rem % (C / 8)) * 8;
+            const int h = h0 - 1 + j;
+            if (h < 0 || h >= H) {
+                r.v[it].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                r.v[it].u4 = *reinterpret_cast<const u32x4*>(
+                    x + (((size_t)n * H + h) * W + ww) * C + cg);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void mm_write(
+        short* __restrict__ x_s, int W, int C, int CS, int LP, int P,
+        int t, MmRegs& r) {
+    const int pieces = (P + 2) * W * (C / 8);
+    #pragma unroll
+    for (int it = 0; it < 5; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (W * (C / 8));
+            const int rem = i % (W * (C / 8));
+            const int ww = rem / (C / 8);
+            const int cg = (rem % (C / 8)) * 8;
+            *reinterpret_cast<u32x4*>(
+                &x_s[(j * LP + 1 + ww) * CS + cg]) = r.v[it].u4;
+        }
+    }
+}
+
 extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
         ushort_t* __restrict__ y, int N, int H, int W, int C, int K,
-        int P, int KT, int spread) {
+        int P, int KT, int spread, int dbuf) {
     extern __shared__ short lds[];
     const int CS = C + 8;            // padded channel stride (16B mult)
     const int LP = W + 2;            // padded line width (pixels)
     const int QR = (P + 2) * LP;     // x window rows
     short* w_lds = lds;              // [KT * 9][CS]
-    short* x_s = lds + KT * 9 * CS;  // [QR][CS]
+    short* x_s0 = lds + KT * 9 * CS; // [QR][CS] buffer 0
+    short* x_s1 = x_s0 + (dbuf ? QR * CS : 0);  // buffer 1 (pipelined)
 
     const int CP = P * W;            // chunk pixels (= 128)
     const int nk = K / KT;
@@ -503,47 +549,50 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
             *reinterpret_cast<u32x4*>(&w_lds[row * CS + cg]) = v.u4;
         }
     }
-    // ---- zero the x pad columns once (cols 0 and W+1 of every line) --
-    for (int i = t; i < (P + 2) * (C / 8) * 2; i += 512) {
-        const int j = i / ((C / 8) * 2);
-        const int rem = i % ((C / 8) * 2);
-        const int col = (rem & 1) ? (W + 1) : 0;
-        const int cg = (rem >> 1) * 8;
-        u32x4 z = {0u, 0u, 0u, 0u};
-        *reinterpret_cast<u32x4*>(
-            &x_s[(j * LP + col) * CS + cg]) = z;
+    // ---- zero x pad columns of both buffers once ----
+    for (int b = 0; b < (dbuf ? 2 : 1); ++b) {
+        short* x_s = b ? x_s1 : x_s0;
+        for (int i = t; i < (P + 2) * (C / 8) * 2; i += 512) {
+            const int j = i / ((C / 8) * 2);
+            const int rem = i % ((C / 8) * 2);
+            const int col = (rem & 1) ? (W + 1) : 0;
+            const int cg = (rem >> 1) * 8;
+            u32x4 z = {0u, 0u, 0u, 0u};
+            *reinterpret_cast<u32x4*>(
+                &x_s[(j * LP + col) * CS + cg]) = z;
+        }
     }
-    __syncthreads();
 
     const int lines_per_img = H / P;
     const long chunks_total = (long)N * lines_per_img;
 
+    MmRegs regs;
+    // Prologue: stage chunk 0 into buffer 0; issue chunk 1's loads.
+    if (sp < chunks_total) {
+        const int n0 = (int)(sp / lines_per_img);
+        const int h00 = (int)(sp % lines_per_img) * P;
+        mm_issue(x, n0, h00, H, W, C, 0, P, t, regs);
+        mm_write(x_s0, W, C, CS, LP, P, t, regs);
+    }
+    __syncthreads();
+    if (dbuf && sp + spread < chunks_total) {
+        const long qn = sp + spread;
+        mm_issue(x, (int)(qn / lines_per_img),
+                 (int)(qn % lines_per_img) * P, H, W, C, 0, P, t, regs);
+    }
+
+    int cur = 0;
     for (long q = sp; q < chunks_total; q += spread) {
         const int n = (int)(q / lines_per_img);
         const int h0 = (int)(q % lines_per_img) * P;
+        const short* x_s = (dbuf && cur) ? x_s1 : x_s0;
 
-        __syncthreads();
-        // ---- stage the x window (pixel-major, natural layout) ----
-        {
-            const int pieces = (P + 2) * W * (C / 8);
-            for (int i = t; i < pieces; i += 512) {
-                const int j = i / (W * (C / 8));    // window line
-                const int rem = i % (W * (C / 8));
-                const int ww = rem / (C / 8);
-                const int cg = (rem % (C / 8)) * 8;
-                const int h = h0 - 1 + j;
-                V16 v;
-                if (h < 0 || h >= H) {
-                    v.u4 = u32x4{0u, 0u, 0u, 0u};
-                } else {
-                    v.u4 = *reinterpret_cast<const u32x4*>(
-                        x + (((size_t)n * H + h) * W + ww) * C + cg);
-                }
-                *reinterpret_cast<u32x4*>(
-                    &x_s[(j * LP + 1 + ww) * CS + cg]) = v.u4;
-            }
+        if (!dbuf && q != sp) {
+            __syncthreads();
+            mm_issue(x, n, h0, H, W, C, 0, P, t, regs);
+            mm_write(x_s0, W, C, CS, LP, P, t, regs);
+            __syncthreads();
         }
-        __syncthreads();
 
         // ---- contraction: 9 taps x C/32 channel chunks ----
         f32x4 acc[2][2];
@@ -606,6 +655,22 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
                 }
             }
         }
+
+        // ---- pipelined: stage q+1 into the other buffer, issue q+2 --
+        if (dbuf) {
+            if (q + spread < chunks_total) {
+                short* nxt = cur ? x_s0 : x_s1;
+                mm_write(nxt, W, C, CS, LP, P, t, regs);
+                if (q + 2 * spread < chunks_total) {
+                    const long q2 = q + 2 * spread;
+                    mm_issue(x, (int)(q2 / lines_per_img),
+                             (int)(q2 % lines_per_img) * P, H, W, C, 0,
+                             P, t, regs);
+                }
+            }
+            __syncthreads();
+            cur ^= 1;
+        }
     }
 }
 
@@ -626,7 +691,13 @@ extern "C" void launch_conv3x3_mm(
     const int KT = (C <= 64) ? 64 : 32;
     const int CS = C + 8;
     const int QR = (P + 2) * (W + 2);
-    const size_t lds_bytes = ((size_t)KT * 9 + QR) * CS * 2;
+    // double-buffer the x window when it fits the 160 KB LDS
+    size_t lds_bytes = ((size_t)KT * 9 + 2 * QR) * CS * 2;
+    int dbuf = 1;
+    if (lds_bytes > 160 * 1024) {
+        dbuf = 0;
+        lds_bytes = ((size_t)KT * 9 + QR) * CS * 2;
+    }
     static int attr_set = 0;
     if (!attr_set) {
         hipFuncSetAttribute(
@@ -643,5 +714,5 @@ extern "C" void launch_conv3x3_mm(
     if (spread < 1) spread = 1;
     hipLaunchKernelGGL(k_conv3x3_mm, dim3((unsigned)(nk * spread)),
                        dim3(512), lds_bytes, s, x, w, y, N, H, W, C, K,
-                       P, KT, (int)spread);
+                       P, KT, (int)spread, dbuf);
 }
