@@ -525,7 +525,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         ushort_t* __restrict__ y, int N, int H, int W, int C, int K,
         int P, int KT, int spread, int dbuf) {
     extern __shared__ short lds[];
-    const int CS = C + 8;            // padded channel stride (16B mult)
+    const int CS = C + 16;           // pad: zero conflicts under the real b128 lane groups
     const int LP = W + 2;            // padded line width (pixels)
     const int QR = (P + 2) * LP;     // x window rows
     short* w_lds = lds;              // [KT * 9][CS]
@@ -700,7 +700,7 @@ extern "C" void launch_conv3x3_mm(
         int N, int H, int W, int C, int K, hipStream_t s) {
     const int P = (W == 32) ? 4 : 8;
     const int KT = (C <= 64) ? 64 : 32;
-    const int CS = C + 8;
+    const int CS = C + 16;
     const int QR = (P + 2) * (W + 2);
     // double-buffer the x window when it fits the 160 KB LDS
     size_t lds_bytes = ((size_t)KT * 9 + 2 * QR) * CS * 2;
