@@ -71,6 +71,11 @@ def _free_port():
         return s.getsockname()[1]
 
 
+def _version():
+    import adaptdl_amd
+    return adaptdl_amd.__version__
+
+
 class JobSpec(object):
     """What to run and how elastic it may be.
 
@@ -351,6 +356,7 @@ class LocalController(object):
                 "ADAPTDL_NUM_NODES": "1",
                 "ADAPTDL_NUM_RESTARTS": str(job.num_restarts),
                 "ADAPTDL_SUPERVISOR_URL": self.supervisor.url,
+                "ADAPTDL_SCHED_VERSION": _version(),
             })
             if spec.gpus_per_replica > 0:
                 mine = gpus[rank * spec.gpus_per_replica:

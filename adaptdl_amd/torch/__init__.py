@@ -50,6 +50,17 @@ __all__ = [
 ]
 
 
+def version_check():
+    """Warn when the scheduler that launched us is a different version
+    (reference torch/__init__.py:43-48 parity)."""
+    import adaptdl_amd
+    sched = adaptdl_amd.env.sched_version()
+    if sched is not None and sched != adaptdl_amd.__version__:
+        LOG.warning(
+            "adaptdl_amd version %s differs from scheduler version %s",
+            adaptdl_amd.__version__, sched)
+
+
 def _pick_free_port():
     with socket.socket(socket.AF_INET, socket.SOCK_STREAM) as s:
         s.bind(("0.0.0.0", 0))
@@ -68,6 +79,7 @@ def init_process_group(backend=None, init_method=None, world_size=None,
     replaces).
     """
     install_signal_handlers()
+    version_check()
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     num_replicas = world_size if world_size is not None \
