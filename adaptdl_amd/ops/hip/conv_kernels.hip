@@ -622,10 +622,16 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
                 for (int r = 0; r < 4; ++r)
                     acc[mf][nf][r] = 0.f;
 
-        for (int tau = 0; tau < 9; ++tau) {
-            const int dh = tau / 3;
-            const int dw = tau % 3;
-            for (int cc = 0; cc < C / 32; ++cc) {
+        // cc outer (runtime trip count, stays rolled); tau inner and
+        // fully unrolled: a rolled tau loop compiled to 4 MFMAs each
+        // preceded by s_waitcnt lgkmcnt(0) — LDS reads and MFMA issue
+        // fully serialized.  Unrolled, the body exposes 36 MFMAs + 36
+        // independent b128 reads of ILP per cc iteration.
+        for (int cc = 0; cc < C / 32; ++cc) {
+            #pragma unroll
+            for (int tau = 0; tau < 9; ++tau) {
+                const int dh = tau / 3;
+                const int dw = tau % 3;
                 bf16x8 bfrag[2];
                 #pragma unroll
                 for (int nf = 0; nf < 2; ++nf) {
