@@ -35,6 +35,10 @@ void launch_bn_bwd(const unsigned short*, const unsigned short*,
                    float*, float*, float*, hipStream_t);
 int conv3x3_wrw_supported(int, int, int, int);
 int conv3x3_mm_supported(int, int, int, int);
+int conv3x3_s2_bwd_supported(int, int, int, int);
+void launch_conv3x3_s2_bwd(const unsigned short*, const unsigned short*,
+                           unsigned short*, int, int, int, int, int,
+                           hipStream_t);
 void launch_conv3x3_mm(const unsigned short*, const unsigned short*,
                        unsigned short*, int, int, int, int, int,
                        hipStream_t);
@@ -309,6 +313,35 @@ void conv_mm(torch::Tensor x, torch::Tensor w, torch::Tensor y) {
                       (int)N, (int)H, (int)Wd, (int)C, (int)K, stream());
 }
 
+bool conv_s2_bwd_ok(long N, long Ho, long Wo, long K, long C) {
+    (void)N;
+    return conv3x3_s2_bwd_supported((int)Ho, (int)Wo, (int)K, (int)C)
+        != 0;
+}
+
+// dx = conv3x3_s2p1_backward_data(dy, w): dy (N,K,Ho,Wo) channels_last
+// bf16; wt is the host-prepared (C,3,3,K) CONTIGUOUS transposed weight
+// (w.permute(1,2,3,0).contiguous()); dx (N,C,2Ho,2Wo) channels_last.
+void conv_s2_bwd(torch::Tensor dy, torch::Tensor wt, torch::Tensor dx) {
+    check_bn_x(dy, "dy"); check_bn_x(dx, "dx");
+    const long N = dy.size(0), K = dy.size(1), Ho = dy.size(2),
+        Wo = dy.size(3);
+    const long C = dx.size(1);
+    TORCH_CHECK(dx.size(0) == N && dx.size(2) == 2 * Ho &&
+                dx.size(3) == 2 * Wo, "dx shape mismatch");
+    TORCH_CHECK(wt.is_cuda() && wt.scalar_type() == torch::kBFloat16 &&
+                wt.is_contiguous() && wt.size(0) == C &&
+                wt.size(1) == 3 && wt.size(2) == 3 && wt.size(3) == K,
+                "wt must be (C,3,3,K) contiguous bf16");
+    TORCH_CHECK(conv_s2_bwd_ok(N, Ho, Wo, K, C),
+                "unsupported s2 bwd-data shape");
+    launch_conv3x3_s2_bwd((const unsigned short*)dy.data_ptr(),
+                          (const unsigned short*)wt.data_ptr(),
+                          (unsigned short*)dx.data_ptr(),
+                          (int)N, (int)Ho, (int)Wo, (int)K, (int)C,
+                          stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -329,4 +362,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("conv_wrw", &conv_wrw, "MFMA 3x3 s1 NHWC weight gradient");
     mod.def("conv_mm_ok", &conv_mm_ok, "3x3 fwd/bwd-data predicate");
     mod.def("conv_mm", &conv_mm, "MFMA 3x3 s1 NHWC conv (fwd/bwd-data)");
+    mod.def("conv_s2_bwd_ok", &conv_s2_bwd_ok,
+            "3x3 stride-2 bwd-data predicate");
+    mod.def("conv_s2_bwd", &conv_s2_bwd,
+            "MFMA 3x3 s2 NHWC backward-data (polyphase)");
 }

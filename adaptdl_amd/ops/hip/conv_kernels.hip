@@ -593,7 +593,13 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         mm_write(x_s0, W, C, CS, LP, P, t, regs);
     }
     __syncthreads();
-    if (dbuf && sp + spread < chunks_total) {
+    // Both paths prefetch chunk sp+spread now: with dbuf it lands in
+    // the other LDS buffer; without, it stays in regs through the
+    // contraction of chunk sp (register double-buffering — the 160 KB
+    // LDS can't hold two x windows at C=128, and an un-prefetched
+    // single-buffer path exposes the full global-load latency between
+    // chunks: layer2 measured 259 us vs layer1's pipelined 174).
+    if (sp + spread < chunks_total) {
         const long qn = sp + spread;
         mm_issue(x, (int)(qn / lines_per_img),
                  (int)(qn % lines_per_img) * P, H, W, C, 0, P, t, regs);
@@ -606,10 +612,18 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         const short* x_s = (dbuf && cur) ? x_s1 : x_s0;
 
         if (!dbuf && q != sp) {
-            __syncthreads();
-            mm_issue(x, n, h0, H, W, C, 0, P, t, regs);
+            // regs already hold chunk q (prefetched during the previous
+            // contraction); commit it to LDS, then start chunk q+spread's
+            // loads so they fly behind this chunk's MFMA work.
+            __syncthreads();   // all waves done with x_s (y-store bounce)
             mm_write(x_s0, W, C, CS, LP, P, t, regs);
             __syncthreads();
+            if (q + spread < chunks_total) {
+                const long qn = q + spread;
+                mm_issue(x, (int)(qn / lines_per_img),
+                         (int)(qn % lines_per_img) * P, H, W, C, 0, P, t,
+                         regs);
+            }
         }
 
         // ---- contraction: 9 taps x C/32 channel chunks ----
@@ -762,4 +776,326 @@ extern "C" void launch_conv3x3_mm(
     hipLaunchKernelGGL(k_conv3x3_mm, dim3((unsigned)(nk * spread)),
                        dim3(512), lds_bytes, s, x, w, y, N, H, W, C, K,
                        P, KT, (int)spread, dbuf);
+}
+
+// =====================================================================
+// MFMA 3x3 STRIDE-2 backward-data (NHWC bf16), polyphase decomposition.
+//
+// dx[n,hi,wi,c] = sum_{k,dh,dw : hi=2ho+dh-1, wi=2wo+dw-1} dy[n,ho,wo,k]
+//                 * w[k,dh,dw,c]
+//
+// Profiling (profiles/bench_r05_kernel_stats.csv) shows the stride-2
+// bwd-data falling to a CK kernel at ~1035 us/call (~75 GF/s) -- the
+// single largest per-call conv entry of the ResNet step.  Splitting dx
+// by pixel parity (a,b) = (hi%2, wi%2) turns the strided scatter into
+// four dense stride-1 mini-convs over dy with disjoint tap sets:
+//
+//   phase (a,b): taps dh in (a ? {0,2} : {1}) x dw in (b ? {0,2} : {1})
+//   ho = h2 + (dh==0), wo = w2 + (dw==0)      (hi=2*h2+a, wi=2*w2+b)
+//
+// (1 + 2 + 2 + 4 taps = the 9 taps, each used exactly once.)
+// Block layout (conv_mm skeleton: natural-layout staging, all fragment
+// reads aligned b128, register-prefetch pipeline, y-store bounce):
+//   - a chunk is P2 phase-rows = 2*P2 full dx lines of one image; the
+//     dy window is P2+1 lines (bottom halo only: tap offsets are >= 0),
+//     pixel-major [line][wo (+8 zero pad cols)][K+16 stride],
+//   - w_lds holds the TRANSPOSED weight tile wt[c][tau][k] (contiguous
+//     in the contraction dim k; prepared host-side as a [C][3][3][K]
+//     tensor -- cheap: weights are tiny),
+//   - waves: 4 pixel-quarters x (CT/16 = 2) c-frags; each quarter owns
+//     one fragment of phase pair {(0,0),(1,1)} or {(0,1),(1,0)} so the
+//     per-wave tap count is 5 vs 4 (1+4 / 2+2) instead of the 4x
+//     imbalance of phase-per-quarter,
+//   - the chunk's 4 phases together tile 2*P2 COMPLETE dx lines, so the
+//     bounce-staged store writes plain contiguous b128 lines.
+// Constraints (v1): Wo == 16 (so a 16-pixel fragment row is one phase
+// line), Ho % P2 == 0, K % 32 == 0, C % 32 == 0, CT = 32.
+// =====================================================================
+
+#define S2_P2 2
+#define S2_CT 32
+
+struct S2Regs {
+    V16 v[2];
+};
+
+__device__ __forceinline__ void s2_issue(
+        const ushort_t* __restrict__ dy, int n, int r0, int Ho, int Wo,
+        int K, int t, S2Regs& r) {
+    const int pieces = (S2_P2 + 1) * Wo * (K / 8);
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (Wo * (K / 8));
+            const int rem = i % (Wo * (K / 8));
+            const int wo = rem / (K / 8);
+            const int kg = (rem % (K / 8)) * 8;
+            const int ho = r0 + j;
+            if (ho >= Ho) {
+                r.v[it].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                r.v[it].u4 = *reinterpret_cast<const u32x4*>(
+                    dy + (((size_t)n * Ho + ho) * Wo + wo) * K + kg);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void s2_write(
+        short* __restrict__ dy_s, int Wo, int K, int KS, int LW, int t,
+        S2Regs& r) {
+    const int pieces = (S2_P2 + 1) * Wo * (K / 8);
+    #pragma unroll
+    for (int it = 0; it < 2; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (Wo * (K / 8));
+            const int rem = i % (Wo * (K / 8));
+            const int wo = rem / (K / 8);
+            const int kg = (rem % (K / 8)) * 8;
+            *reinterpret_cast<u32x4*>(
+                &dy_s[(j * LW + wo) * KS + kg]) = r.v[it].u4;
+        }
+    }
+    // Refresh the zero pad columns every staging: the dx bounce store
+    // reuses this buffer and clobbers them.
+    const int zp = (S2_P2 + 1) * 8 * (K / 8);
+    for (int i = t; i < zp; i += 512) {
+        const int j = i / (8 * (K / 8));
+        const int rem = i % (8 * (K / 8));
+        const int pc = rem / (K / 8);
+        const int kg = (rem % (K / 8)) * 8;
+        u32x4 z = {0u, 0u, 0u, 0u};
+        *reinterpret_cast<u32x4*>(
+            &dy_s[(j * LW + Wo + pc) * KS + kg]) = z;
+    }
+}
+
+// Contraction for wave-quarter Q.  Fragment f's phase: quarters 0..1
+// own {(0,0),(1,1)}, quarters 2..3 own {(0,1),(1,0)}; h2 = Q&1.
+template <int Q>
+__device__ __forceinline__ void s2_contract(
+        const short* __restrict__ dy_s, const short* __restrict__ w_lds,
+        int K, int KS, int KS2, int LW, int wn, int row16, int slot8,
+        f32x4 (&acc)[2]) {
+    const int h2 = Q & 1;
+    const int w2 = row16;            // Wo == 16: fragment row == w2
+    #pragma unroll 2
+    for (int cc = 0; cc < K / 32; ++cc) {
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) {
+            constexpr int PH0 = (Q < 2) ? 0 : 1;   // f == 0 phase
+            constexpr int PH1 = (Q < 2) ? 3 : 2;   // f == 1 phase
+            const int ph = f ? PH1 : PH0;
+            const int a = ph >> 1, b = ph & 1;
+            #pragma unroll
+            for (int dh = 0; dh < 3; ++dh) {
+                if (a ? (dh == 1) : (dh != 1)) continue;
+                #pragma unroll
+                for (int dw = 0; dw < 3; ++dw) {
+                    if (b ? (dw == 1) : (dw != 1)) continue;
+                    const int j = h2 + (dh == 0 ? 1 : 0);
+                    const int col = w2 + (dw == 0 ? 1 : 0);
+                    const bf16x8 afrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &dy_s[(j * LW + col) * KS + cc * 32
+                                  + slot8]);
+                    const int clocal = wn * 16 + row16;
+                    const int tau = dh * 3 + dw;
+                    const bf16x8 bfrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &w_lds[((clocal * 9) + tau) * KS2
+                                   + cc * 32 + slot8]);
+                    acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        afrag, bfrag, acc[f], 0, 0, 0);
+                }
+            }
+        }
+    }
+}
+
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd(
+        const ushort_t* __restrict__ dy, const ushort_t* __restrict__ wt,
+        ushort_t* __restrict__ dx, int N, int Ho, int Wo, int K, int C,
+        int spread, int dbuf) {
+    extern __shared__ short lds[];
+    const int KS = K + 16;           // dy_s row stride (zero conflicts)
+    const int KS2 = K + 8;           // w_lds row stride
+    const int LW = Wo + 8;           // dy line width incl. zero pad cols
+    const int WIN = (S2_P2 + 1) * LW;
+    short* w_lds = lds;              // [S2_CT * 9][KS2]
+    short* dy_s0 = lds + S2_CT * 9 * KS2;
+    short* dy_s1 = dy_s0 + (dbuf ? WIN * KS : 0);
+
+    const int Hi = 2 * Ho, Wi = 2 * Wo;
+    const int nct = C / S2_CT;
+    const int ct = blockIdx.x / spread;
+    const int sp = blockIdx.x % spread;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wq = wid >> 1;         // pixel quarter (phase pair + h2)
+    const int wn = wid & 1;          // c fragment within the CT tile
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+
+    // ---- stage the transposed weight tile wt[c][tau][k] once ----
+    {
+        const int pieces = S2_CT * 9 * (K / 8);
+        const ushort_t* wg = wt + (size_t)ct * S2_CT * 9 * K;
+        for (int i = t; i < pieces; i += 512) {
+            const int row = i / (K / 8);          // clocal * 9 + tau
+            const int kg = (i % (K / 8)) * 8;
+            *reinterpret_cast<u32x4*>(&w_lds[row * KS2 + kg]) =
+                *reinterpret_cast<const u32x4*>(
+                    wg + (size_t)row * K + kg);
+        }
+    }
+    // (pad columns are zeroed inside s2_write on every staging)
+    const int rows_per_img = Ho / S2_P2;
+    const long chunks_total = (long)N * rows_per_img;
+
+    S2Regs regs;
+    if (sp < chunks_total) {
+        s2_issue(dy, (int)(sp / rows_per_img),
+                 (int)(sp % rows_per_img) * S2_P2, Ho, Wo, K, t, regs);
+        s2_write(dy_s0, Wo, K, KS, LW, t, regs);
+    }
+    __syncthreads();
+    if (sp + spread < chunks_total) {
+        const long qn = sp + spread;
+        s2_issue(dy, (int)(qn / rows_per_img),
+                 (int)(qn % rows_per_img) * S2_P2, Ho, Wo, K, t, regs);
+    }
+
+    int cur = 0;
+    for (long q = sp; q < chunks_total; q += spread) {
+        const int n = (int)(q / rows_per_img);
+        const int r0 = (int)(q % rows_per_img) * S2_P2;
+        short* dy_s = (dbuf && cur) ? dy_s1 : dy_s0;
+
+        if (!dbuf && q != sp) {
+            __syncthreads();
+            s2_write(dy_s0, Wo, K, KS, LW, t, regs);
+            __syncthreads();
+            if (q + spread < chunks_total) {
+                const long qn = q + spread;
+                s2_issue(dy, (int)(qn / rows_per_img),
+                         (int)(qn % rows_per_img) * S2_P2, Ho, Wo, K, t,
+                         regs);
+            }
+        }
+
+        f32x4 acc[2];
+        #pragma unroll
+        for (int f = 0; f < 2; ++f)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                acc[f][r] = 0.f;
+
+        switch (wq) {
+        case 0: s2_contract<0>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                               slot8, acc); break;
+        case 1: s2_contract<1>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                               slot8, acc); break;
+        case 2: s2_contract<2>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                               slot8, acc); break;
+        default: s2_contract<3>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                                slot8, acc); break;
+        }
+
+        // ---- bounce the dx tile through the (now free) dy_s buffer so
+        // global stores go out as whole 16B lines ----
+        ushort_t* ystage = reinterpret_cast<ushort_t*>(dy_s);
+        __syncthreads();             // all waves done reading dy_s
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) {
+            const int ph = (wq < 2) ? (f ? 3 : 0) : (f ? 2 : 1);
+            const int h2 = wq & 1;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int w2 = (lane >> 4) * 4 + r;
+                const int c = wn * 16 + (lane & 15);
+                const int s = (ph * S2_P2 + h2) * Wo + w2;
+                ystage[s * S2_CT + c] = f2b(acc[f][r]);
+            }
+        }
+        __syncthreads();
+        {
+            const int pieces = 2 * S2_P2 * Wi * (S2_CT / 8);
+            const size_t base = ((size_t)n * Hi + 2 * r0) * Wi;
+            for (int i = t; i < pieces; i += 512) {
+                const int hl = i / (Wi * (S2_CT / 8));
+                const int rem = i % (Wi * (S2_CT / 8));
+                const int wi = rem / (S2_CT / 8);
+                const int cg = (rem % (S2_CT / 8)) * 8;
+                const int ph = ((hl & 1) << 1) | (wi & 1);
+                const int s = (ph * S2_P2 + (hl >> 1)) * Wo + (wi >> 1);
+                *reinterpret_cast<u32x4*>(
+                    dx + (base + (size_t)hl * Wi + wi) * C
+                    + ct * S2_CT + cg) =
+                    *reinterpret_cast<const u32x4*>(
+                        &ystage[s * S2_CT + cg]);
+            }
+        }
+
+        if (dbuf) {
+            if (q + spread < chunks_total) {
+                short* nxt = cur ? dy_s0 : dy_s1;
+                s2_write(nxt, Wo, K, KS, LW, t, regs);
+                if (q + 2 * spread < chunks_total) {
+                    const long q2 = q + 2 * spread;
+                    s2_issue(dy, (int)(q2 / rows_per_img),
+                             (int)(q2 % rows_per_img) * S2_P2, Ho, Wo, K,
+                             t, regs);
+                }
+            }
+            __syncthreads();
+            cur ^= 1;
+        }
+    }
+}
+
+extern "C" int conv3x3_s2_bwd_supported(int Ho, int Wo, int K, int C) {
+    if (Wo != 16) return 0;
+    if (Ho % S2_P2) return 0;
+    if (K % 32 || C % S2_CT) return 0;
+    const int pieces = (S2_P2 + 1) * Wo * (K / 8);
+    if (pieces > 1024) return 0;     // s2_issue register budget
+    // LDS: weight tile + at least one dy window must fit.
+    size_t need = ((size_t)S2_CT * 9 * (K + 8)
+                   + (size_t)(S2_P2 + 1) * (Wo + 8) * (K + 16)) * 2;
+    if (need > 160 * 1024) return 0;
+    return 1;
+}
+
+extern "C" void launch_conv3x3_s2_bwd(
+        const ushort_t* dy, const ushort_t* wt, ushort_t* dx,
+        int N, int Ho, int Wo, int K, int C, hipStream_t s) {
+    const int KS = K + 16, KS2 = K + 8, LW = Wo + 8;
+    const size_t wbytes = (size_t)S2_CT * 9 * KS2 * 2;
+    const size_t dybytes = (size_t)(S2_P2 + 1) * LW * KS * 2;
+    int dbuf = 1;
+    size_t lds_bytes = wbytes + 2 * dybytes;
+    if (lds_bytes > 160 * 1024) {
+        dbuf = 0;
+        lds_bytes = wbytes + dybytes;
+    }
+    static int attr_set = 0;
+    if (!attr_set) {
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_s2_bwd),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = 1;
+    }
+    const long chunks = (long)N * (Ho / S2_P2);
+    const int nct = C / S2_CT;
+    long spread = 256 / nct;
+    if (spread > chunks) spread = chunks;
+    if (spread < 1) spread = 1;
+    hipLaunchKernelGGL(k_conv3x3_s2_bwd, dim3((unsigned)(nct * spread)),
+                       dim3(512), lds_bytes, s, dy, wt, dx, N, Ho, Wo,
+                       K, C, (int)spread, dbuf);
 }

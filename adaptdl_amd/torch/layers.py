@@ -191,6 +191,46 @@ class _FusedConvFunction(torch.autograd.Function):
         return dx, dw, None, None, None, None
 
 
+class _S2ConvFunction(torch.autograd.Function):
+    """3x3 stride-2 conv with backward-data on the polyphase MFMA kernel.
+
+    The stride-2 backward-data otherwise falls to a CK scatter kernel
+    measured at ~1035 us/call (~75 GF/s) -- the largest per-call conv
+    entry of the ResNet step (profiles/bench_r05_kernel_stats.csv).
+    Forward and the weight gradient stay on MIOpen.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding, dilation, groups):
+        wb = weight.detach().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        y = torch.nn.functional.conv2d(x, wb, None, stride, padding,
+                                       dilation, groups)
+        ctx.save_for_backward(x, wb)
+        ctx.conv_args = (stride, padding, dilation, groups)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops._load_extension()
+        x, wb = ctx.saved_tensors
+        stride, padding, dilation, groups = ctx.conv_args
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            # dx[even/odd phases] = dense mini-convs of dy with the
+            # (C,3,3,K)-transposed weight (no flip: taps keep their
+            # original (dh, dw) indices in the polyphase decomposition).
+            wt = wb.permute(1, 2, 3, 0).contiguous()
+            dx = torch.empty_like(x)
+            ext.conv_s2_bwd(dy, wt, dx)
+        if ctx.needs_input_grad[1]:
+            dw = torch.ops.aten.convolution_backward(
+                dy, x, wb, None, stride, padding, dilation, False,
+                [0, 0], groups, [False, True, False])[1]
+        return dx, dw, None, None, None, None
+
+
 class FusedConv2d(nn.Conv2d):
     """Conv2d whose 3x3/s1 weight gradient runs the MFMA wrw kernel."""
 
@@ -206,6 +246,21 @@ class FusedConv2d(nn.Conv2d):
         n, c, h, w = x.shape
         return bool(ext.conv_wrw_ok(n, h, w, c, self.out_channels))
 
+    def _s2_ok(self, x):
+        if not (x.is_cuda and x.dtype == torch.bfloat16 and
+                x.dim() == 4 and ops.has_extension() and
+                self.bias is None and self.groups == 1 and
+                self.kernel_size == (3, 3) and self.stride == (2, 2) and
+                self.padding == (1, 1) and self.dilation == (1, 1) and
+                x.is_contiguous(memory_format=torch.channels_last)):
+            return False
+        ext = ops._load_extension()
+        n, c, h, w = x.shape
+        if h % 2 or w % 2:
+            return False
+        return bool(ext.conv_s2_bwd_ok(n, h // 2, w // 2,
+                                       self.out_channels, c))
+
     def forward(self, x):
         if torch.is_autocast_enabled() and x.is_cuda and \
                 x.dtype != torch.bfloat16 and \
@@ -216,4 +271,8 @@ class FusedConv2d(nn.Conv2d):
             return _FusedConvFunction.apply(x, self.weight, self.stride,
                                             self.padding, self.dilation,
                                             self.groups)
+        if x.requires_grad and self._s2_ok(x):
+            return _S2ConvFunction.apply(x, self.weight, self.stride,
+                                         self.padding, self.dilation,
+                                         self.groups)
         return super().forward(x)

@@ -205,7 +205,10 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": elapsed * 1000.0 / args.steps,
             "higher_is_better": True,
-            "scaling": "weak",
+            # The named config caps the GLOBAL batch at 4096 for every
+            # replica count (autoscale_batch_size(4096, ...)), so total
+            # work per step is fixed as N grows: strong scaling.
+            "scaling": "strong",
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",

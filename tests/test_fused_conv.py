@@ -150,3 +150,57 @@ def test_conv_mm_full_autograd_roundtrip(monkeypatch):
     assert torch.allclose(x.grad.float(), x32.grad, atol=0.3, rtol=5e-2)
     assert torch.allclose(conv.weight.grad, ref.weight.grad,
                           atol=0.3, rtol=5e-2)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 16, 16, 128),     # layer2.0.conv1-like (dy side)
+    (3, 32, 16, 16, 64),      # small channels
+])
+def test_s2_bwd_kernel_matches_fp32(shape):
+    """Polyphase stride-2 backward-data vs fp32 autograd reference."""
+    from adaptdl_amd import ops
+    ext = ops._load_extension()
+    torch.manual_seed(5)
+    n, c, ho, wo, k = shape
+    hi, wi = 2 * ho, 2 * wo
+    dev = torch.device("cuda")
+    dy = (torch.randn(n, k, ho, wo, device=dev) * 0.5) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(k, c, 3, 3, device=dev) * 0.2).to(torch.bfloat16)
+
+    assert ext.conv_s2_bwd_ok(n, ho, wo, k, c)
+    wt = w.permute(1, 2, 3, 0).contiguous()
+    dx = torch.empty(n, c, hi, wi, dtype=torch.bfloat16, device=dev) \
+        .contiguous(memory_format=torch.channels_last)
+    ext.conv_s2_bwd(dy, wt, dx)
+
+    ref = torch.nn.grad.conv2d_input(
+        (n, c, hi, wi), w.float(), dy.float(), stride=2, padding=1)
+    assert torch.allclose(dx.float(), ref, atol=0.1, rtol=5e-2), \
+        (dx.float() - ref).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_s2_conv_module_backward():
+    """FusedConv2d stride-2 path: dx/dw through _S2ConvFunction."""
+    torch.manual_seed(6)
+    dev = torch.device("cuda")
+    conv = FusedConv2d(64, 128, 3, stride=2, padding=1, bias=False).to(dev)
+    x = torch.randn(2, 64, 32, 32, device=dev).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv(x)
+    assert y.shape == (2, 128, 16, 16)
+    y.backward(torch.randn_like(y))
+    assert x.grad is not None and conv.weight.grad is not None
+
+    # reference grads on the same bf16-rounded dy
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = conv.weight.detach().float().requires_grad_(True)
+    dy = torch.randn(2, 128, 16, 16, device=dev).to(torch.bfloat16)
+    x.grad = None
+    conv.weight.grad = None
+    conv(x).backward(dy.contiguous(memory_format=torch.channels_last))
+    y32 = F.conv2d(x32, w32, stride=2, padding=1)
+    y32.backward(dy.float())
+    assert torch.allclose(x.grad.float(), x32.grad, atol=0.1, rtol=5e-2)

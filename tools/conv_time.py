@@ -28,3 +28,37 @@ for n, c, h, w, k in shapes:
     us_m = t0.elapsed_time(t1) * 100
     gf = 2 * n * k * c * 9 * h * w / 1e9
     print(f"fwd N{n} C{c} {h}x{w} K{k}: ours {us:7.1f}us ({gf/us*1e3:6.0f} GF/s) miopen {us_m:7.1f}us ({gf/us_m*1e3:6.0f} GF/s)")
+
+# stride-2 backward-data: polyphase MFMA kernel vs library (CK) path
+for n, c, ho, wo, k in [(1024, 64, 16, 16, 128)]:
+    hi, wi = 2 * ho, 2 * wo
+    dy = torch.randn(n, k, ho, wo, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(k, c, 3, 3, device=dev) * 0.1).to(torch.bfloat16)
+    wt = w.permute(1, 2, 3, 0).contiguous()
+    dx = torch.empty(n, c, hi, wi, dtype=torch.bfloat16, device=dev).contiguous(memory_format=torch.channels_last)
+    assert ext.conv_s2_bwd_ok(n, ho, wo, k, c)
+    for _ in range(3):
+        ext.conv_s2_bwd(dy, wt, dx)
+    torch.cuda.synchronize()
+    t0 = torch.cuda.Event(enable_timing=True); t1 = torch.cuda.Event(enable_timing=True)
+    t0.record()
+    for _ in range(10):
+        ext.conv_s2_bwd(dy, wt, dx)
+    t1.record(); torch.cuda.synchronize()
+    us = t0.elapsed_time(t1) * 100
+    wcl = w.contiguous(memory_format=torch.channels_last)
+    xref = torch.randn(n, c, hi, wi, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    def lib():
+        return torch.ops.aten.convolution_backward(
+            dy, xref, wcl, None, [2, 2], [1, 1], [1, 1], False, [0, 0], 1,
+            [True, False, False])[0]
+    for _ in range(3):
+        lib()
+    torch.cuda.synchronize()
+    t0.record()
+    for _ in range(10):
+        lib()
+    t1.record(); torch.cuda.synchronize()
+    us_m = t0.elapsed_time(t1) * 100
+    gf = 2 * n * k * c * 9 * ho * wo / 1e9
+    print(f"s2bwd N{n} K{k} {ho}x{wo}->C{c}: ours {us:7.1f}us ({gf/us*1e3:6.0f} GF/s) lib {us_m:7.1f}us ({gf/us_m*1e3:6.0f} GF/s)")
