@@ -874,15 +874,18 @@ __device__ __forceinline__ void s2_write(
 
 // Contraction for wave-quarter Q.  Fragment f's phase: quarters 0..1
 // own {(0,0),(1,1)}, quarters 2..3 own {(0,1),(1,0)}; h2 = Q&1.
-template <int Q>
+// KC = K/32 as a compile-time trip count (a runtime bound kept the loop
+// rolled -- conv_mm lesson); cc-parity-split accumulators double the
+// independent MFMA chains per wave (2 -> 4) to cover MFMA latency.
+template <int Q, int KC>
 __device__ __forceinline__ void s2_contract(
         const short* __restrict__ dy_s, const short* __restrict__ w_lds,
-        int K, int KS, int KS2, int LW, int wn, int row16, int slot8,
-        f32x4 (&acc)[2]) {
+        int KS, int KS2, int LW, int wn, int row16, int slot8,
+        f32x4 (&acc)[2][2]) {
     const int h2 = Q & 1;
     const int w2 = row16;            // Wo == 16: fragment row == w2
-    #pragma unroll 2
-    for (int cc = 0; cc < K / 32; ++cc) {
+    #pragma unroll
+    for (int cc = 0; cc < KC; ++cc) {
         #pragma unroll
         for (int f = 0; f < 2; ++f) {
             constexpr int PH0 = (Q < 2) ? 0 : 1;   // f == 0 phase
@@ -907,11 +910,28 @@ __device__ __forceinline__ void s2_contract(
                         *reinterpret_cast<const bf16x8*>(
                             &w_lds[((clocal * 9) + tau) * KS2
                                    + cc * 32 + slot8]);
-                    acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        afrag, bfrag, acc[f], 0, 0, 0);
+                    acc[f][cc & 1] =
+                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            afrag, bfrag, acc[f][cc & 1], 0, 0, 0);
                 }
             }
         }
+    }
+}
+
+template <int Q>
+__device__ __forceinline__ void s2_contract_k(
+        const short* __restrict__ dy_s, const short* __restrict__ w_lds,
+        int K, int KS, int KS2, int LW, int wn, int row16, int slot8,
+        f32x4 (&acc)[2][2]) {
+    switch (K / 32) {
+    case 2: s2_contract<Q, 2>(dy_s, w_lds, KS, KS2, LW, wn, row16,
+                              slot8, acc); break;
+    case 4: s2_contract<Q, 4>(dy_s, w_lds, KS, KS2, LW, wn, row16,
+                              slot8, acc); break;
+    case 8: s2_contract<Q, 8>(dy_s, w_lds, KS, KS2, LW, wn, row16,
+                              slot8, acc); break;
+    default: break;  // unreachable: supported() gates K to {64,128,256}
     }
 }
 
@@ -988,22 +1008,24 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd(
             }
         }
 
-        f32x4 acc[2];
+        f32x4 acc[2][2];
         #pragma unroll
         for (int f = 0; f < 2; ++f)
             #pragma unroll
-            for (int r = 0; r < 4; ++r)
-                acc[f][r] = 0.f;
+            for (int pp = 0; pp < 2; ++pp)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[f][pp][r] = 0.f;
 
         switch (wq) {
-        case 0: s2_contract<0>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
-                               slot8, acc); break;
-        case 1: s2_contract<1>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
-                               slot8, acc); break;
-        case 2: s2_contract<2>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
-                               slot8, acc); break;
-        default: s2_contract<3>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
-                                slot8, acc); break;
+        case 0: s2_contract_k<0>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                                 slot8, acc); break;
+        case 1: s2_contract_k<1>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                                 slot8, acc); break;
+        case 2: s2_contract_k<2>(dy_s, w_lds, K, KS, KS2, LW, wn, row16,
+                                 slot8, acc); break;
+        default: s2_contract_k<3>(dy_s, w_lds, K, KS, KS2, LW, wn,
+                                  row16, slot8, acc); break;
         }
 
         // ---- bounce the dx tile through the (now free) dy_s buffer so
@@ -1019,7 +1041,8 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd(
                 const int w2 = (lane >> 4) * 4 + r;
                 const int c = wn * 16 + (lane & 15);
                 const int s = (ph * S2_P2 + h2) * Wo + w2;
-                ystage[s * S2_CT + c] = f2b(acc[f][r]);
+                ystage[s * S2_CT + c] =
+                    f2b(acc[f][0][r] + acc[f][1][r]);
             }
         }
         __syncthreads();

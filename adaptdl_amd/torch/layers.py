@@ -247,6 +247,8 @@ class FusedConv2d(nn.Conv2d):
         return bool(ext.conv_wrw_ok(n, h, w, c, self.out_channels))
 
     def _s2_ok(self, x):
+        if os.getenv("ADAPTDL_S2_BWD") == "0":   # A/B escape hatch
+            return False
         if not (x.is_cuda and x.dtype == torch.bfloat16 and
                 x.dim() == 4 and ops.has_extension() and
                 self.bias is None and self.groups == 1 and
