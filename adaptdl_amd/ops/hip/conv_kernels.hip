@@ -1081,24 +1081,257 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd(
     }
 }
 
+// ---- Wo == 8 variant: one 16-pixel fragment is a WHOLE phase image
+// of the chunk (2 rows x 8), so waves split the K contraction in
+// quarters instead of the c tile, and the four partial dx tiles merge
+// through an f32 LDS stage (ds atomics) before the bf16 bounce store.
+// CT = 16 keeps the K=256 weight tile inside LDS (76 KB + 2x26 KB
+// double-buffered dy windows = 128 KB).
+
+#define S2W8_CT 16
+
+template <int G, int KQ, int KC>
+__device__ __forceinline__ void s2w8_contract(
+        const short* __restrict__ dy_s, const short* __restrict__ w_lds,
+        int KS, int KS2, int LW, int row16, int slot8,
+        f32x4 (&acc)[2][2]) {
+    const int h2 = row16 >> 3;
+    const int w2 = row16 & 7;
+    #pragma unroll
+    for (int cc = 0; cc < KC; ++cc) {
+        const int kbase = KQ * (KC * 32) + cc * 32;
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) {
+            constexpr int PH0 = (G == 0) ? 0 : 1;
+            constexpr int PH1 = (G == 0) ? 3 : 2;
+            const int ph = f ? PH1 : PH0;
+            const int a = ph >> 1, b = ph & 1;
+            #pragma unroll
+            for (int dh = 0; dh < 3; ++dh) {
+                if (a ? (dh == 1) : (dh != 1)) continue;
+                #pragma unroll
+                for (int dw = 0; dw < 3; ++dw) {
+                    if (b ? (dw == 1) : (dw != 1)) continue;
+                    const int j = h2 + (dh == 0 ? 1 : 0);
+                    const int col = w2 + (dw == 0 ? 1 : 0);
+                    const bf16x8 afrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &dy_s[(j * LW + col) * KS + kbase + slot8]);
+                    const int tau = dh * 3 + dw;
+                    const bf16x8 bfrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &w_lds[(row16 * 9 + tau) * KS2 + kbase
+                                   + slot8]);
+                    acc[f][cc & 1] =
+                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            afrag, bfrag, acc[f][cc & 1], 0, 0, 0);
+                }
+            }
+        }
+    }
+}
+
+template <int KC>
+__device__ __forceinline__ void s2w8_contract_g(
+        int wg, const short* __restrict__ dy_s,
+        const short* __restrict__ w_lds, int KS, int KS2, int LW,
+        int row16, int slot8, f32x4 (&acc)[2][2]) {
+    switch (wg) {
+    case 0: s2w8_contract<0, 0, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 1: s2w8_contract<0, 1, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 2: s2w8_contract<0, 2, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 3: s2w8_contract<0, 3, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 4: s2w8_contract<1, 0, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 5: s2w8_contract<1, 1, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    case 6: s2w8_contract<1, 2, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                    slot8, acc); break;
+    default: s2w8_contract<1, 3, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    }
+}
+
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd_w8(
+        const ushort_t* __restrict__ dy, const ushort_t* __restrict__ wt,
+        ushort_t* __restrict__ dx, int N, int Ho, int K, int C,
+        int spread, int dbuf) {
+    extern __shared__ short lds[];
+    const int Wo = 8;
+    const int KS = K + 16;
+    const int KS2 = K + 8;
+    const int LW = Wo + 8;
+    const int WIN = (S2_P2 + 1) * LW;
+    short* w_lds = lds;              // [S2W8_CT * 9][KS2]
+    short* dy_s0 = lds + S2W8_CT * 9 * KS2;
+    short* dy_s1 = dy_s0 + (dbuf ? WIN * KS : 0);
+
+    const int Hi = 2 * Ho, Wi = 2 * Wo;
+    const int ct = blockIdx.x / spread;
+    const int sp = blockIdx.x % spread;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;          // = 4*group + k-quarter
+    const int wg2 = wid >> 2;        // phase-pair group
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+
+    // CP2 = 4 phases * P2 * Wo = 64 chunk pixels
+    const int CP2 = 4 * S2_P2 * Wo;
+
+    {
+        const int pieces = S2W8_CT * 9 * (K / 8);
+        const ushort_t* wg = wt + (size_t)ct * S2W8_CT * 9 * K;
+        for (int i = t; i < pieces; i += 512) {
+            const int row = i / (K / 8);
+            const int kg = (i % (K / 8)) * 8;
+            *reinterpret_cast<u32x4*>(&w_lds[row * KS2 + kg]) =
+                *reinterpret_cast<const u32x4*>(
+                    wg + (size_t)row * K + kg);
+        }
+    }
+
+    const int rows_per_img = Ho / S2_P2;
+    const long chunks_total = (long)N * rows_per_img;
+
+    S2Regs regs;
+    if (sp < chunks_total) {
+        s2_issue(dy, (int)(sp / rows_per_img),
+                 (int)(sp % rows_per_img) * S2_P2, Ho, Wo, K, t, regs);
+        s2_write(dy_s0, Wo, K, KS, LW, t, regs);
+    }
+    __syncthreads();
+    if (sp + spread < chunks_total) {
+        const long qn = sp + spread;
+        s2_issue(dy, (int)(qn / rows_per_img),
+                 (int)(qn % rows_per_img) * S2_P2, Ho, Wo, K, t, regs);
+    }
+
+    int cur = 0;
+    for (long q = sp; q < chunks_total; q += spread) {
+        const int n = (int)(q / rows_per_img);
+        const int r0 = (int)(q % rows_per_img) * S2_P2;
+        short* dy_s = (dbuf && cur) ? dy_s1 : dy_s0;
+
+        if (!dbuf && q != sp) {
+            __syncthreads();
+            s2_write(dy_s0, Wo, K, KS, LW, t, regs);
+            __syncthreads();
+            if (q + spread < chunks_total) {
+                const long qn = q + spread;
+                s2_issue(dy, (int)(qn / rows_per_img),
+                         (int)(qn % rows_per_img) * S2_P2, Ho, Wo, K, t,
+                         regs);
+            }
+        }
+
+        f32x4 acc[2][2];
+        #pragma unroll
+        for (int f = 0; f < 2; ++f)
+            #pragma unroll
+            for (int pp = 0; pp < 2; ++pp)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[f][pp][r] = 0.f;
+
+        if (K / 32 == 8)
+            s2w8_contract_g<2>(wid & 7, dy_s, w_lds, KS, KS2, LW,
+                               row16, slot8, acc);
+        else
+            s2w8_contract_g<1>(wid & 7, dy_s, w_lds, KS, KS2, LW,
+                               row16, slot8, acc);
+
+        // ---- merge the 4 k-quarter partials in an f32 LDS stage ----
+        float* fstage = reinterpret_cast<float*>(dy_s);  // [64][CT]
+        __syncthreads();             // all waves done reading dy_s
+        for (int i = t; i < CP2 * S2W8_CT; i += 512)
+            fstage[i] = 0.f;
+        __syncthreads();
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) {
+            const int ph = (wg2 == 0) ? (f ? 3 : 0) : (f ? 2 : 1);
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = (lane >> 4) * 4 + r;
+                const int h2 = m >> 3, w2 = m & 7;
+                const int c = lane & 15;
+                const int s = (ph * S2_P2 + h2) * Wo + w2;
+                atomicAdd(&fstage[s * S2W8_CT + c],
+                          acc[f][0][r] + acc[f][1][r]);
+            }
+        }
+        __syncthreads();
+        {
+            // convert + interleaved contiguous store (8 c per piece)
+            const int pieces = 2 * S2_P2 * Wi * (S2W8_CT / 8);
+            const size_t base = ((size_t)n * Hi + 2 * r0) * Wi;
+            for (int i = t; i < pieces; i += 512) {
+                const int hl = i / (Wi * (S2W8_CT / 8));
+                const int rem = i % (Wi * (S2W8_CT / 8));
+                const int wi = rem / (S2W8_CT / 8);
+                const int cg = (rem % (S2W8_CT / 8)) * 8;
+                const int ph = ((hl & 1) << 1) | (wi & 1);
+                const int s = (ph * S2_P2 + (hl >> 1)) * Wo
+                    + (wi >> 1);
+                ushort_t out[8];
+                #pragma unroll
+                for (int jj = 0; jj < 8; ++jj)
+                    out[jj] = f2b(fstage[s * S2W8_CT + cg + jj]);
+                *reinterpret_cast<u32x4*>(
+                    dx + (base + (size_t)hl * Wi + wi) * C
+                    + ct * S2W8_CT + cg) =
+                    *reinterpret_cast<const u32x4*>(out);
+            }
+        }
+
+        if (dbuf) {
+            if (q + spread < chunks_total) {
+                short* nxt = cur ? dy_s0 : dy_s1;
+                s2_write(nxt, Wo, K, KS, LW, t, regs);
+                if (q + 2 * spread < chunks_total) {
+                    const long q2 = q + 2 * spread;
+                    s2_issue(dy, (int)(q2 / rows_per_img),
+                             (int)(q2 % rows_per_img) * S2_P2, Ho, Wo,
+                             K, t, regs);
+                }
+            }
+            __syncthreads();
+            cur ^= 1;
+        }
+    }
+}
+
 extern "C" int conv3x3_s2_bwd_supported(int Ho, int Wo, int K, int C) {
-    if (Wo != 16) return 0;
     if (Ho % S2_P2) return 0;
-    if (K % 32 || C % S2_CT) return 0;
-    const int pieces = (S2_P2 + 1) * Wo * (K / 8);
-    if (pieces > 1024) return 0;     // s2_issue register budget
-    // LDS: weight tile + at least one dy window must fit.
-    size_t need = ((size_t)S2_CT * 9 * (K + 8)
-                   + (size_t)(S2_P2 + 1) * (Wo + 8) * (K + 16)) * 2;
-    if (need > 160 * 1024) return 0;
-    return 1;
+    if (Wo == 16) {
+        // K must hit an instantiated contraction template exactly.
+        if ((K != 64 && K != 128 && K != 256) || C % S2_CT) return 0;
+        if ((S2_P2 + 1) * Wo * (K / 8) > 1024) return 0;  // reg budget
+        size_t need = ((size_t)S2_CT * 9 * (K + 8)
+                       + (size_t)(S2_P2 + 1) * (Wo + 8) * (K + 16)) * 2;
+        return need <= 160 * 1024;
+    }
+    if (Wo == 8) {
+        if ((K != 128 && K != 256) || C % S2W8_CT) return 0;
+        if ((S2_P2 + 1) * Wo * (K / 8) > 1024) return 0;
+        size_t need = ((size_t)S2W8_CT * 9 * (K + 8)
+                       + (size_t)(S2_P2 + 1) * (Wo + 8) * (K + 16)) * 2;
+        return need <= 160 * 1024;
+    }
+    return 0;
 }
 
 extern "C" void launch_conv3x3_s2_bwd(
         const ushort_t* dy, const ushort_t* wt, ushort_t* dx,
         int N, int Ho, int Wo, int K, int C, hipStream_t s) {
+    const int CT = (Wo == 8) ? S2W8_CT : S2_CT;
     const int KS = K + 16, KS2 = K + 8, LW = Wo + 8;
-    const size_t wbytes = (size_t)S2_CT * 9 * KS2 * 2;
+    const size_t wbytes = (size_t)CT * 9 * KS2 * 2;
     const size_t dybytes = (size_t)(S2_P2 + 1) * LW * KS * 2;
     int dbuf = 1;
     size_t lds_bytes = wbytes + 2 * dybytes;
@@ -1111,14 +1344,25 @@ extern "C" void launch_conv3x3_s2_bwd(
         hipFuncSetAttribute(
             reinterpret_cast<const void*>(&k_conv3x3_s2_bwd),
             hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_s2_bwd_w8),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
         attr_set = 1;
     }
     const long chunks = (long)N * (Ho / S2_P2);
-    const int nct = C / S2_CT;
+    const int nct = C / CT;
     long spread = 256 / nct;
     if (spread > chunks) spread = chunks;
     if (spread < 1) spread = 1;
-    hipLaunchKernelGGL(k_conv3x3_s2_bwd, dim3((unsigned)(nct * spread)),
-                       dim3(512), lds_bytes, s, dy, wt, dx, N, Ho, Wo,
-                       K, C, (int)spread, dbuf);
+    if (Wo == 8) {
+        hipLaunchKernelGGL(k_conv3x3_s2_bwd_w8,
+                           dim3((unsigned)(nct * spread)), dim3(512),
+                           lds_bytes, s, dy, wt, dx, N, Ho, K, C,
+                           (int)spread, dbuf);
+    } else {
+        hipLaunchKernelGGL(k_conv3x3_s2_bwd,
+                           dim3((unsigned)(nct * spread)), dim3(512),
+                           lds_bytes, s, dy, wt, dx, N, Ho, Wo, K, C,
+                           (int)spread, dbuf);
+    }
 }

@@ -156,6 +156,8 @@ def test_conv_mm_full_autograd_roundtrip(monkeypatch):
 @pytest.mark.parametrize("shape", [
     (2, 64, 16, 16, 128),     # layer2.0.conv1-like (dy side)
     (3, 32, 16, 16, 64),      # small channels
+    (2, 128, 8, 8, 256),      # layer3.0.conv1-like (Wo=8 k-split path)
+    (3, 32, 8, 8, 128),       # Wo=8, K=128
 ])
 def test_s2_bwd_kernel_matches_fp32(shape):
     """Polyphase stride-2 backward-data vs fp32 autograd reference."""
