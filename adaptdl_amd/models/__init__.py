@@ -7,3 +7,4 @@ from adaptdl_amd.models.bert import (BertConfig, BertModel,  # noqa: F401
                                      BertForMaskedLM, BertForPreTraining)
 from adaptdl_amd.models.ncf import NeuMF  # noqa: F401
 from adaptdl_amd.models.dcgan import Generator, Discriminator  # noqa: F401
+from adaptdl_amd.models.cifar import CIFAR_MODELS  # noqa: F401,E402

@@ -25,6 +25,7 @@ import torch.nn.functional as F
 import adaptdl_amd.env as env
 import adaptdl_amd.torch as adl
 from adaptdl_amd.models import ResNet18
+from adaptdl_amd.models.cifar import CIFAR_MODELS
 
 
 class SyntheticCifar(torch.utils.data.Dataset):
@@ -46,6 +47,9 @@ def main():
     parser.add_argument("--bs", type=int, default=128)
     parser.add_argument("--max-bs", type=int, default=4096)
     parser.add_argument("--lr", type=float, default=0.1)
+    parser.add_argument("--model", default="resnet18",
+                        choices=["resnet18"] + sorted(CIFAR_MODELS),
+                        help="architecture from the CIFAR model zoo")
     parser.add_argument("--synthetic", action="store_true", default=True)
     parser.add_argument("--samples", type=int, default=50000)
     args = parser.parse_args()
@@ -55,7 +59,9 @@ def main():
     device = torch.device("cuda" if use_gpu else "cpu")
 
     torch.manual_seed(1234)
-    model = ResNet18().to(device)
+    ctor = ResNet18 if args.model == "resnet18" \
+        else CIFAR_MODELS[args.model]
+    model = ctor().to(device)
     if use_gpu:
         model = model.to(memory_format=torch.channels_last)
     optim = adl.FusedSGD(model.parameters(), lr=args.lr,
