@@ -15,9 +15,15 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
 
 
 import argparse
+import faulthandler
 import math
+import signal
 
 import torch
+
+# Dump all thread stacks on SIGUSR1 (hung-collective debugging aid;
+# reference transformer.py:34-35 parity).
+faulthandler.register(signal.SIGUSR1)
 import torch.nn.functional as F
 
 import adaptdl_amd.env as env
