@@ -36,6 +36,10 @@ void launch_bn_bwd(const unsigned short*, const unsigned short*,
 int conv3x3_wrw_supported(int, int, int, int);
 int conv3x3_mm_supported(int, int, int, int);
 int conv3x3_s2_bwd_supported(int, int, int, int);
+int conv3x3_s2_fwd_supported(int, int, int, int);
+void launch_conv3x3_s2_fwd(const unsigned short*, const unsigned short*,
+                           unsigned short*, int, int, int, int, int,
+                           hipStream_t);
 void launch_conv3x3_s2_bwd(const unsigned short*, const unsigned short*,
                            unsigned short*, int, int, int, int, int,
                            hipStream_t);
@@ -342,6 +346,34 @@ void conv_s2_bwd(torch::Tensor dy, torch::Tensor wt, torch::Tensor dx) {
                           stream());
 }
 
+bool conv_s2_fwd_ok(long N, long H, long W, long C, long K) {
+    (void)N;
+    return conv3x3_s2_fwd_supported((int)H, (int)W, (int)C, (int)K) != 0;
+}
+
+// y = conv3x3_s2p1(x, w): x (N,C,H,W) channels_last bf16, w (K,C,3,3)
+// channels_last bf16, y (N,K,H/2,W/2) channels_last (experimental --
+// production forward stays on MIOpen pending a per-shape win).
+void conv_s2_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor y) {
+    check_bn_x(x, "x"); check_bn_x(y, "y");
+    const long N = x.size(0), C = x.size(1), H = x.size(2),
+        Wd = x.size(3);
+    const long K = y.size(1);
+    TORCH_CHECK(y.size(0) == N && y.size(2) == H / 2 &&
+                y.size(3) == Wd / 2, "y shape mismatch");
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
+                w.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                w.size(0) == K && w.size(1) == C && w.size(2) == 3 &&
+                w.size(3) == 3, "w must be (K,C,3,3) channels_last bf16");
+    TORCH_CHECK(conv_s2_fwd_ok(N, H, Wd, C, K),
+                "unsupported s2 fwd shape");
+    launch_conv3x3_s2_fwd((const unsigned short*)x.data_ptr(),
+                          (const unsigned short*)w.data_ptr(),
+                          (unsigned short*)y.data_ptr(),
+                          (int)N, (int)H, (int)Wd, (int)C, (int)K,
+                          stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -362,6 +394,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("conv_wrw", &conv_wrw, "MFMA 3x3 s1 NHWC weight gradient");
     mod.def("conv_mm_ok", &conv_mm_ok, "3x3 fwd/bwd-data predicate");
     mod.def("conv_mm", &conv_mm, "MFMA 3x3 s1 NHWC conv (fwd/bwd-data)");
+    mod.def("conv_s2_fwd_ok", &conv_s2_fwd_ok,
+            "3x3 stride-2 forward predicate (experimental)");
+    mod.def("conv_s2_fwd", &conv_s2_fwd,
+            "MFMA 3x3 s2 NHWC forward (experimental)");
     mod.def("conv_s2_bwd_ok", &conv_s2_bwd_ok,
             "3x3 stride-2 bwd-data predicate");
     mod.def("conv_s2_bwd", &conv_s2_bwd,

@@ -1306,6 +1306,278 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_bwd_w8(
     }
 }
 
+// =====================================================================
+// MFMA 3x3 STRIDE-2 FORWARD (NHWC bf16), experimental.
+//
+//   y[n,ho,wo,k] = sum_{dh,dw,c} x[n,2ho+dh-1,2wo+dw-1,c] * w[k,tau,c]
+//
+// Unlike backward-data, the forward is a pure gather: with pixel-major
+// LDS staging each tap is just a per-lane address offset.  The staging
+// splits every input row into even|odd COLUMN sections ("E|z|O", row
+// stride LS2 = 2*Wo+2, the z column holds the wo=0 left-pad zero), so
+// a fragment's 16 lanes step by ONE column slot = CS2 = C+8 shorts
+// (36 dwords at C=64: 16 distinct banks) instead of two raw columns
+// (2*(C+16) = 80 dwords: 4 banks, 4-way conflicts).  Tap decode:
+//   dw=1 -> E[wo];  dw=0 -> O[wo-1];  dw=2 -> O[wo]
+//   local row L = 2*(ho-r0) + dh  (L=0 is input row 2*r0-1; top halo
+//   only -- the bottom tap row 2*(r0+P2-1)+1 is always in range).
+// Weight tile w_lds[ko][tau][ci] and the wave layout mirror conv_mm;
+// chunk = P2F y-rows (CP = P2F*Wo = 64 pixels: 4 one-fragment pixel
+// quarters x 2 ko-halves of the KT=64 tile).  Single-buffered x window
+// with register prefetch (the w tile + two windows exceed 160 KB).
+// Constraints: Wo == 16 (fragment row == one y line), Ho % P2F == 0,
+// C % 32 == 0, C <= 128, K % 64 == 0.
+// =====================================================================
+
+#define S2F_P2 4
+
+struct S2FRegs {
+    V16 v[5];
+};
+
+__device__ __forceinline__ void s2f_issue(
+        const ushort_t* __restrict__ x, int n, int r0, int H, int W,
+        int C, int t, S2FRegs& r) {
+    // stage input rows 2*r0-1 .. 2*r0+2*P2F-1 (2*P2F+1 rows)
+    const int pieces = (2 * S2F_P2 + 1) * W * (C / 8);
+    #pragma unroll
+    for (int it = 0; it < 5; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (W * (C / 8));
+            const int rem = i % (W * (C / 8));
+            const int wcol = rem / (C / 8);
+            const int cg = (rem % (C / 8)) * 8;
+            const int h = 2 * r0 - 1 + j;
+            if (h < 0 || h >= H) {
+                r.v[it].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                r.v[it].u4 = *reinterpret_cast<const u32x4*>(
+                    x + (((size_t)n * H + h) * W + wcol) * C + cg);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void s2f_write(
+        short* __restrict__ x_s, int W, int C, int CS2, int LS2, int t,
+        S2FRegs& r) {
+    const int Wo = W / 2;
+    const int pieces = (2 * S2F_P2 + 1) * W * (C / 8);
+    #pragma unroll
+    for (int it = 0; it < 5; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (W * (C / 8));
+            const int rem = i % (W * (C / 8));
+            const int wcol = rem / (C / 8);
+            const int cg = (rem % (C / 8)) * 8;
+            // E section: slots 0..Wo-1; zero col at Wo; O: Wo+1..2Wo
+            const int slot = (wcol & 1) ? (Wo + 1 + (wcol >> 1))
+                                        : (wcol >> 1);
+            *reinterpret_cast<u32x4*>(
+                &x_s[(j * LS2 + slot) * CS2 + cg]) = r.v[it].u4;
+        }
+    }
+    // left-pad zero column (O[-1]) of every staged row
+    const int zp = (2 * S2F_P2 + 1) * (C / 8);
+    for (int i = t; i < zp; i += 512) {
+        const int j = i / (C / 8);
+        const int cg = (i % (C / 8)) * 8;
+        u32x4 z = {0u, 0u, 0u, 0u};
+        *reinterpret_cast<u32x4*>(
+            &x_s[(j * LS2 + Wo) * CS2 + cg]) = z;
+    }
+}
+
+template <int CBLK>
+__device__ __forceinline__ void s2f_contract(
+        const short* __restrict__ x_s, const short* __restrict__ w_lds,
+        int CS2, int WS, int LS2, int Wo, int wq, int wk, int row16,
+        int slot8, f32x4 (&acc)[2]) {
+    const int ho_loc = wq;           // quarter = one y line (Wo == 16)
+    const int wo = row16;
+    #pragma unroll
+    for (int cc = 0; cc < CBLK; ++cc) {
+        #pragma unroll
+        for (int dh = 0; dh < 3; ++dh) {
+            #pragma unroll
+            for (int dw = 0; dw < 3; ++dw) {
+                const int L = 2 * ho_loc + dh;
+                const int slot = (dw == 1) ? wo
+                    : (Wo + 1 + wo - (dw == 0 ? 1 : 0));
+                const bf16x8 afrag =
+                    *reinterpret_cast<const bf16x8*>(
+                        &x_s[(L * LS2 + slot) * CS2 + cc * 32
+                             + slot8]);
+                const int tau = dh * 3 + dw;
+                #pragma unroll
+                for (int nf = 0; nf < 2; ++nf) {
+                    const int ko = wk * 32 + nf * 16 + row16;
+                    const bf16x8 bfrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &w_lds[(ko * 9 + tau) * WS + cc * 32
+                                   + slot8]);
+                    acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        afrag, bfrag, acc[nf], 0, 0, 0);
+                }
+            }
+        }
+    }
+}
+
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_fwd(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
+        ushort_t* __restrict__ y, int N, int H, int W, int C, int K,
+        int spread) {
+    extern __shared__ short lds[];
+    const int KT = 64;
+    const int Wo = W / 2, Ho = H / 2;
+    const int CS2 = C + 8;
+    const int WS = C + 8;
+    const int LS2 = 2 * Wo + 2;
+    short* w_lds = lds;              // [KT * 9][WS]
+    short* x_s = lds + KT * 9 * WS;  // [(2*P2F+1) rows][LS2][CS2]
+
+    const int nk = K / KT;
+    const int kt = blockIdx.x / spread;
+    const int sp = blockIdx.x % spread;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wq = wid >> 1;         // y-line quarter of the chunk
+    const int wk = wid & 1;          // ko half of the KT tile
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+    const int CP = S2F_P2 * Wo;      // 64 chunk pixels
+
+    {
+        const int pieces = KT * 9 * (C / 8);
+        const ushort_t* wg = w + (size_t)kt * KT * 9 * C;
+        for (int i = t; i < pieces; i += 512) {
+            const int row = i / (C / 8);
+            const int cg = (i % (C / 8)) * 8;
+            *reinterpret_cast<u32x4*>(&w_lds[row * WS + cg]) =
+                *reinterpret_cast<const u32x4*>(
+                    wg + (size_t)row * C + cg);
+        }
+    }
+
+    const int rows_per_img = Ho / S2F_P2;
+    const long chunks_total = (long)N * rows_per_img;
+
+    S2FRegs regs;
+    if (sp < chunks_total) {
+        s2f_issue(x, (int)(sp / rows_per_img),
+                  (int)(sp % rows_per_img) * S2F_P2, H, W, C, t, regs);
+        s2f_write(x_s, W, C, CS2, LS2, t, regs);
+    }
+    __syncthreads();
+    if (sp + spread < chunks_total) {
+        const long qn = sp + spread;
+        s2f_issue(x, (int)(qn / rows_per_img),
+                  (int)(qn % rows_per_img) * S2F_P2, H, W, C, t, regs);
+    }
+
+    for (long q = sp; q < chunks_total; q += spread) {
+        const int n = (int)(q / rows_per_img);
+        const int r0 = (int)(q % rows_per_img) * S2F_P2;
+
+        if (q != sp) {
+            __syncthreads();         // y bounce of q-1 fully read
+            s2f_write(x_s, W, C, CS2, LS2, t, regs);
+            __syncthreads();
+            if (q + spread < chunks_total) {
+                const long qn = q + spread;
+                s2f_issue(x, (int)(qn / rows_per_img),
+                          (int)(qn % rows_per_img) * S2F_P2, H, W, C, t,
+                          regs);
+            }
+        }
+
+        f32x4 acc[2];
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+            #pragma unroll
+            for (int r = 0; r < 4; ++r)
+                acc[nf][r] = 0.f;
+
+        switch (C / 32) {
+        case 2: s2f_contract<2>(x_s, w_lds, CS2, WS, LS2, Wo, wq, wk,
+                                row16, slot8, acc); break;
+        case 3: s2f_contract<3>(x_s, w_lds, CS2, WS, LS2, Wo, wq, wk,
+                                row16, slot8, acc); break;
+        default: s2f_contract<4>(x_s, w_lds, CS2, WS, LS2, Wo, wq, wk,
+                                 row16, slot8, acc); break;
+        }
+
+        // bounce y tile through x_s: [64 px][KT] bf16 = 8 KB
+        ushort_t* ystage = reinterpret_cast<ushort_t*>(x_s);
+        __syncthreads();
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int wo2 = (lane >> 4) * 4 + r;
+                const int ko = wk * 32 + nf * 16 + (lane & 15);
+                ystage[(wq * Wo + wo2) * KT + ko] = (short)f2b(acc[nf][r]);
+            }
+        }
+        __syncthreads();
+        {
+            const int pieces = CP * (KT / 8);
+            const size_t base = ((size_t)n * Ho + r0) * Wo;
+            for (int i = t; i < pieces; i += 512) {
+                const int p = i / (KT / 8);
+                const int kg = (i % (KT / 8)) * 8;
+                *reinterpret_cast<u32x4*>(
+                    y + (base + p) * K + kt * KT + kg) =
+                    *reinterpret_cast<const u32x4*>(
+                        &ystage[p * KT + kg]);
+            }
+        }
+    }
+}
+
+extern "C" int conv3x3_s2_fwd_supported(int H, int W, int C, int K) {
+    if (W != 32) return 0;           // Wo == 16
+    const int Ho = H / 2;
+    if (H % 2 || Ho % S2F_P2) return 0;
+    if (C % 32 || C > 128 || K % 64) return 0;
+    const int pieces = (2 * S2F_P2 + 1) * W * (C / 8);
+    if (pieces > 2560) return 0;     // s2f_issue register budget
+    size_t need = ((size_t)64 * 9 * (C + 8)
+                   + (size_t)(2 * S2F_P2 + 1) * (2 * (W / 2) + 2)
+                     * (C + 8)) * 2;
+    return need <= 160 * 1024;
+}
+
+extern "C" void launch_conv3x3_s2_fwd(
+        const ushort_t* x, const ushort_t* w, ushort_t* y,
+        int N, int H, int W, int C, int K, hipStream_t s) {
+    const int KT = 64;
+    const int LS2 = 2 * (W / 2) + 2;
+    const size_t lds_bytes = ((size_t)KT * 9 * (C + 8)
+                              + (size_t)(2 * S2F_P2 + 1) * LS2
+                                * (C + 8)) * 2;
+    static int attr_set = 0;
+    if (!attr_set) {
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_s2_fwd),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = 1;
+    }
+    const long chunks = (long)N * ((H / 2) / S2F_P2);
+    const int nk = K / KT;
+    long spread = 256 / nk;
+    if (spread > chunks) spread = chunks;
+    if (spread < 1) spread = 1;
+    hipLaunchKernelGGL(k_conv3x3_s2_fwd, dim3((unsigned)(nk * spread)),
+                       dim3(512), lds_bytes, s, x, w, y, N, H, W, C, K,
+                       (int)spread);
+}
+
 extern "C" int conv3x3_s2_bwd_supported(int Ho, int Wo, int K, int C) {
     if (Ho % S2_P2) return 0;
     if (Wo == 16) {

@@ -206,3 +206,24 @@ def test_s2_conv_module_backward():
     y32 = F.conv2d(x32, w32, stride=2, padding=1)
     y32.backward(dy.float())
     assert torch.allclose(x.grad.float(), x32.grad, atol=0.1, rtol=5e-2)
+
+
+@pytest.mark.gpu
+def test_s2_fwd_kernel_matches_fp32():
+    """Experimental polyphase stride-2 FORWARD vs fp32 reference."""
+    from adaptdl_amd import ops
+    ext = ops._load_extension()
+    torch.manual_seed(7)
+    n, c, h, w_, k = 3, 64, 32, 32, 128
+    dev = torch.device("cuda")
+    x = (torch.randn(n, c, h, w_, device=dev) * 0.5).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(k, c, 3, 3, device=dev) * 0.2).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    assert ext.conv_s2_fwd_ok(n, h, w_, c, k)
+    y = torch.empty(n, k, h // 2, w_ // 2, dtype=torch.bfloat16,
+                    device=dev).contiguous(memory_format=torch.channels_last)
+    ext.conv_s2_fwd(x, w, y)
+    ref = F.conv2d(x.float(), w.float(), stride=2, padding=1)
+    assert torch.allclose(y.float(), ref, atol=0.1, rtol=5e-2), \
+        (y.float() - ref).abs().max().item()
