@@ -17,8 +17,7 @@ class FakeWriter:
 
 
 def test_adp_and_loader_exporters(tmp_ckpt_env):
-    if not collective.initialized():
-        collective.initialize(master_addr="127.0.0.1")
+    collective.initialize(master_addr="127.0.0.1")
     model = torch.nn.Linear(4, 2)
     optim = torch.optim.SGD(model.parameters(), lr=0.1)
     adp = adl.AdaptiveDataParallel(model, optim, name="tb-test")
@@ -42,3 +41,4 @@ def test_adp_and_loader_exporters(tmp_ckpt_env):
     assert any("Batch" in t for t in tags), tags
     assert all(torch.isfinite(torch.tensor(v)) for v, _ in
                w.scalars.values())
+    collective.teardown()   # forked tests re-initialize their own
