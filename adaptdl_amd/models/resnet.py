@@ -33,8 +33,8 @@ class BasicBlock(nn.Module):
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
-                nn.Conv2d(in_planes, self.expansion * planes, 1,
-                          stride=stride, bias=False),
+                FusedConv2d(in_planes, self.expansion * planes, 1,
+                            stride=stride, bias=False),
                 _bn(self.expansion * planes))
 
     def forward(self, x):
@@ -59,8 +59,8 @@ class Bottleneck(nn.Module):
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != self.expansion * planes:
             self.shortcut = nn.Sequential(
-                nn.Conv2d(in_planes, self.expansion * planes, 1,
-                          stride=stride, bias=False),
+                FusedConv2d(in_planes, self.expansion * planes, 1,
+                            stride=stride, bias=False),
                 _bn(self.expansion * planes))
 
     def forward(self, x):

@@ -231,6 +231,50 @@ class _S2ConvFunction(torch.autograd.Function):
         return dx, dw, None, None, None, None
 
 
+class _S2Conv1x1Function(torch.autograd.Function):
+    """1x1 stride-2 conv (ResNet downsample shortcut) as plain GEMMs.
+
+    The strided 1x1 conv touches only the even-index pixels, so all
+    three directions are matmuls over the subsampled image (hipBLASLt on
+    MI355X) instead of MIOpen/CK's strided conv kernels:
+        y            = x[:, ::2, ::2] @ W^T
+        dx[::2,::2]  = dy @ W          (zeros elsewhere)
+        dW           = dy^T @ x[:, ::2, ::2]
+    Semantics are identical on CPU, so the path is CPU-testable;
+    measured-perf gating happens at the caller.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight):
+        wb = weight.detach().to(x.dtype)
+        xs = x[:, :, ::2, ::2]                      # (N, C, Ho, Wo)
+        xs_nhwc = xs.permute(0, 2, 3, 1)
+        y = (xs_nhwc @ wb.view(wb.shape[0], wb.shape[1]).t()) \
+            .permute(0, 3, 1, 2) \
+            .contiguous(memory_format=torch.channels_last)
+        ctx.save_for_backward(x, wb)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wb = ctx.saved_tensors
+        k, c = wb.shape[0], wb.shape[1]
+        w2d = wb.view(k, c)
+        dy_nhwc = dy.permute(0, 2, 3, 1)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.zeros_like(x)
+            dx[:, :, ::2, ::2] = (dy_nhwc @ w2d.to(dy.dtype)) \
+                .permute(0, 3, 1, 2)
+        if ctx.needs_input_grad[1]:
+            # bf16 GEMM (fp32 accumulation inside), fp32 master grad out
+            xs_nhwc = x[:, :, ::2, ::2].permute(0, 2, 3, 1)
+            dw = (dy_nhwc.reshape(-1, k).t()
+                  @ xs_nhwc.reshape(-1, c).to(dy.dtype)) \
+                .float().view(k, c, 1, 1)
+        return dx, dw
+
+
 class FusedConv2d(nn.Conv2d):
     """Conv2d whose 3x3/s1 weight gradient runs the MFMA wrw kernel."""
 
@@ -270,6 +314,19 @@ class FusedConv2d(nn.Conv2d):
         return bool(ext.conv_s2_bwd_ok(n, h // 2, w // 2,
                                        self.out_channels, c))
 
+    def _s2_1x1_ok(self, x):
+        # GEMM path for the 1x1 stride-2 downsample shortcut.  Gated
+        # off by default pending an in-context A/B next round (the
+        # Wo=8 s2_bwd lesson: never default-on an unmeasured path).
+        if os.getenv("ADAPTDL_S2_1X1") != "1":
+            return False
+        return (x.is_cuda and x.dtype == torch.bfloat16 and
+                x.dim() == 4 and self.bias is None and
+                self.groups == 1 and self.kernel_size == (1, 1) and
+                self.stride == (2, 2) and self.padding == (0, 0) and
+                x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0 and
+                x.is_contiguous(memory_format=torch.channels_last))
+
     def forward(self, x):
         if torch.is_autocast_enabled() and x.is_cuda and \
                 x.dtype != torch.bfloat16 and \
@@ -284,4 +341,6 @@ class FusedConv2d(nn.Conv2d):
             return _S2ConvFunction.apply(x, self.weight, self.stride,
                                          self.padding, self.dilation,
                                          self.groups)
+        if self._s2_1x1_ok(x):
+            return _S2Conv1x1Function.apply(x, self.weight)
         return super().forward(x)

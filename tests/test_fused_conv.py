@@ -238,3 +238,25 @@ def test_s2_fwd_kernel_matches_fp32():
     ref = F.conv2d(x.float(), w.float(), stride=2, padding=1)
     assert torch.allclose(y.float(), ref, atol=0.1, rtol=5e-2), \
         (y.float() - ref).abs().max().item()
+
+
+def test_s2_1x1_gemm_function_matches_autograd():
+    """1x1 stride-2 conv as GEMMs (_S2Conv1x1Function) vs conv2d
+    autograd, on CPU fp32 (identical semantics on GPU bf16)."""
+    from adaptdl_amd.torch.layers import _S2Conv1x1Function
+    torch.manual_seed(8)
+    n, c, h, w_, k = 3, 16, 8, 8, 32
+    x = torch.randn(n, c, h, w_, requires_grad=True)
+    weight = torch.randn(k, c, 1, 1, requires_grad=True)
+    y = _S2Conv1x1Function.apply(x, weight)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = weight.detach().clone().requires_grad_(True)
+    y2 = F.conv2d(x2, w2, stride=2)
+    y2.backward(dy)
+
+    assert torch.allclose(y, y2, atol=1e-4)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4)
+    assert torch.allclose(weight.grad, w2.grad, atol=1e-3)
