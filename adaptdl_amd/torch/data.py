@@ -63,9 +63,11 @@ class ElasticSampler(Sampler):
             indices = list(range(len(self.dataset)))
         base_index = self.index % len(self.dataset)
         local_indices = indices[base_index + self.rank::self.num_replicas]
-        # Pad so every replica yields the same number of samples.
+        # Pad so every replica yields the same number of samples.  The
+        # modulo covers num_replicas > len(dataset) (a rank with no
+        # samples of its own still yields one padding sample).
         if len(local_indices) < len(self):
-            local_indices.append(indices[self.rank])
+            local_indices.append(indices[self.rank % len(indices)])
         assert len(local_indices) == len(self)
         return iter(local_indices)
 
