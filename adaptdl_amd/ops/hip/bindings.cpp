@@ -37,6 +37,11 @@ int conv3x3_wrw_supported(int, int, int, int);
 int conv3x3_mm_supported(int, int, int, int);
 int conv3x3_s2_bwd_supported(int, int, int, int);
 int conv3x3_s2_fwd_supported(int, int, int, int);
+int conv3x3_s2_wrw_supported(int, int, int, int);
+int conv3x3_s2_wrw_nsplit(int, int, int, int, int);
+void launch_conv3x3_s2_wrw(const unsigned short*, const unsigned short*,
+                           float*, float*, int, int, int, int, int,
+                           hipStream_t);
 void launch_conv3x3_s2_fwd(const unsigned short*, const unsigned short*,
                            unsigned short*, int, int, int, int, int,
                            hipStream_t);
@@ -374,6 +379,44 @@ void conv_s2_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor y) {
                           stream());
 }
 
+bool conv_s2_wrw_ok(long N, long Ho, long Wo, long C, long K) {
+    (void)N;
+    return conv3x3_s2_wrw_supported((int)Ho, (int)Wo, (int)C, (int)K)
+        != 0;
+}
+
+long conv_s2_wrw_nsplit(long N, long Ho, long Wo, long C, long K) {
+    return conv3x3_s2_wrw_nsplit((int)N, (int)Ho, (int)Wo, (int)C,
+                                 (int)K);
+}
+
+// dW = wrw(x, dy) for the stride-2 3x3 conv: x (N,C,2Ho,2Wo)
+// channels_last bf16, dy (N,K,Ho,Wo) channels_last bf16, ws fp32
+// workspace (nsplit*K*9*C), dw (K,C,3,3) channels_last fp32 output.
+void conv_s2_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor ws,
+                 torch::Tensor dw) {
+    check_bn_x(x, "x"); check_bn_x(dy, "dy");
+    const long N = dy.size(0), K = dy.size(1), Ho = dy.size(2),
+        Wo = dy.size(3);
+    const long C = x.size(1);
+    TORCH_CHECK(x.size(0) == N && x.size(2) == 2 * Ho &&
+                x.size(3) == 2 * Wo, "x shape mismatch");
+    TORCH_CHECK(conv_s2_wrw_ok(N, Ho, Wo, C, K),
+                "unsupported s2 wrw shape");
+    const long nsplit = conv_s2_wrw_nsplit(N, Ho, Wo, C, K);
+    TORCH_CHECK(ws.is_cuda() && ws.scalar_type() == torch::kFloat32 &&
+                ws.numel() >= nsplit * K * 9 * C, "ws too small");
+    TORCH_CHECK(dw.is_cuda() && dw.scalar_type() == torch::kFloat32 &&
+                dw.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                dw.size(0) == K && dw.size(1) == C, "dw must be "
+                "(K,C,3,3) channels_last fp32");
+    launch_conv3x3_s2_wrw((const unsigned short*)x.data_ptr(),
+                          (const unsigned short*)dy.data_ptr(),
+                          ws.data_ptr<float>(), dw.data_ptr<float>(),
+                          (int)N, (int)Ho, (int)Wo, (int)C, (int)K,
+                          stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -398,6 +441,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             "3x3 stride-2 forward predicate (experimental)");
     mod.def("conv_s2_fwd", &conv_s2_fwd,
             "MFMA 3x3 s2 NHWC forward (experimental)");
+    mod.def("conv_s2_wrw_ok", &conv_s2_wrw_ok,
+            "3x3 stride-2 wrw predicate (experimental)");
+    mod.def("conv_s2_wrw_nsplit", &conv_s2_wrw_nsplit,
+            "s2 wrw workspace splits");
+    mod.def("conv_s2_wrw", &conv_s2_wrw,
+            "MFMA 3x3 s2 NHWC weight gradient (experimental)");
     mod.def("conv_s2_bwd_ok", &conv_s2_bwd_ok,
             "3x3 stride-2 bwd-data predicate");
     mod.def("conv_s2_bwd", &conv_s2_bwd,

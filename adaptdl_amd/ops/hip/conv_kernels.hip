@@ -1638,3 +1638,398 @@ extern "C" void launch_conv3x3_s2_bwd(
                            (int)spread, dbuf);
     }
 }
+
+// =====================================================================
+// MFMA 3x3 STRIDE-2 weight gradient (NHWC bf16), experimental.
+//
+//   dW[k][dh][dw][c] = sum_{n,ho,wo} dy[n,ho,wo,k]
+//                      * x[n, 2*ho+dh-1, 2*wo+dw-1, c]
+//
+// Same implicit-GEMM structure as the stride-1 wrw kernel (transposed
+// dy_t/x_t staging, 36 accumulator fragments per wave, 5/4 tau-split
+// wave pairs, split-K fp32 slabs + the SAME k_wrw_reduce), with the
+// x side staged POLYPHASE: each of the 2P+1 input lines of a chunk is
+// stored as [E: even cols (Wo)][4 zero pads][O: odd cols (Wo)][4 pads]
+// (row stride LS2 = 2*Wo+8), so the dw=1 tap is one aligned b128 read
+// (E) and dw=0/2 come from one aligned 16-value O window via the
+// stride-1 kernel's byte-rotation extracts (dw=0 -> shift 6, dw=2 ->
+// shift 8).  x global loads pair SAME-PARITY pixels (w, w+2) so the
+// packed b32 transpose writes stay adjacent in E/O.  The x_t stride is
+// runtime (content (2P+1)*LS2 > the stride-1 search's 288): the
+// launcher picks a multiple of 8 whose dword stride has gcd 4 with the
+// 64 LDS banks (16 distinct banks per 16-lane fragment group); no XOR
+// swizzle in v1 (measure first).  dy_t reuses DY_STRIDE + DSWZ as-is.
+// Constraints: dy Wo in {8, 16}, Ho % 4 == 0, C % 64, K % 64.
+// Experimental: gated off until an in-context A/B vs MIOpen's s2 wrw.
+// =====================================================================
+
+#define S2W_P 4
+
+struct S2WRegs {
+    V16 vdy[1][2];
+    V16 vx[3][2];
+};
+
+__device__ __forceinline__ void s2w_issue(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+        int n, int lo0, int Ho, int Wo, int C, int K, int kt, int ct,
+        int CP, int t, S2WRegs& r) {
+    const int dyn8 = CP * 4;         // pixel-pairs x 8 k-groups
+    #pragma unroll
+    for (int it = 0; it < 1; ++it) {
+        const int i = t + it * 512;
+        if (i < dyn8) {
+            const int p = (i >> 3) * 2;
+            const ushort_t* g = dy +
+                (((size_t)n * Ho + (lo0 + p / Wo)) * Wo + p % Wo) * K
+                + (size_t)kt * 64 + (i & 7) * 8;
+            r.vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+            r.vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
+        }
+    }
+    const int Hi = 2 * Ho, Wi = 2 * Wo;
+    const int xn8 = (2 * S2W_P + 1) * (Wi / 2) * 8;
+    #pragma unroll
+    for (int it = 0; it < 3; ++it) {
+        const int i = t + it * 512;
+        if (i < xn8) {
+            const int j = i / ((Wi / 2) * 8);
+            const int rem = i % ((Wi / 2) * 8);
+            const int pr = rem >> 3;
+            const int wbase = (pr & 1) + 4 * (pr >> 1);
+            const int h = 2 * lo0 - 1 + j;
+            if (h < 0 || h >= Hi) {
+                r.vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
+                r.vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                const ushort_t* g = x +
+                    (((size_t)n * Hi + h) * Wi + wbase) * C
+                    + (size_t)ct * 64 + (rem & 7) * 8;
+                r.vx[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+                r.vx[it][1].u4 =
+                    *reinterpret_cast<const u32x4*>(g + 2 * C);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void s2w_write(
+        short* __restrict__ dy_t, short* __restrict__ x_t,
+        int Wo, int CP, int XT2S, int LS2, int t, S2WRegs& r) {
+    const int dyn8 = CP * 4;
+    #pragma unroll
+    for (int it = 0; it < 1; ++it) {
+        const int i = t + it * 512;
+        if (i < dyn8) {
+            const int p = (i >> 3) * 2;
+            const int kg = (i & 7) * 8;
+            #pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                const unsigned int packed =
+                    (unsigned int)(unsigned short)r.vdy[it][0].s[j]
+                    | ((unsigned int)(unsigned short)
+                       r.vdy[it][1].s[j] << 16);
+                *reinterpret_cast<unsigned int*>(
+                    &dy_t[(kg + j) * DY_STRIDE
+                          + DSWZ(kg + j, p)]) = packed;
+            }
+        }
+    }
+    const int Wi = 2 * Wo;
+    const int xn8 = (2 * S2W_P + 1) * (Wi / 2) * 8;
+    #pragma unroll
+    for (int it = 0; it < 3; ++it) {
+        const int i = t + it * 512;
+        if (i < xn8) {
+            const int j = i / ((Wi / 2) * 8);
+            const int rem = i % ((Wi / 2) * 8);
+            const int pr = rem >> 3;
+            const int cg = (rem & 7) * 8;
+            const int wbase = (pr & 1) + 4 * (pr >> 1);
+            // same-parity pair (wbase, wbase+2) -> adjacent E/O slots
+            const int col = (wbase & 1)
+                ? (Wo + 4 + ((wbase - 1) >> 1))   // O section
+                : (wbase >> 1);                   // E section
+            #pragma unroll
+            for (int jj = 0; jj < 8; ++jj) {
+                const unsigned int packed =
+                    (unsigned int)(unsigned short)r.vx[it][0].s[jj]
+                    | ((unsigned int)(unsigned short)
+                       r.vx[it][1].s[jj] << 16);
+                *reinterpret_cast<unsigned int*>(
+                    &x_t[(cg + jj) * XT2S + j * LS2 + col]) = packed;
+            }
+        }
+    }
+}
+
+template <int WT2>
+__device__ __forceinline__ void s2w_mfma_phase(
+        const short* __restrict__ dy_t, const short* __restrict__ x_t,
+        int kchunks, int Wo, int XT2S, int LS2, int wk, int wc,
+        int row16, int slot8, f32x4 (&acc)[2][2][5]) {
+    for (int kc = 0; kc < kchunks; ++kc) {
+        const int p0 = kc * 32 + slot8;
+        const int li = p0 / Wo;
+        const int w0 = p0 % Wo;
+        bf16x8 afrag[2];
+        #pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+            const int row = wk * 32 + mf * 16 + row16;
+            afrag[mf] = *reinterpret_cast<const bf16x8*>(
+                &dy_t[row * DY_STRIDE + DSWZ(row, p0)]);
+        }
+        int ti = 0;
+        #pragma unroll
+        for (int g = 0; g < 2; ++g) {
+            constexpr int dh0 = WT2 ? 1 : 0;
+            constexpr int dh1 = WT2 ? 2 : 1;
+            const int dh = g ? dh1 : dh0;
+            constexpr int lo0 = WT2 ? 2 : 0;
+            constexpr int hi1 = WT2 ? 3 : 2;
+            const int lo = g ? 0 : lo0;
+            const int hi = g ? hi1 : 3;
+            const int slot = 2 * li + dh;    // input line slot (halo 0)
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf) {
+                const int crow = wc * 32 + nf * 16 + row16;
+                const short* xrow = &x_t[crow * XT2S + slot * LS2];
+                // E: the dw=1 tap, one aligned b128
+                bf16x8 efrag;
+                if (lo <= 1 && 1 < hi)
+                    efrag = *reinterpret_cast<const bf16x8*>(
+                        &xrow[w0]);
+                // O window: taps dw=0 (shift 6) and dw=2 (shift 8)
+                V16 vlo, vhi;
+                if (lo < 1 || hi > 2) {
+                    vlo.u4 = *reinterpret_cast<const u32x4*>(
+                        &xrow[Wo + w0]);
+                    vhi.u4 = *reinterpret_cast<const u32x4*>(
+                        &xrow[Wo + w0 + 8]);
+                }
+                unsigned int win[8] = {vlo.u[0], vlo.u[1], vlo.u[2],
+                                       vlo.u[3], vhi.u[0], vhi.u[1],
+                                       vhi.u[2], vhi.u[3]};
+                int tj = ti;
+                #pragma unroll
+                for (int dw = 0; dw < 3; ++dw) {
+                    if (dw < lo || dw >= hi) continue;
+                    bf16x8 bfrag;
+                    if (dw == 1) {
+                        bfrag = efrag;
+                    } else {
+                        u32x4 frag;
+                        const int sh = (dw == 0) ? 6 : 8;
+                        const int d0 = sh >> 2;
+                        const int rem2 = sh & 3;
+                        if (rem2 == 0) {
+                            frag[0] = win[d0]; frag[1] = win[d0 + 1];
+                            frag[2] = win[d0 + 2];
+                            frag[3] = win[d0 + 3];
+                        } else {
+                            frag[0] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 1], win[d0], rem2);
+                            frag[1] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 2], win[d0 + 1], rem2);
+                            frag[2] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 3], win[d0 + 2], rem2);
+                            frag[3] = __builtin_amdgcn_alignbyte(
+                                win[d0 + 4], win[d0 + 3], rem2);
+                        }
+                        bfrag = *reinterpret_cast<const bf16x8*>(&frag);
+                    }
+                    #pragma unroll
+                    for (int mf = 0; mf < 2; ++mf)
+                        acc[mf][nf][tj] =
+                            __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                                afrag[mf], bfrag, acc[mf][nf][tj],
+                                0, 0, 0);
+                    ++tj;
+                }
+            }
+            ti += (g ? hi1 : 3) - (g ? 0 : lo0);
+        }
+    }
+}
+
+extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_s2_wrw(
+        const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
+        float* __restrict__ ws, int N, int Ho, int Wo, int C, int K,
+        int XT2S, int nsplit) {
+    extern __shared__ short lds[];
+    const int LS2 = 2 * Wo + 8;
+    const int CP = S2W_P * Wo;
+    const int kchunks = CP / 32;
+    const int BUF = 64 * DY_STRIDE + 64 * XT2S;
+    const int nc = C / 64;
+    const int tile = blockIdx.x / nsplit;
+    const int split = blockIdx.x % nsplit;
+    const int kt = tile / nc;
+    const int ct = tile % nc;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wk = wid >> 2;
+    const int wc = (wid >> 1) & 1;
+    const int wt2 = wid & 1;
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+
+    f32x4 acc[2][2][5];
+    #pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+        #pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+            #pragma unroll
+            for (int ti = 0; ti < 5; ++ti)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[mf][nf][ti][r] = 0.f;
+
+    const int rows_per_img = Ho / S2W_P;
+    const long chunks_total = (long)N * rows_per_img;
+
+    // Zero the O-section front pads (cols Wo..Wo+3: the j=-1 slot read
+    // by the dw=0 tap at w0=0) of both x_t buffers once.
+    for (int b = 0; b < 2; ++b) {
+        short* x_t = lds + b * BUF + 64 * DY_STRIDE;
+        for (int i = t; i < 64 * (2 * S2W_P + 1) * 2; i += 512) {
+            const int cc = i / ((2 * S2W_P + 1) * 2);
+            const int rem = i % ((2 * S2W_P + 1) * 2);
+            const int j = rem >> 1;
+            const int half = (rem & 1) * 2;
+            *reinterpret_cast<unsigned int*>(
+                &x_t[cc * XT2S + j * LS2 + Wo + half]) = 0u;
+        }
+    }
+
+    S2WRegs regs;
+    #define S2W_NH(qq) \
+        const int n_ = (int)((qq) / rows_per_img); \
+        const int lo_ = (int)((qq) % rows_per_img) * S2W_P;
+
+    if (split < chunks_total) {
+        S2W_NH(split)
+        s2w_issue(x, dy, n_, lo_, Ho, Wo, C, K, kt, ct, CP, t, regs);
+        s2w_write(lds, lds + 64 * DY_STRIDE, Wo, CP, XT2S, LS2, t, regs);
+    }
+    __syncthreads();
+    if (split + nsplit < chunks_total) {
+        S2W_NH(split + nsplit)
+        s2w_issue(x, dy, n_, lo_, Ho, Wo, C, K, kt, ct, CP, t, regs);
+    }
+
+    int cur = 0;
+    for (long q = split; q < chunks_total; q += nsplit, cur ^= 1) {
+        const short* dy_t = lds + cur * BUF;
+        const short* x_t = dy_t + 64 * DY_STRIDE;
+
+        if (wt2 == 0)
+            s2w_mfma_phase<0>(dy_t, x_t, kchunks, Wo, XT2S, LS2, wk, wc,
+                              row16, slot8, acc);
+        else
+            s2w_mfma_phase<1>(dy_t, x_t, kchunks, Wo, XT2S, LS2, wk, wc,
+                              row16, slot8, acc);
+
+        if (q + nsplit < chunks_total) {
+            short* ndy = lds + (cur ^ 1) * BUF;
+            s2w_write(ndy, ndy + 64 * DY_STRIDE, Wo, CP, XT2S, LS2, t,
+                      regs);
+            if (q + 2 * nsplit < chunks_total) {
+                S2W_NH(q + 2 * nsplit)
+                s2w_issue(x, dy, n_, lo_, Ho, Wo, C, K, kt, ct, CP, t,
+                          regs);
+            }
+        }
+        __syncthreads();
+    }
+    #undef S2W_NH
+
+    // fp32 partial slab (same layout as the stride-1 wrw epilogue)
+    float* slab = ws + (size_t)split * K * 9 * C;
+    #pragma unroll
+    for (int ti = 0; ti < 5; ++ti) {
+        if (wt2 && ti >= 4)
+            continue;
+        const int tau = (wt2 ? 5 : 0) + ti;
+        #pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+            #pragma unroll
+            for (int nf = 0; nf < 2; ++nf) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int klocal = wk * 32 + mf * 16
+                        + ((lane >> 4) * 4 + r);
+                    const int clocal = wc * 32 + nf * 16
+                        + (lane & 15);
+                    slab[((size_t)(kt * 64 + klocal) * 9 + tau) * C
+                         + ct * 64 + clocal] = acc[mf][nf][ti][r];
+                }
+            }
+        }
+    }
+}
+
+static int s2w_xt_stride(int Wo) {
+    // content (2P+1)*LS2; stride mult of 8 whose dword count has
+    // gcd(., 64) == 4 -> 16 distinct banks across a 16-lane fragment
+    const int content = (2 * S2W_P + 1) * (2 * Wo + 8);
+    int s = (content + 7) / 8 * 8;
+    while ((s / 2) % 4 != 0 || ((s / 2) % 64) % 8 != 4) {
+        s += 8;
+        if (s > content + 128) break;   // give up: take mult-of-8
+    }
+    return s;
+}
+
+extern "C" int conv3x3_s2_wrw_supported(int Ho, int Wo, int C, int K) {
+    if (Wo != 8 && Wo != 16) return 0;
+    if (Ho % S2W_P) return 0;
+    if (C % 64 || K % 64) return 0;
+    const int XT2S = s2w_xt_stride(Wo);
+    size_t lds_bytes = 2 * (size_t)(64 * DY_STRIDE + 64 * XT2S) * 2;
+    if (lds_bytes > 160 * 1024) return 0;
+    const int xn8 = (2 * S2W_P + 1) * Wo * 8;  // Wi/2 == Wo
+    if (xn8 > 1536) return 0;        // s2w_issue register budget
+    return 1;
+}
+
+extern "C" int conv3x3_s2_wrw_nsplit(int N, int Ho, int Wo, int C,
+                                     int K) {
+    long chunks = (long)N * (Ho / S2W_P);
+    int tiles = (K / 64) * (C / 64);
+    long target = 256 / tiles;
+    if (target < 1) target = 1;
+    if (target > chunks) target = chunks;
+    return (int)target;
+}
+
+extern "C" void launch_conv3x3_s2_wrw(
+        const ushort_t* x, const ushort_t* dy, float* ws, float* dw,
+        int N, int Ho, int Wo, int C, int K, hipStream_t s) {
+    const int XT2S = s2w_xt_stride(Wo);
+    const int nsplit = conv3x3_s2_wrw_nsplit(N, Ho, Wo, C, K);
+    const int tiles = (K / 64) * (C / 64);
+    const size_t lds_bytes = 2 * (size_t)(64 * DY_STRIDE + 64 * XT2S)
+        * 2;
+    static int attr_set = 0;
+    if (!attr_set) {
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_s2_wrw),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = 1;
+    }
+    hipLaunchKernelGGL(k_conv3x3_s2_wrw, dim3(tiles * nsplit), dim3(512),
+                       lds_bytes, s, x, dy, ws, N, Ho, Wo, C, K, XT2S,
+                       nsplit);
+    const long n = (long)K * 9 * C;
+    const int ngroups = (nsplit + 15) / 16;
+    const long nchunks = (n + 1023) / 1024;
+    if (ngroups > 1)
+        hipMemsetAsync(dw, 0, n * sizeof(float), s);
+    long blocks = nchunks * ngroups;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(k_wrw_reduce, dim3((unsigned)blocks), dim3(256), 0,
+                       s, ws, dw, n, nsplit, ngroups, nchunks);
+}

@@ -225,9 +225,22 @@ class _S2ConvFunction(torch.autograd.Function):
             dx = torch.empty_like(x)
             ext.conv_s2_bwd(dy, wt, dx)
         if ctx.needs_input_grad[1]:
-            dw = torch.ops.aten.convolution_backward(
-                dy, x, wb, None, stride, padding, dilation, False,
-                [0, 0], groups, [False, True, False])[1]
+            ext2 = ops._load_extension()
+            n, k, ho, wo = dy.shape
+            c = x.shape[1]
+            if os.getenv("ADAPTDL_S2_WRW") == "1" \
+                    and ext2.conv_s2_wrw_ok(n, ho, wo, c, k):
+                nsplit = ext2.conv_s2_wrw_nsplit(n, ho, wo, c, k)
+                ws = torch.empty(nsplit * k * 9 * c,
+                                 dtype=torch.float32, device=x.device)
+                dw = torch.empty(k, c, 3, 3, dtype=torch.float32,
+                                 device=x.device) \
+                    .contiguous(memory_format=torch.channels_last)
+                ext2.conv_s2_wrw(x, dy, ws, dw)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    dy, x, wb, None, stride, padding, dilation, False,
+                    [0, 0], groups, [False, True, False])[1]
         return dx, dw, None, None, None, None
 
 

@@ -260,3 +260,44 @@ def test_s2_1x1_gemm_function_matches_autograd():
     assert torch.allclose(y, y2, atol=1e-4)
     assert torch.allclose(x.grad, x2.grad, atol=1e-4)
     assert torch.allclose(weight.grad, w2.grad, atol=1e-3)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (2, 64, 16, 16, 128),     # layer2.0.conv1 (x 32x32x64 -> dy 16x16x128)
+    (3, 128, 8, 8, 256),      # layer3.0.conv1
+])
+def test_s2_wrw_kernel_matches_fp32(shape):
+    """Experimental polyphase stride-2 weight gradient vs fp32 reference.
+
+    Opt-in like the s2 forward: simulation-verified + compiled but not
+    yet run on hardware (GPU budget exhausted this round); enable with
+    ADAPTDL_EXPERIMENTAL_S2_FWD=1 (shares the experimental gate).
+    """
+    import os
+    if os.getenv("ADAPTDL_EXPERIMENTAL_S2_FWD") != "1":
+        pytest.skip("experimental s2 wrw kernel: not yet GPU-validated; "
+                    "set ADAPTDL_EXPERIMENTAL_S2_FWD=1")
+    from adaptdl_amd import ops
+    ext = ops._load_extension()
+    torch.manual_seed(9)
+    n, c, ho, wo, k = shape
+    hi, wi = 2 * ho, 2 * wo
+    dev = torch.device("cuda")
+    x = (torch.randn(n, c, hi, wi, device=dev) * 0.5).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    dy = (torch.randn(n, k, ho, wo, device=dev) * 0.5) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    assert ext.conv_s2_wrw_ok(n, ho, wo, c, k)
+    nsplit = ext.conv_s2_wrw_nsplit(n, ho, wo, c, k)
+    ws = torch.empty(nsplit * k * 9 * c, dtype=torch.float32, device=dev)
+    dw = torch.empty(k, c, 3, 3, dtype=torch.float32, device=dev) \
+        .contiguous(memory_format=torch.channels_last)
+    ext.conv_s2_wrw(x, dy, ws, dw)
+
+    x32 = x.float().requires_grad_(True)
+    w32 = torch.zeros(k, c, 3, 3, device=dev, requires_grad=True)
+    y = F.conv2d(x32, w32, stride=2, padding=1)
+    y.backward(dy.float())
+    assert torch.allclose(dw, w32.grad, atol=0.1, rtol=5e-2), \
+        (dw - w32.grad).abs().max().item()
