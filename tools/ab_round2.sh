@@ -1,0 +1,26 @@
+#!/bin/bash
+# Round-2 opener: validate + measure the three pre-built gated paths in
+# ONE gpurun call (budget-efficient).  Usage on the GPU box:
+#   bash tools/ab_round2.sh
+# Writes everything under gpurun_out/ for copy-back.
+set -x
+cd "$(dirname "$0")/.."
+mkdir -p gpurun_out
+
+# 1. Numerics of the unvalidated experimental kernels (s2 fwd + s2 wrw).
+ADAPTDL_EXPERIMENTAL_S2_FWD=1 timeout 180 python -m pytest \
+    tests/test_fused_conv.py -q -k "s2_fwd or s2_wrw" \
+    2>&1 | tail -3 | tee gpurun_out/ab_numerics.log
+
+# 2. Per-shape timings vs the library (appends s2 rows).
+timeout 180 python tools/conv_time.py > gpurun_out/ab_conv_time.log 2>&1
+tail -8 gpurun_out/ab_conv_time.log
+
+# 3. In-context bench A/B: baseline, then each gated path alone.
+for cfg in "" "ADAPTDL_S2_1X1=1" "ADAPTDL_S2_WRW=1"; do
+    name=${cfg:-baseline}; name=${name%%=*}
+    env $cfg timeout 260 python bench.py --steps 20 --warmup 12 \
+        > "gpurun_out/ab_bench_${name}.log" 2>&1
+    grep '"metric"' "gpurun_out/ab_bench_${name}.log" | tail -1
+done
+# Keep only measured wins (the bench_r10 rule); see ROADMAP.md item 3.
