@@ -1,5 +1,9 @@
 """Fused flat-bucket optimizers for the gradient engine.
 
+Reference parity note: the reference has no custom optimizers (its
+examples use plain torch.optim.SGD/Adam, which also work here); these
+exist purely as the MI355X-native fast path.
+
 When attached to a GradSyncEngine (AdaptiveDataParallel does this
 automatically), parameters and optimizer state are flattened into the
 same persistent per-(group, dtype) buffers the gradient buckets use, so
