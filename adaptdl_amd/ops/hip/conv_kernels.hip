@@ -627,14 +627,20 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
         }
 
         // ---- contraction: 9 taps x C/32 channel chunks ----
-        f32x4 acc[2][2];
+        // acc split by tau parity: a fragment's 9 taps form one serial
+        // MFMA dependency chain per cc iteration otherwise (KT=32 has
+        // only 2 (mf) chains per wave -- suspected cause of layer2's
+        // 255 us, ROADMAP item 1).  Parity doubles the chains.
+        f32x4 acc[2][2][2];
         #pragma unroll
         for (int mf = 0; mf < 2; ++mf)
             #pragma unroll
             for (int nf = 0; nf < 2; ++nf)
                 #pragma unroll
-                for (int r = 0; r < 4; ++r)
-                    acc[mf][nf][r] = 0.f;
+                for (int tp = 0; tp < 2; ++tp)
+                    #pragma unroll
+                    for (int r = 0; r < 4; ++r)
+                        acc[mf][nf][tp][r] = 0.f;
 
         // cc outer (runtime trip count, stays rolled); tau inner and
         // fully unrolled: a rolled tau loop compiled to 4 MFMAs each
@@ -668,9 +674,9 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
                     #pragma unroll
                     for (int nf = 0; nf < 2; ++nf)
                         if (nf < NF)
-                            acc[mf][nf] =
+                            acc[mf][nf][tau & 1] =
                                 __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                                    afrag, bfrag[nf], acc[mf][nf],
+                                    afrag, bfrag[nf], acc[mf][nf][tau & 1],
                                     0, 0, 0);
                 }
             }
@@ -697,7 +703,8 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_mm(
                         + r;
                     const int ko = wk * KH + nf * 16 + (lane & 15);
                     ystage[p * KT + ko] =
-                        (short)f2b(acc[mf][nf][r]);
+                        (short)f2b(acc[mf][nf][0][r]
+                                   + acc[mf][nf][1][r]);
                 }
             }
         }
