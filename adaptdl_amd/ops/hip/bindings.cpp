@@ -38,6 +38,10 @@ int conv3x3_mm_supported(int, int, int, int);
 int conv3x3_s2_bwd_supported(int, int, int, int);
 int conv3x3_s2_fwd_supported(int, int, int, int);
 int conv3x3_s2_wrw_supported(int, int, int, int);
+int conv3x3_s2_bwd_w8b_supported(int, int, int, int);
+void launch_conv3x3_s2_bwd_w8b(const unsigned short*,
+                               const unsigned short*, unsigned short*,
+                               int, int, int, int, hipStream_t);
 int conv3x3_s2_wrw_nsplit(int, int, int, int, int);
 void launch_conv3x3_s2_wrw(const unsigned short*, const unsigned short*,
                            float*, float*, int, int, int, int, int,
@@ -417,6 +421,33 @@ void conv_s2_wrw(torch::Tensor x, torch::Tensor dy, torch::Tensor ws,
                           stream());
 }
 
+bool conv_s2_bwd_w8b_ok(long N, long Ho, long Wo, long K, long C) {
+    (void)N;
+    return conv3x3_s2_bwd_w8b_supported((int)Ho, (int)Wo, (int)K,
+                                        (int)C) != 0;
+}
+
+// Wo=8 stride-2 bwd-data, P2=4 redesign (see conv_kernels.hip).
+void conv_s2_bwd_w8b(torch::Tensor dy, torch::Tensor wt,
+                     torch::Tensor dx) {
+    check_bn_x(dy, "dy"); check_bn_x(dx, "dx");
+    const long N = dy.size(0), K = dy.size(1), Ho = dy.size(2),
+        Wo = dy.size(3);
+    const long C = dx.size(1);
+    TORCH_CHECK(dx.size(0) == N && dx.size(2) == 2 * Ho &&
+                dx.size(3) == 2 * Wo, "dx shape mismatch");
+    TORCH_CHECK(wt.is_cuda() && wt.scalar_type() == torch::kBFloat16 &&
+                wt.is_contiguous() && wt.size(0) == C &&
+                wt.size(3) == K, "wt must be (C,3,3,K) contiguous bf16");
+    TORCH_CHECK(conv_s2_bwd_w8b_ok(N, Ho, Wo, K, C),
+                "unsupported w8b shape");
+    launch_conv3x3_s2_bwd_w8b((const unsigned short*)dy.data_ptr(),
+                              (const unsigned short*)wt.data_ptr(),
+                              (unsigned short*)dx.data_ptr(),
+                              (int)N, (int)Ho, (int)K, (int)C,
+                              stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -447,6 +478,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
             "s2 wrw workspace splits");
     mod.def("conv_s2_wrw", &conv_s2_wrw,
             "MFMA 3x3 s2 NHWC weight gradient (experimental)");
+    mod.def("conv_s2_bwd_w8b_ok", &conv_s2_bwd_w8b_ok,
+            "Wo=8 s2 bwd-data P2=4 redesign predicate");
+    mod.def("conv_s2_bwd_w8b", &conv_s2_bwd_w8b,
+            "Wo=8 s2 bwd-data P2=4 redesign (experimental)");
     mod.def("conv_s2_bwd_ok", &conv_s2_bwd_ok,
             "3x3 stride-2 bwd-data predicate");
     mod.def("conv_s2_bwd", &conv_s2_bwd,

@@ -2040,3 +2040,314 @@ extern "C" void launch_conv3x3_s2_wrw(
     hipLaunchKernelGGL(k_wrw_reduce, dim3((unsigned)blocks), dim3(256), 0,
                        s, ws, dw, n, nsplit, ngroups, nchunks);
 }
+
+// =====================================================================
+// Wo == 8 stride-2 backward-data, SECOND attempt (experimental).
+//
+// The shipped k_conv3x3_s2_bwd_w8 measured a 3.8 ms/step regression in
+// context (profiles/bench_r10): P2=2 makes 4096 tiny 64-pixel chunks
+// with 5 barriers + an f32 merge each.  This variant uses P2=4
+// (128-pixel chunks, half the barrier rounds, 2x work per round) with
+// frag-quad wave groups ({ph00, ph11} x both row-pairs vs the two
+// mixed phases: 10 vs 8 MFMAs per cc, 1.25x balance) and keeps the
+// 4-way k-split + f32 LDS merge.  Single-buffered dy window with
+// register prefetch (76 KB weights + 43.5 KB window; two windows do
+// not fit).  Gated: numerics test requires ADAPTDL_EXPERIMENTAL_S2_FWD,
+// production use requires winning a round-2 in-context A/B.
+// =====================================================================
+
+#define S2W8B_P2 4
+
+struct S2W8bRegs {
+    V16 v[3];
+};
+
+__device__ __forceinline__ void s2w8b_issue(
+        const ushort_t* __restrict__ dy, int n, int r0, int Ho, int Wo,
+        int K, int t, S2W8bRegs& r) {
+    const int pieces = (S2W8B_P2 + 1) * Wo * (K / 8);
+    #pragma unroll
+    for (int it = 0; it < 3; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (Wo * (K / 8));
+            const int rem = i % (Wo * (K / 8));
+            const int wo = rem / (K / 8);
+            const int kg = (rem % (K / 8)) * 8;
+            const int ho = r0 + j;
+            if (ho >= Ho) {
+                r.v[it].u4 = u32x4{0u, 0u, 0u, 0u};
+            } else {
+                r.v[it].u4 = *reinterpret_cast<const u32x4*>(
+                    dy + (((size_t)n * Ho + ho) * Wo + wo) * K + kg);
+            }
+        }
+    }
+}
+
+__device__ __forceinline__ void s2w8b_write(
+        short* __restrict__ dy_s, int Wo, int K, int KS, int LW, int t,
+        S2W8bRegs& r) {
+    const int pieces = (S2W8B_P2 + 1) * Wo * (K / 8);
+    #pragma unroll
+    for (int it = 0; it < 3; ++it) {
+        const int i = t + it * 512;
+        if (i < pieces) {
+            const int j = i / (Wo * (K / 8));
+            const int rem = i % (Wo * (K / 8));
+            const int wo = rem / (K / 8);
+            const int kg = (rem % (K / 8)) * 8;
+            *reinterpret_cast<u32x4*>(
+                &dy_s[(j * LW + wo) * KS + kg]) = r.v[it].u4;
+        }
+    }
+    const int zp = (S2W8B_P2 + 1) * 8 * (K / 8);   // pad col refresh
+    for (int i = t; i < zp; i += 512) {
+        const int j = i / (8 * (K / 8));
+        const int rem = i % (8 * (K / 8));
+        const int pc = rem / (K / 8);
+        const int kg = (rem % (K / 8)) * 8;
+        u32x4 z = {0u, 0u, 0u, 0u};
+        *reinterpret_cast<u32x4*>(
+            &dy_s[(j * LW + Wo + pc) * KS + kg]) = z;
+    }
+}
+
+// Wave group G owns four fragments: G=0 -> phases {0,3} x row-pairs,
+// G=1 -> {1,2} x row-pairs.  KQ = k-quarter, KC = per-wave cc count.
+template <int G, int KQ, int KC>
+__device__ __forceinline__ void s2w8b_contract(
+        const short* __restrict__ dy_s, const short* __restrict__ w_lds,
+        int KS, int KS2, int LW, int row16, int slot8,
+        f32x4 (&acc)[4][2]) {
+    const int w2 = row16 & 7;
+    const int h2base = row16 >> 3;   // 0..1 within the row pair
+    #pragma unroll
+    for (int cc = 0; cc < KC; ++cc) {
+        const int kbase = KQ * (KC * 32) + cc * 32;
+        #pragma unroll
+        for (int fi = 0; fi < 4; ++fi) {
+            constexpr int PHA = (G == 0) ? 0 : 1;
+            constexpr int PHB = (G == 0) ? 3 : 2;
+            const int ph = (fi >> 1) ? PHB : PHA;
+            const int f = fi & 1;                // row pair
+            const int h2 = 2 * f + h2base;
+            const int a = ph >> 1, b = ph & 1;
+            #pragma unroll
+            for (int dh = 0; dh < 3; ++dh) {
+                if (a ? (dh == 1) : (dh != 1)) continue;
+                #pragma unroll
+                for (int dw = 0; dw < 3; ++dw) {
+                    if (b ? (dw == 1) : (dw != 1)) continue;
+                    const int j = h2 + (dh == 0 ? 1 : 0);
+                    const int col = w2 + (dw == 0 ? 1 : 0);
+                    const bf16x8 afrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &dy_s[(j * LW + col) * KS + kbase + slot8]);
+                    const int tau = dh * 3 + dw;
+                    const bf16x8 bfrag =
+                        *reinterpret_cast<const bf16x8*>(
+                            &w_lds[(row16 * 9 + tau) * KS2 + kbase
+                                   + slot8]);
+                    acc[fi][cc & 1] =
+                        __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                            afrag, bfrag, acc[fi][cc & 1], 0, 0, 0);
+                }
+            }
+        }
+    }
+}
+
+template <int KC>
+__device__ __forceinline__ void s2w8b_contract_g(
+        int wid7, const short* __restrict__ dy_s,
+        const short* __restrict__ w_lds, int KS, int KS2, int LW,
+        int row16, int slot8, f32x4 (&acc)[4][2]) {
+    switch (wid7) {
+    case 0: s2w8b_contract<0, 0, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 1: s2w8b_contract<0, 1, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 2: s2w8b_contract<0, 2, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 3: s2w8b_contract<0, 3, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 4: s2w8b_contract<1, 0, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 5: s2w8b_contract<1, 1, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    case 6: s2w8b_contract<1, 2, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                     slot8, acc); break;
+    default: s2w8b_contract<1, 3, KC>(dy_s, w_lds, KS, KS2, LW, row16,
+                                      slot8, acc); break;
+    }
+}
+
+extern "C" __global__ __launch_bounds__(512, 2)
+void k_conv3x3_s2_bwd_w8b(
+        const ushort_t* __restrict__ dy, const ushort_t* __restrict__ wt,
+        ushort_t* __restrict__ dx, int N, int Ho, int K, int C,
+        int spread) {
+    extern __shared__ short lds[];
+    const int Wo = 8;
+    const int KS = K + 16;
+    const int KS2 = K + 8;
+    const int LW = Wo + 8;
+    short* w_lds = lds;              // [S2W8_CT * 9][KS2]
+    short* dy_s = lds + S2W8_CT * 9 * KS2;   // single buffer
+
+    const int Hi = 2 * Ho, Wi = 2 * Wo;
+    const int ct = blockIdx.x / spread;
+    const int sp = blockIdx.x % spread;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63;
+    const int wid = t >> 6;
+    const int wg2 = wid >> 2;        // frag-quad group
+    const int row16 = lane & 15;
+    const int slot8 = (lane >> 4) * 8;
+    const int CP2 = 4 * S2W8B_P2 * Wo;   // 128 chunk pixels
+
+    {
+        const int pieces = S2W8_CT * 9 * (K / 8);
+        const ushort_t* wg = wt + (size_t)ct * S2W8_CT * 9 * K;
+        for (int i = t; i < pieces; i += 512) {
+            const int row = i / (K / 8);
+            const int kg = (i % (K / 8)) * 8;
+            *reinterpret_cast<u32x4*>(&w_lds[row * KS2 + kg]) =
+                *reinterpret_cast<const u32x4*>(
+                    wg + (size_t)row * K + kg);
+        }
+    }
+
+    const int rows_per_img = Ho / S2W8B_P2;
+    const long chunks_total = (long)N * rows_per_img;
+
+    S2W8bRegs regs;
+    if (sp < chunks_total) {
+        s2w8b_issue(dy, (int)(sp / rows_per_img),
+                    (int)(sp % rows_per_img) * S2W8B_P2, Ho, Wo, K, t,
+                    regs);
+        s2w8b_write(dy_s, Wo, K, KS, LW, t, regs);
+    }
+    __syncthreads();
+    if (sp + spread < chunks_total) {
+        const long qn = sp + spread;
+        s2w8b_issue(dy, (int)(qn / rows_per_img),
+                    (int)(qn % rows_per_img) * S2W8B_P2, Ho, Wo, K, t,
+                    regs);
+    }
+
+    for (long q = sp; q < chunks_total; q += spread) {
+        const int n = (int)(q / rows_per_img);
+        const int r0 = (int)(q % rows_per_img) * S2W8B_P2;
+
+        if (q != sp) {
+            __syncthreads();         // merge/store of q-1 fully read
+            s2w8b_write(dy_s, Wo, K, KS, LW, t, regs);
+            __syncthreads();
+            if (q + spread < chunks_total) {
+                const long qn = q + spread;
+                s2w8b_issue(dy, (int)(qn / rows_per_img),
+                            (int)(qn % rows_per_img) * S2W8B_P2, Ho, Wo,
+                            K, t, regs);
+            }
+        }
+
+        f32x4 acc[4][2];
+        #pragma unroll
+        for (int fi = 0; fi < 4; ++fi)
+            #pragma unroll
+            for (int pp = 0; pp < 2; ++pp)
+                #pragma unroll
+                for (int r = 0; r < 4; ++r)
+                    acc[fi][pp][r] = 0.f;
+
+        if (K / 32 == 8)
+            s2w8b_contract_g<2>(wid & 7, dy_s, w_lds, KS, KS2, LW,
+                                row16, slot8, acc);
+        else
+            s2w8b_contract_g<1>(wid & 7, dy_s, w_lds, KS, KS2, LW,
+                                row16, slot8, acc);
+
+        float* fstage = reinterpret_cast<float*>(dy_s);  // [128][CT]
+        __syncthreads();
+        for (int i = t; i < CP2 * S2W8_CT; i += 512)
+            fstage[i] = 0.f;
+        __syncthreads();
+        #pragma unroll
+        for (int fi = 0; fi < 4; ++fi) {
+            const int ph = (fi >> 1) ? ((wg2 == 0) ? 3 : 2)
+                                     : ((wg2 == 0) ? 0 : 1);
+            const int f = fi & 1;
+            #pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                const int m = (lane >> 4) * 4 + r;
+                const int h2 = 2 * f + (m >> 3);
+                const int w2 = m & 7;
+                const int c = lane & 15;
+                const int s = (ph * S2W8B_P2 + h2) * Wo + w2;
+                atomicAdd(&fstage[s * S2W8_CT + c],
+                          acc[fi][0][r] + acc[fi][1][r]);
+            }
+        }
+        __syncthreads();
+        {
+            const int pieces = 2 * S2W8B_P2 * Wi * (S2W8_CT / 8);
+            const size_t base = ((size_t)n * Hi + 2 * r0) * Wi;
+            for (int i = t; i < pieces; i += 512) {
+                const int hl = i / (Wi * (S2W8_CT / 8));
+                const int rem = i % (Wi * (S2W8_CT / 8));
+                const int wi = rem / (S2W8_CT / 8);
+                const int cg = (rem % (S2W8_CT / 8)) * 8;
+                const int ph = ((hl & 1) << 1) | (wi & 1);
+                const int s = (ph * S2W8B_P2 + (hl >> 1)) * Wo
+                    + (wi >> 1);
+                ushort_t out[8];
+                #pragma unroll
+                for (int jj = 0; jj < 8; ++jj)
+                    out[jj] = f2b(fstage[s * S2W8_CT + cg + jj]);
+                *reinterpret_cast<u32x4*>(
+                    dx + (base + (size_t)hl * Wi + wi) * C
+                    + ct * S2W8_CT + cg) =
+                    *reinterpret_cast<const u32x4*>(out);
+            }
+        }
+    }
+}
+
+extern "C" int conv3x3_s2_bwd_w8b_supported(int Ho, int Wo, int K,
+                                            int C) {
+    if (Wo != 8) return 0;
+    if (Ho % S2W8B_P2) return 0;
+    if ((K != 128 && K != 256) || C % S2W8_CT) return 0;
+    if ((S2W8B_P2 + 1) * Wo * (K / 8) > 1536) return 0;
+    size_t need = ((size_t)S2W8_CT * 9 * (K + 8)
+                   + (size_t)(S2W8B_P2 + 1) * (Wo + 8) * (K + 16)) * 2;
+    return need <= 160 * 1024;
+}
+
+extern "C" void launch_conv3x3_s2_bwd_w8b(
+        const ushort_t* dy, const ushort_t* wt, ushort_t* dx,
+        int N, int Ho, int K, int C, hipStream_t s) {
+    const int Wo = 8;
+    const size_t lds_bytes = ((size_t)S2W8_CT * 9 * (K + 8)
+                              + (size_t)(S2W8B_P2 + 1) * (Wo + 8)
+                                * (K + 16)) * 2;
+    static int attr_set = 0;
+    if (!attr_set) {
+        hipFuncSetAttribute(
+            reinterpret_cast<const void*>(&k_conv3x3_s2_bwd_w8b),
+            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        attr_set = 1;
+    }
+    const long chunks = (long)N * (Ho / S2W8B_P2);
+    const int nct = C / S2W8_CT;
+    long spread = 256 / nct;
+    if (spread > chunks) spread = chunks;
+    if (spread < 1) spread = 1;
+    hipLaunchKernelGGL(k_conv3x3_s2_bwd_w8b,
+                       dim3((unsigned)(nct * spread)), dim3(512),
+                       lds_bytes, s, dy, wt, dx, N, Ho, K, C,
+                       (int)spread);
+}

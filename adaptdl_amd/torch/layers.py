@@ -223,7 +223,13 @@ class _S2ConvFunction(torch.autograd.Function):
             # original (dh, dw) indices in the polyphase decomposition).
             wt = wb.permute(1, 2, 3, 0).contiguous()
             dx = torch.empty_like(x)
-            ext.conv_s2_bwd(dy, wt, dx)
+            n2, k2, ho2, wo2 = dy.shape
+            if os.getenv("ADAPTDL_S2_W8B") == "1" and wo2 == 8 \
+                    and ext.conv_s2_bwd_w8b_ok(n2, ho2, wo2, k2,
+                                               x.shape[1]):
+                ext.conv_s2_bwd_w8b(dy, wt, dx)   # P2=4 redesign A/B
+            else:
+                ext.conv_s2_bwd(dy, wt, dx)
         if ctx.needs_input_grad[1]:
             ext2 = ops._load_extension()
             n, k, ho, wo = dy.shape

@@ -301,3 +301,35 @@ def test_s2_wrw_kernel_matches_fp32(shape):
     y.backward(dy.float())
     assert torch.allclose(dw, w32.grad, atol=0.1, rtol=5e-2), \
         (dw - w32.grad).abs().max().item()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("shape", [
+    (2, 128, 8, 8, 256),
+    (3, 32, 8, 8, 128),
+])
+def test_s2_bwd_w8b_matches_fp32(shape):
+    """P2=4 redesign of the Wo=8 stride-2 bwd-data (experimental;
+    shares the unvalidated-kernel gate)."""
+    import os
+    if os.getenv("ADAPTDL_EXPERIMENTAL_S2_FWD") != "1":
+        pytest.skip("experimental w8b kernel: not yet GPU-validated; "
+                    "set ADAPTDL_EXPERIMENTAL_S2_FWD=1")
+    from adaptdl_amd import ops
+    ext = ops._load_extension()
+    torch.manual_seed(10)
+    n, c, ho, wo, k = shape
+    hi, wi = 2 * ho, 2 * wo
+    dev = torch.device("cuda")
+    dy = (torch.randn(n, k, ho, wo, device=dev) * 0.5) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(k, c, 3, 3, device=dev) * 0.2).to(torch.bfloat16)
+    assert ext.conv_s2_bwd_w8b_ok(n, ho, wo, k, c)
+    wt = w.permute(1, 2, 3, 0).contiguous()
+    dx = torch.empty(n, c, hi, wi, dtype=torch.bfloat16, device=dev) \
+        .contiguous(memory_format=torch.channels_last)
+    ext.conv_s2_bwd_w8b(dy, wt, dx)
+    ref = torch.nn.grad.conv2d_input(
+        (n, c, hi, wi), w.float(), dy.float(), stride=2, padding=1)
+    assert torch.allclose(dx.float(), ref, atol=0.1, rtol=5e-2), \
+        (dx.float() - ref).abs().max().item()
