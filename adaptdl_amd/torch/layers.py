@@ -328,7 +328,8 @@ class FusedConv2d(nn.Conv2d):
         # CT=16 makes each block re-read dy 8x across 128 tiny chunk
         # iterations of 5 barriers + an f32 merge each.  Off by default
         # until the per-chunk overhead is redesigned (ROADMAP item 2).
-        if w // 2 != 16 and os.getenv("ADAPTDL_S2_W8") != "1":
+        if w // 2 != 16 and os.getenv("ADAPTDL_S2_W8") != "1" \
+                and os.getenv("ADAPTDL_S2_W8B") != "1":
             return False
         return bool(ext.conv_s2_bwd_ok(n, h // 2, w // 2,
                                        self.out_channels, c))
