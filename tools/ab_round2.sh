@@ -9,7 +9,7 @@ mkdir -p gpurun_out
 
 # 1. Numerics of the unvalidated experimental kernels (s2 fwd + s2 wrw).
 ADAPTDL_EXPERIMENTAL_S2_FWD=1 timeout 180 python -m pytest \
-    tests/test_fused_conv.py -q -k "s2_fwd or s2_wrw" \
+    tests/test_fused_conv.py -q -k "s2_fwd or s2_wrw or w8b" \
     2>&1 | tail -3 | tee gpurun_out/ab_numerics.log
 
 # 2. Per-shape timings vs the library (appends s2 rows).
