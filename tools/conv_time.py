@@ -148,3 +148,26 @@ for n, c, hi, wi, k in [(1024, 64, 32, 32, 128)]:
             fn()
         t1.record(); torch.cuda.synchronize()
         print(f"s2 1x1 N{n} C{c}->K{k} fwd+bwd {tag}: {t0.elapsed_time(t1)*100:8.1f}us")
+
+# Wo=8 s2 bwd-data: P2=4 redesign vs shipped w8 kernel vs library
+for n, c, ho, wo, k in [(1024, 128, 8, 8, 256)]:
+    if not ext.conv_s2_bwd_w8b_ok(n, ho, wo, k, c):
+        print("w8b: shape unsupported"); break
+    hi, wi = 2 * ho, 2 * wo
+    dy = torch.randn(n, k, ho, wo, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(k, c, 3, 3, device=dev) * 0.1).to(torch.bfloat16)
+    wt2 = w.permute(1, 2, 3, 0).contiguous()
+    dx = torch.empty(n, c, hi, wi, dtype=torch.bfloat16, device=dev).contiguous(memory_format=torch.channels_last)
+    gf = 2 * n * k * c * 9 * ho * wo / 1e9
+    for fn, tag in ((lambda: ext.conv_s2_bwd_w8b(dy, wt2, dx), "w8b"),
+                    (lambda: ext.conv_s2_bwd(dy, wt2, dx), "w8 ")):
+        for _ in range(3):
+            fn()
+        torch.cuda.synchronize()
+        t0 = torch.cuda.Event(enable_timing=True); t1 = torch.cuda.Event(enable_timing=True)
+        t0.record()
+        for _ in range(10):
+            fn()
+        t1.record(); torch.cuda.synchronize()
+        us = t0.elapsed_time(t1) * 100
+        print(f"s2bwd-w8 N{n} K{k} {ho}x{wo}->C{c} {tag}: {us:7.1f}us ({gf/us*1e3:6.0f} GF/s)")
