@@ -72,13 +72,17 @@ def main():
     loader = adl.AdaptiveDataLoader(SyntheticCifar(args.samples),
                                     batch_size=args.bs, shuffle=True,
                                     drop_last=True)
-    loader.autoscale_batch_size(args.max_bs, local_bsz_bounds=(32, 1024),
-                                gradient_accumulation=True)
+    if args.max_bs > 0:   # --max-bs 0 disables adaptive batch sizing
+        loader.autoscale_batch_size(args.max_bs,
+                                    local_bsz_bounds=(32, 1024),
+                                    gradient_accumulation=True)
 
     stats = adl.Accumulator()
     for epoch in adl.remaining_epochs_until(args.epochs):
         model.train()
+        bsz = None
         for x, y in loader:
+            bsz = loader.current_batch_size
             x, y = x.to(device), y.to(device)
             if use_gpu:
                 x = x.contiguous(memory_format=torch.channels_last)
@@ -96,8 +100,7 @@ def main():
         with stats.synchronized():
             if int(os.getenv("ADAPTDL_REPLICA_RANK", "0")) == 0:
                 print("epoch {} replicas {} batch {} loss {:.4f}".format(
-                    epoch, env.num_replicas(),
-                    loader.current_batch_size,
+                    epoch, env.num_replicas(), bsz,
                     stats["loss_sum"] / max(stats["count"], 1)))
             stats["loss_sum"] = stats["count"] = 0
 
