@@ -62,6 +62,7 @@ def main():
     for epoch in adl.remaining_epochs_until(args.epochs):
         total, count = 0.0, 0
         for users, items, labels in loader:
+            bsz = loader.current_batch_size
             users, items = users.to(device), items.to(device)
             labels = labels.to(device)
             optim.zero_grad()
@@ -73,7 +74,7 @@ def main():
             count += 1
         if env.replica_rank() == 0 and count:
             print("epoch {} bce {:.4f} batch {} accum {}".format(
-                epoch, total / count, loader.current_batch_size,
+                epoch, total / count, bsz,
                 loader.accumulation_steps))
 
 
