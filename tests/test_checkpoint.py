@@ -51,3 +51,40 @@ def _run_ckpt():
 
 def test_checkpoint_restarts():
     _run_ckpt()
+
+
+def test_on_disk_format_contract(tmp_path, monkeypatch):
+    """The reference's on-disk layout (BASELINE.json: 'checkpoint format
+    parity'): one ``checkpoint-<num_restarts>`` directory per save,
+    atomically renamed from a ``_checkpoint`` staging dir, holding one
+    file per named State; older checkpoint dirs are pruned; loading
+    picks the latest K.
+    """
+    import os
+    monkeypatch.setenv("ADAPTDL_CHECKPOINT_PATH", str(tmp_path))
+    monkeypatch.setenv("ADAPTDL_REPLICA_RANK", "0")
+    monkeypatch.setenv("ADAPTDL_NUM_REPLICAS", "1")
+    monkeypatch.setenv("ADAPTDL_NUM_RESTARTS", "0")
+    checkpoint._REGISTRY.clear()
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+
+    s1, s2 = _CounterState("alpha"), _CounterState("beta")
+    s1.count, s2.count = 7, 11
+    checkpoint.save_all_states()
+    assert sorted(os.listdir(tmp_path)) == ["checkpoint-0"]
+    assert sorted(os.listdir(tmp_path / "checkpoint-0")) == \
+        ["alpha", "beta"]
+
+    # second save at a higher restart count prunes the old dir
+    monkeypatch.setenv("ADAPTDL_NUM_RESTARTS", "3")
+    s1.count = 8
+    checkpoint.save_all_states()
+    assert sorted(os.listdir(tmp_path)) == ["checkpoint-3"]
+
+    # a fresh process (new registry) loads the latest checkpoint
+    checkpoint._REGISTRY.clear()
+    fresh = _CounterState("alpha")
+    checkpoint.load_state(fresh)
+    assert fresh.count == 8
+    collective.teardown()
