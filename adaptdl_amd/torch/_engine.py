@@ -112,6 +112,13 @@ class GradSyncEngine(object):
         cap_bytes = int(cap_mb * 1024 * 1024)
         self.require_sync = True
         self.accum_count = 0       # completed (un-synced) microbatches
+        # hipGraph-capture mode (set by torch.graph_step.GraphedStepper):
+        # device work proceeds normally but host-visible side effects that
+        # are illegal or meaningless during stream capture are deferred —
+        # no hipEvent timing records, and _on_sync_done (which host-syncs
+        # via stats.cpu()) is left for the stepper to invoke after replay.
+        self.graph_mode = False
+        self.pending_sync_done = False
         self._callback_queued = False
         self._hooks = []
         self.buckets = []
@@ -219,7 +226,11 @@ class GradSyncEngine(object):
             self._stats_work = torch.distributed.all_reduce(
                 self.stats[0], op=torch.distributed.ReduceOp.SUM,
                 group=self._pg, async_op=True)
-        if self._use_events:
+        if self.graph_mode:
+            # Timing events cannot be recorded inside a stream capture.
+            self._sync_start_ev = None
+            self._sync_start_t = None
+        elif self._use_events:
             self._sync_start_ev = torch.cuda.Event(enable_timing=True)
             self._sync_start_ev.record()
         else:
@@ -238,6 +249,15 @@ class GradSyncEngine(object):
         if self._stats_work is not None:
             self._stats_work.wait()
             self._stats_work = None
+        if self.graph_mode:
+            # Defer the host-side GNS update: _on_sync_done host-syncs via
+            # stats.cpu(), which is illegal during stream capture.  The
+            # GraphedStepper invokes it after each replay, when the stats
+            # tensor holds this step's values.
+            self.last_sync_time = 0.0
+            self._sync_end_ev = None
+            self.pending_sync_done = True
+            return
         if self._use_events:
             self._sync_end_ev = torch.cuda.Event(enable_timing=True)
             self._sync_end_ev.record()

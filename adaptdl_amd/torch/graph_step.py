@@ -1,0 +1,330 @@
+"""hipGraph capture of the steady-state training microbatch cycle.
+
+EXPERIMENTAL, env-gated with ``ADAPTDL_HIPGRAPH=1`` (ROADMAP item 5):
+trace-union analysis of the flagship bench (profiles/bench_r08.json
+discussion) measured 4.3% of step time as GPU idle gaps between the many
+small kernel launches of forward+backward.  Capturing the pinned
+steady-state microbatch into a hipGraph replays the whole kernel sequence
+with one launch, closing those gaps.
+
+Scope and shape of the capture
+------------------------------
+
+One optimizer-step cycle with ``A`` gradient-accumulation microbatches has
+``A + 1`` microbatch positions whose *device* work differs (the engine's
+fused statistic kernels depend on the position — see
+``gradient_noise_scale.py`` ``_accum_stat``/``_local_stat``):
+
+==========  ====================================================
+kind        captured device work
+==========  ====================================================
+"first"     bucket+stats zeroing, forward+backward,
+            per-bucket ``sqsum`` + prev snapshot       (pos 0)
+"mid"       forward+backward, per-bucket ``sqsum_diff``
+                                                (pos 1 .. A-1)
+"sync"      forward+backward, local-stat flush, bucket
+            all-reduce (world>1), scale + total-``sqsum``
+                                                       (pos A)
+"solo"      the whole cycle when A == 0 (zeroing + sync)
+==========  ====================================================
+
+Each kind is captured once (lazily, at its first occurrence after one
+eager warmup cycle) into its own hipGraph; all graphs share one memory
+pool and are replayed strictly in cycle order.  Everything *host-side*
+stays eager: the optimizer step (patched LR scaling + fused kernels),
+the GNS host math (``_on_sync_done`` — deferred by the engine's
+``graph_mode`` flag and invoked here after the sync-position replay,
+when the stats tensor holds the step's values), and the dataloader's
+profile/exit-flag machinery, which all run between replays exactly as
+they do between eager microbatches.
+
+Because graph replay executes no Python, the host-visible engine state a
+replayed microbatch would have produced (``accum_count``, zero-grad
+bookkeeping, sync flags) is applied explicitly after each replay from
+the known cycle position.  The values are *absolute* (not increments),
+so the same code path is correct both for true hipGraph replays (no
+Python ran) and for the eager test backend (Python ran and already
+advanced the state to the same values).
+
+Safety rules (all checked per cycle; violation falls back to eager):
+
+- the cycle signature — (atomic batch size, accumulation steps, input
+  shapes/dtypes) — must match the captured one; a change (e.g. the
+  goodput model re-choosing the batch size) drops the graphs and
+  re-warms,
+- the observed sync position must match the expected one,
+- AMP GradScaler (mp_scaler) is not supported (its inf-check host logic
+  inspects gradients between backward and step); construction refuses,
+- any capture error permanently disables the stepper for the run (the
+  current microbatch is re-run eagerly).
+
+Reference note: the reference has no equivalent (it delegates launch
+scheduling to torch DDP); this is an MI355X-native addition.
+"""
+
+import contextlib
+import logging
+import os
+
+import torch
+
+from adaptdl_amd.torch.data import current_dataloader
+
+__all__ = ["GraphedStepper", "maybe_graphed_stepper"]
+
+LOG = logging.getLogger(__name__)
+
+
+class HipGraphBackend(object):
+    """Real stream-capture backend (requires a GPU).
+
+    ``capture`` records the callable's kernels into a hipGraph WITHOUT
+    executing them (host-side Python runs once, recording); the caller
+    replays immediately afterwards so the microbatch happens exactly
+    once.  All graphs share one memory pool; replays must follow the
+    capture order.
+    """
+
+    def __init__(self):
+        self._pool = torch.cuda.graph_pool_handle()
+        self._side = torch.cuda.Stream()
+
+    @contextlib.contextmanager
+    def warmup(self):
+        """Eager warmup on a side stream (the torch.cuda.graph protocol:
+        lazy inits must not land on the capture stream)."""
+        self._side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._side):
+            yield
+        torch.cuda.current_stream().wait_stream(self._side)
+
+    def capture(self, fn):
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph, pool=self._pool):
+            out = fn()
+        return graph, out
+
+    def replay(self, graph):
+        graph.replay()
+
+
+class EagerBackend(object):
+    """Test backend: "capture" stores the callable, "replay" executes it.
+
+    The capture-then-immediately-replay flow therefore executes the
+    microbatch exactly once, like the real backend — but through the
+    full Python path with ``engine.graph_mode`` set, exercising the
+    deferred-host-work and absolute-state bookkeeping on CPU.
+    """
+
+    @contextlib.contextmanager
+    def warmup(self):
+        yield
+
+    def capture(self, fn):
+        return fn, None
+
+    def replay(self, fn):
+        fn()
+
+
+class GraphedStepper(object):
+    """Routes training microbatches through captured hipGraphs.
+
+    Arguments:
+        adp: the AdaptiveDataParallel instance.
+        optimizer: the (patched) optimizer; its ``step()`` stays eager
+            and must be called by the user after each microbatch, as in
+            a plain training loop.
+        fwd_bwd: callable ``fwd_bwd(*tensors) -> loss`` performing ONE
+            microbatch: ``optimizer.zero_grad()``, forward, loss,
+            ``loss.backward()`` — and nothing else.
+        backend: capture backend; defaults to HipGraphBackend on GPU.
+        warmup_cycles: eager cycles run before capturing (after enable
+            and after every signature change).
+
+    Usage::
+
+        stepper = GraphedStepper(adp, optim, fwd_bwd)
+        for batch in loader:
+            stepper.microbatch(batch)   # replaces fwd_bwd(batch)
+            optim.step()
+    """
+
+    def __init__(self, adp, optimizer, fwd_bwd, backend=None,
+                 warmup_cycles=1):
+        if getattr(adp.gns, "_mp_scaler", None) is not None:
+            raise ValueError("GraphedStepper does not support mp_scaler "
+                             "(GradScaler host logic is not capturable)")
+        self._adp = adp
+        self._gns = adp.gns
+        self._engine = adp.gns.engine
+        self._optimizer = optimizer
+        self._fwd_bwd = fwd_bwd
+        self._backend = backend or HipGraphBackend()
+        self._warmup_cycles = warmup_cycles
+        self._disabled = False
+
+        self._sig = None          # captured cycle signature
+        self._accum = 0           # A of the captured cycle
+        self._pos = 0             # position within the current cycle
+        self._warm_left = warmup_cycles
+        self._graphs = {}         # kind -> (graph, loss_out)
+        self._static = None       # static input buffers
+        self.stats = {"captures": 0, "replays": 0, "eager": 0,
+                      "fallbacks": 0}
+        # Pre-allocate the differenced-estimator snapshots so they are
+        # plain allocator memory, not graph-pool memory.
+        for bucket in self._engine.buckets:
+            bucket.ensure_prev()
+
+    # ---- public entry ----------------------------------------------------
+
+    def microbatch(self, *tensors):
+        """Run one training microbatch (replayed when possible).
+
+        Returns the loss tensor when available (eager runs and real
+        captures return the static loss buffer; replays return the same
+        buffer, which holds the value after the replay completes).
+        """
+        dl = current_dataloader()
+        if self._disabled or dl is None or not dl.training:
+            return self._run_eager(*tensors)
+
+        accum = dl.accumulation_steps
+        is_sync = dl.is_optim_step()
+        if self._pos == 0:
+            sig = (dl.current_local_bsz, accum,
+                   tuple((tuple(t.shape), t.dtype) for t in tensors))
+            if sig != self._sig:
+                self._reset(sig, accum)
+        # Desync guard: position math must agree with the dataloader.
+        expected_sync = (self._accum == 0) or (self._pos == self._accum)
+        if is_sync != expected_sync or accum != self._accum:
+            LOG.warning("graph stepper desync (pos %d, accum %d->%d, "
+                        "sync %s); falling back to eager this cycle",
+                        self._pos, self._accum, accum, is_sync)
+            self.stats["fallbacks"] += 1
+            self._sig = None      # force re-signature at next cycle start
+            self._pos = 0 if is_sync else self._pos + 1
+            return self._run_eager(*tensors)
+
+        kind = ("solo" if self._accum == 0 else
+                "sync" if is_sync else
+                "first" if self._pos == 0 else "mid")
+
+        if self._warm_left > 0:
+            with self._backend.warmup():
+                loss = self._run_eager(*tensors)
+            if is_sync:
+                self._warm_left -= 1
+        elif kind not in self._graphs:
+            loss = self._capture(kind, *tensors)
+        else:
+            loss = self._replay(kind, *tensors)
+
+        self._pos = 0 if is_sync else self._pos + 1
+        return loss
+
+    # ---- internals -------------------------------------------------------
+
+    def _run_eager(self, *tensors):
+        self._engine.graph_mode = False
+        self.stats["eager"] += 1
+        return self._fwd_bwd(*tensors)
+
+    def _reset(self, sig, accum):
+        self._graphs.clear()
+        self._static = None
+        self._sig = sig
+        self._accum = accum
+        self._warm_left = self._warmup_cycles
+        self._pos = 0
+
+    def _ensure_static(self, tensors):
+        if self._static is None:
+            self._static = [torch.empty_like(t) for t in tensors]
+        for dst, src in zip(self._static, tensors):
+            dst.copy_(src, non_blocking=True)
+
+    def _set_pre_state(self, kind):
+        sync = kind in ("sync", "solo")
+        self._adp.require_backward_grad_sync = sync
+        self._engine.require_sync = sync
+        self._engine.graph_mode = True
+
+    def _apply_post_state(self, kind):
+        """Absolute host state after a capture/replay of ``kind``.
+
+        Graph replay executes no Python, so the engine's host-visible
+        bookkeeping is applied here; on the eager backend the Python
+        path already advanced it to these same values.
+        """
+        engine = self._engine
+        engine._callback_queued = False
+        for bucket in engine.buckets:
+            bucket.ready = 0
+        if kind == "first":
+            engine.accum_count = 1
+            self._gns._should_zero_grad = False
+        elif kind == "mid":
+            engine.accum_count = self._pos + 1
+            self._gns._should_zero_grad = False
+        else:  # sync / solo
+            engine.accum_count = \
+                self._accum + 1 if kind == "sync" else 1
+            for bucket in engine.buckets:
+                bucket.work = None
+            engine.last_sync_time = 0.0
+            engine.pending_sync_done = False
+            engine.graph_mode = False
+            # Deferred host finalize: GNS estimates, scaling-rule gain,
+            # progress, grad-param reporting (reads the stats tensor the
+            # replay just produced).
+            self._gns._on_sync_done()
+
+    def _capture(self, kind, *tensors):
+        self._ensure_static(tensors)
+        self._set_pre_state(kind)
+        static = self._static
+        try:
+            graph, out = self._backend.capture(
+                lambda: self._fwd_bwd(*static))
+        except Exception:
+            LOG.exception("hipGraph capture failed for %r; disabling "
+                          "graph stepping for this run", kind)
+            self._disabled = True
+            self._engine.graph_mode = False
+            self.stats["fallbacks"] += 1
+            return self._run_eager(*tensors)
+        self._graphs[kind] = (graph, out)
+        self.stats["captures"] += 1
+        # Capture records without executing: replay immediately so this
+        # microbatch happens exactly once, on the data just copied in.
+        self._backend.replay(graph)
+        self._apply_post_state(kind)
+        return out
+
+    def _replay(self, kind, *tensors):
+        self._ensure_static(tensors)
+        self._set_pre_state(kind)
+        graph, out = self._graphs[kind]
+        self._backend.replay(graph)
+        self.stats["replays"] += 1
+        self._apply_post_state(kind)
+        return out
+
+
+def maybe_graphed_stepper(adp, optimizer, fwd_bwd):
+    """Build a GraphedStepper iff ``ADAPTDL_HIPGRAPH=1`` and a GPU is
+    available; returns None (caller keeps the eager path) otherwise."""
+    if os.getenv("ADAPTDL_HIPGRAPH") != "1":
+        return None
+    if not torch.cuda.is_available():
+        LOG.warning("ADAPTDL_HIPGRAPH=1 ignored: no GPU available")
+        return None
+    try:
+        return GraphedStepper(adp, optimizer, fwd_bwd)
+    except ValueError as exc:
+        LOG.warning("ADAPTDL_HIPGRAPH=1 ignored: %s", exc)
+        return None

@@ -101,8 +101,9 @@ def main():
                                     local_bsz_bounds=(lo, hi),
                                     gradient_accumulation=True)
 
-    def train_step(idx):
-        idx = (idx % args.pool).to(device, non_blocking=True)
+    graph_stepper = None  # ADAPTDL_HIPGRAPH=1: set before the probe pass
+
+    def fwd_bwd(idx):
         x = pool_x[idx]
         if channels_last:
             x = x.contiguous(memory_format=torch.channels_last)
@@ -114,6 +115,14 @@ def main():
         else:
             loss = F.cross_entropy(adp(x), y)
         loss.backward()
+        return loss
+
+    def train_step(idx):
+        idx = (idx % args.pool).to(device, non_blocking=True)
+        if graph_stepper is not None:
+            graph_stepper.microbatch(idx)
+        else:
+            fwd_bwd(idx)
         optim.step()
 
     def barrier_sync():
@@ -156,15 +165,26 @@ def main():
                     break
             if restart_pass:
                 restart_pass = False
+                # Experimental hipGraph capture of the steady microbatch
+                # cycle (ADAPTDL_HIPGRAPH=1): created before the probe
+                # pass so warmup + capture complete outside the timed
+                # region (probe runs 5 cycles instead of 2: one eager
+                # warmup cycle, one capture cycle, three replay cycles).
+                if os.getenv("ADAPTDL_HIPGRAPH") == "1" and use_gpu:
+                    from adaptdl_amd.torch.graph_step import \
+                        maybe_graphed_stepper
+                    graph_stepper = maybe_graphed_stepper(adp, optim,
+                                                          fwd_bwd)
+                probe_target = 2 if graph_stepper is None else 5
                 # One probe pass: let _sync_local_bsz adopt the fitted
-                # model's choice, run 2 steps to settle caches, then time.
+                # model's choice, run steps to settle caches, then time.
                 probe = 0
                 for idx in loader:
                     is_optim = loader._elastic.is_optim_step()
                     train_step(idx)
                     if is_optim:
                         probe += 1
-                    if probe >= 2:
+                    if probe >= probe_target:
                         break
                 global_batch = loader._elastic.current_batch_size
                 # Pin the adaptive choice for the timed region: the metric
@@ -225,6 +245,8 @@ def main():
                 "parallelism": "dp{}".format(world),
             },
         }
+        if graph_stepper is not None:
+            result["config"]["hipgraph"] = dict(graph_stepper.stats)
         print(json.dumps(result))
     if world > 1:
         torch.distributed.destroy_process_group()

@@ -1,0 +1,188 @@
+"""GraphedStepper cycle bookkeeping on CPU (EagerBackend).
+
+The hipGraph stepper's risk is not the captured kernels (replay runs the
+identical sequence) but the host-side bookkeeping it takes over from the
+engine: deferred ``_on_sync_done``, absolute accum-count state, cycle
+position tracking, signature-change fallback.  The EagerBackend executes
+the full Python path with ``engine.graph_mode`` set — exactly the deferred
+flow — so training through it must be bit-identical to plain eager
+training.  (The real capture backend is validated by the env-gated GPU
+test in tests/test_gpu_kernels.py.)
+"""
+
+import os
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+import adaptdl_amd.collective as collective
+
+from conftest import elastic_multiprocessing
+
+
+def _build(accum_steps):
+    """Tiny deterministic training stack with forced accumulation."""
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.data import AdaptiveDataLoaderHelper
+
+    # Force a fixed (atomic_bsz, accum_steps) choice instead of the
+    # goodput model's, so eager and graphed runs see identical cycles.
+    def fake_sync(self):
+        self._state.current_local_bsz = 8
+        self._state.accumulation_steps = accum_steps
+        return 8
+
+    AdaptiveDataLoaderHelper._sync_local_bsz = fake_sync
+
+    torch.manual_seed(0)
+    xs = torch.randn(48, 8)
+    ys = torch.randint(0, 4, (48,))
+    dataset = torch.utils.data.TensorDataset(xs, ys)
+
+    torch.manual_seed(1)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                torch.nn.Linear(16, 4))
+    optim = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    loader = adl.AdaptiveDataLoader(dataset, batch_size=8, shuffle=True)
+    return adl, model, optim, adp, loader
+
+
+def _train(accum_steps, graphed, out_path):
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+
+    collective.initialize()
+    adl, model, optim, adp, loader = _build(accum_steps)
+
+    def fwd_bwd(x, y):
+        optim.zero_grad()
+        loss = F.cross_entropy(adp(x), y)
+        loss.backward()
+        return loss
+
+    stepper = None
+    if graphed:
+        stepper = GraphedStepper(adp, optim, fwd_bwd,
+                                 backend=EagerBackend(), warmup_cycles=1)
+
+    for _epoch in adl.remaining_epochs_until(3):
+        for x, y in loader:
+            if stepper is not None:
+                stepper.microbatch(x, y)
+            else:
+                fwd_bwd(x, y)
+            optim.step()
+
+    gns_state = optim.state["gns"]
+    result = {
+        "weights": [p.detach().clone() for p in model.parameters()],
+        "sqr_avg": np.array(gns_state["sqr_avg"]),
+        "var_avg": np.array(gns_state["var_avg"]),
+        "progress": float(gns_state["progress"]),
+        "gain": float(adp.gain),
+        "stats": dict(stepper.stats) if stepper is not None else None,
+    }
+    torch.save(result, out_path)
+    collective.teardown()
+    return 0
+
+
+@elastic_multiprocessing
+def _train_child(accum_steps, graphed, out_path):
+    return _train(accum_steps, graphed, out_path)
+
+
+def _compare(tmp_path, accum_steps, expect_stats):
+    eager_path = str(tmp_path / "eager.pt")
+    graph_path = str(tmp_path / "graphed.pt")
+    _train_child(accum_steps, False, eager_path)
+    _train_child(accum_steps, True, graph_path)
+    eager = torch.load(eager_path, weights_only=False)
+    graphed = torch.load(graph_path, weights_only=False)
+
+    for we, wg in zip(eager["weights"], graphed["weights"]):
+        assert torch.equal(we, wg), "weights diverged"
+    np.testing.assert_allclose(eager["sqr_avg"], graphed["sqr_avg"])
+    np.testing.assert_allclose(eager["var_avg"], graphed["var_avg"])
+    assert eager["progress"] == graphed["progress"]
+    assert eager["gain"] == graphed["gain"]
+    assert graphed["stats"] == expect_stats
+
+
+def test_graphed_equals_eager_with_accumulation(tmp_path):
+    """A=2 cycle: kinds first/mid/sync; bit-identical to eager.
+
+    Expected stats over 3 epochs x 6 microbatches: mb1 runs before the
+    loader is marked training (eager), which consumes one accumulation
+    slot and shortens the first cycle — the stepper detects the desync
+    at mb3 and falls back for that cycle (1 fallback, by design); mb4-6
+    are the warmup cycle, mb7-9 capture the three kinds, mb10-18 replay.
+    """
+    _compare(tmp_path, accum_steps=2, expect_stats={
+        "captures": 3, "replays": 9, "eager": 6, "fallbacks": 1})
+
+
+def test_graphed_equals_eager_no_accumulation(tmp_path):
+    """A=0 cycle: single "solo" kind (differenced GNS estimator path).
+
+    mb1 eager (loader not yet marked training), mb2 warmup cycle, mb3
+    captures, mb4-18 replay; the solo cycle never desyncs.
+    """
+    _compare(tmp_path, accum_steps=0, expect_stats={
+        "captures": 1, "replays": 15, "eager": 2, "fallbacks": 0})
+
+
+@elastic_multiprocessing
+def _train_resize_child(out_path):
+    """Batch-size change mid-training: graphs must drop and re-capture."""
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+    from adaptdl_amd.torch.data import AdaptiveDataLoaderHelper
+
+    collective.initialize()
+    adl, model, optim, adp, loader = _build(accum_steps=0)
+
+    def fwd_bwd(x, y):
+        optim.zero_grad()
+        loss = F.cross_entropy(adp(x), y)
+        loss.backward()
+        return loss
+
+    stepper = GraphedStepper(adp, optim, fwd_bwd,
+                             backend=EagerBackend(), warmup_cycles=1)
+
+    for epoch in adl.remaining_epochs_until(4):
+        if epoch == 2:
+            # Grow the atomic batch size: new shapes from epoch 2 on.
+            def fake_sync(self):
+                self._state.current_local_bsz = 16
+                self._state.accumulation_steps = 0
+                return 16
+            AdaptiveDataLoaderHelper._sync_local_bsz = fake_sync
+        for x, y in loader:
+            stepper.microbatch(x, y)
+            optim.step()
+        for p in model.parameters():
+            assert torch.isfinite(p).all()
+
+    # Two signatures -> two capture generations, each preceded by one
+    # warmup cycle; no desyncs.
+    assert stepper.stats["captures"] == 2, stepper.stats
+    assert stepper.stats["fallbacks"] == 0, stepper.stats
+    assert stepper.stats["replays"] > 0
+    torch.save({"ok": True}, out_path)
+    collective.teardown()
+    return 0
+
+
+def test_graphed_recaptures_on_batch_size_change(tmp_path):
+    out = str(tmp_path / "resize.pt")
+    _train_resize_child(out)
+    assert torch.load(out, weights_only=False)["ok"]
+
+
+def test_env_gate_off_returns_none():
+    """maybe_graphed_stepper is a no-op without ADAPTDL_HIPGRAPH=1."""
+    from adaptdl_amd.torch.graph_step import maybe_graphed_stepper
+    assert os.getenv("ADAPTDL_HIPGRAPH") != "1"
+    assert maybe_graphed_stepper(None, None, None) is None

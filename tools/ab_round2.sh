@@ -16,8 +16,15 @@ ADAPTDL_EXPERIMENTAL_S2_FWD=1 timeout 180 python -m pytest \
 timeout 180 python tools/conv_time.py > gpurun_out/ab_conv_time.log 2>&1
 tail -8 gpurun_out/ab_conv_time.log
 
+# 2b. hipGraph stepper: capture-vs-eager numerics A/B on hardware
+# (tests/test_gpu_e2e.py::test_hipgraph_stepper_matches_eager).
+ADAPTDL_HIPGRAPH=1 timeout 240 python -m pytest \
+    tests/test_gpu_e2e.py -q -k hipgraph -m gpu \
+    2>&1 | tail -3 | tee gpurun_out/ab_hipgraph.log
+
 # 3. In-context bench A/B: baseline, then each gated path alone.
-for cfg in "" "ADAPTDL_S2_1X1=1" "ADAPTDL_S2_WRW=1" "ADAPTDL_S2_W8B=1"; do
+for cfg in "" "ADAPTDL_S2_1X1=1" "ADAPTDL_S2_WRW=1" "ADAPTDL_S2_W8B=1" \
+           "ADAPTDL_HIPGRAPH=1"; do
     name=${cfg:-baseline}; name=${name%%=*}
     env $cfg timeout 260 python bench.py --steps 20 --warmup 12 \
         > "gpurun_out/ab_bench_${name}.log" 2>&1
