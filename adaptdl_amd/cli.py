@@ -12,6 +12,7 @@ worker processes under adaptdl_amd.sched.LocalController:
   adaptdl-amd logs NAME [--rank R]                # print worker logs
   adaptdl-amd rescale NAME N                      # force a replica count
   adaptdl-amd stop                                # shut the daemon down
+  adaptdl-amd tensorboard NAME|--logdir DIR       # TensorBoard on a job dir
 
 The daemon exposes a small JSON/HTTP admin API (POST /jobs, GET /jobs,
 GET /jobs/{name}, GET /jobs/{name}/logs, POST /jobs/{name}/rescale,
@@ -242,6 +243,36 @@ def cmd_stop(args):
     print("daemon stopping")
 
 
+def cmd_tensorboard(args):
+    """Point TensorBoard at a job's directory.
+
+    Counterpart of the reference's ``adaptdl tensorboard`` (which
+    manages an in-cluster TensorBoard Deployment per instance,
+    cli/adaptdl_cli/tensorboard.py); locally the job_dir already holds
+    the event files written by the ``to_tensorboard`` exporters, so
+    this resolves the directory (from the daemon, or --logdir) and
+    execs ``tensorboard`` against it — or prints the exact command if
+    the tensorboard package is not installed.
+    """
+    import shutil
+    if args.logdir:
+        logdir = args.logdir
+    else:
+        if not args.name:
+            raise SystemExit("tensorboard: a job NAME or --logdir "
+                             "is required")
+        logdir = _request("{}/jobs/{}".format(args.url,
+                                              args.name))["job_dir"]
+    argv = ["tensorboard", "--logdir", logdir,
+            "--port", str(args.port)]
+    exe = shutil.which("tensorboard")
+    if exe is None:
+        print("tensorboard is not installed; run:")
+        print("  " + " ".join(argv))
+        return
+    os.execv(exe, argv)
+
+
 def main(argv=None):
     argv, command = _split_command(list(argv
                                         if argv is not None
@@ -290,6 +321,14 @@ def main(argv=None):
     p = sub.add_parser("stop", help="shut the daemon down")
     p.add_argument("--url", default=DEFAULT_URL)
 
+    p = sub.add_parser("tensorboard",
+                       help="launch TensorBoard on a job's directory")
+    p.add_argument("name", nargs="?")
+    p.add_argument("--logdir", default=None,
+                   help="explicit log directory (no daemon needed)")
+    p.add_argument("--port", type=int, default=6006)
+    p.add_argument("--url", default=DEFAULT_URL)
+
     args = parser.parse_args(argv)
     if args.cmd == "daemon":
         cmd_daemon(args)
@@ -305,6 +344,8 @@ def main(argv=None):
         cmd_rescale(args)
     elif args.cmd == "stop":
         cmd_stop(args)
+    elif args.cmd == "tensorboard":
+        cmd_tensorboard(args)
 
 
 def _has_gpu():

@@ -174,7 +174,8 @@ class LocalController(object):
             job = self._jobs[name]
             return {"state": job.state, "replicas": job.num_replicas,
                     "restarts": job.num_restarts,
-                    "allocation": list(job.allocation)}
+                    "allocation": list(job.allocation),
+                    "job_dir": job.spec.job_dir}
 
     def jobs(self):
         with self._lock:

@@ -67,6 +67,7 @@ def test_daemon_submit_ls_logs(tmp_path):
             assert time.time() < deadline, st
             time.sleep(0.3)
         assert st["state"] == "Succeeded"
+        assert st["job_dir"]  # consumed by `adaptdl-amd tensorboard`
 
         jobs = _req(url + "/jobs")
         assert "hello" in jobs
@@ -99,3 +100,24 @@ def test_cli_run_foreground(tmp_path):
     assert p.returncode == 0, p.stdout + p.stderr
     assert "Succeeded" in p.stdout
     assert "ran fine" in p.stdout
+
+
+def test_tensorboard_command_prints_fallback(tmp_path, capsys,
+                                             monkeypatch):
+    """`adaptdl-amd tensorboard --logdir D` without the tensorboard
+    package prints the command instead of exec'ing it."""
+    import shutil
+    from adaptdl_amd import cli
+    monkeypatch.setattr(shutil, "which", lambda name: None)
+    cli.main(["tensorboard", "--logdir", str(tmp_path), "--port",
+              "7007"])
+    out = capsys.readouterr().out
+    assert "tensorboard is not installed" in out
+    assert "--logdir {}".format(tmp_path) in out
+    assert "--port 7007" in out
+
+
+def test_tensorboard_command_requires_name_or_logdir():
+    from adaptdl_amd import cli
+    with pytest.raises(SystemExit):
+        cli.main(["tensorboard"])
