@@ -41,3 +41,35 @@ def test_bench_single_process_json_contract(tmp_path):
     for field in ("model", "global_batch", "parallelism"):
         assert field in cfg, field
     assert cfg["parallelism"] == "dp1"
+
+
+def test_bench_two_rank_torchrun_contract(tmp_path):
+    """The driver's N>1 launch: torch.distributed.run with 2 CPU/gloo
+    ranks through the FULL bench (warmup, fit, probe, timed region,
+    max-over-ranks aggregation, rank-0-only JSON)."""
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    env = dict(os.environ)
+    env["ADAPTDL_CHECKPOINT_PATH"] = str(tmp_path)
+    env["OMP_NUM_THREADS"] = "1"
+    env.pop("ADAPTDL_NUM_REPLICAS", None)
+    env.pop("ADAPTDL_REPLICA_RANK", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "5",
+         "--init-batch", "64", "--max-batch", "128",
+         "--dataset-size", "512", "--pool", "64"],
+        env=env, cwd=REPO, capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, out.stdout[-2000:]  # rank 0 only
+    result = json.loads(lines[0])
+    assert result["n_gpus"] == 2
+    assert result["config"]["parallelism"] == "dp2"
+    assert result["value"] > 0
+    assert result["ms_per_step"] > 0
