@@ -7,7 +7,7 @@ position tracking, signature-change fallback.  The EagerBackend executes
 the full Python path with ``engine.graph_mode`` set — exactly the deferred
 flow — so training through it must be bit-identical to plain eager
 training.  (The real capture backend is validated by the env-gated GPU
-test in tests/test_gpu_kernels.py.)
+test in tests/test_gpu_e2e.py.)
 """
 
 import os
@@ -186,3 +186,96 @@ def test_env_gate_off_returns_none():
     from adaptdl_amd.torch.graph_step import maybe_graphed_stepper
     assert os.getenv("ADAPTDL_HIPGRAPH") != "1"
     assert maybe_graphed_stepper(None, None, None) is None
+
+
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+@elastic_multiprocessing
+def _train_two_replicas(graphed, out_path):
+    """2-replica gloo training: the stepper's bookkeeping must hold with
+    real collectives in the cycle (per-bucket all-reduce at the sync
+    position, stats all-reduce, module-state broadcast)."""
+    import adaptdl_amd.env as env
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+
+    collective.initialize()
+    if env.num_restarts() == 0:
+        collective.teardown()
+        return 2
+    torch.distributed.init_process_group(
+        "gloo", init_method="tcp://127.0.0.1:{}".format(
+            collective.broadcast(_free_port())),
+        world_size=env.num_replicas(), rank=env.replica_rank())
+
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.data import AdaptiveDataLoaderHelper
+
+    def fake_sync(self):
+        self._state.current_local_bsz = 8
+        self._state.accumulation_steps = 2
+        return 8
+
+    AdaptiveDataLoaderHelper._sync_local_bsz = fake_sync
+
+    torch.manual_seed(0)
+    xs = torch.randn(48, 8)
+    ys = torch.randint(0, 4, (48,))
+    dataset = torch.utils.data.TensorDataset(xs, ys)
+    torch.manual_seed(1 + env.replica_rank())  # broadcast must fix this
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                torch.nn.Linear(16, 4))
+    optim = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    loader = adl.AdaptiveDataLoader(dataset, batch_size=16)
+
+    def fwd_bwd(x, y):
+        optim.zero_grad()
+        loss = F.cross_entropy(adp(x), y)
+        loss.backward()
+        return loss
+
+    stepper = GraphedStepper(adp, optim, fwd_bwd,
+                             backend=EagerBackend(), warmup_cycles=1) \
+        if graphed else None
+
+    for _epoch in adl.remaining_epochs_until(4):
+        for x, y in loader:
+            if stepper is not None:
+                stepper.microbatch(x, y)
+            else:
+                fwd_bwd(x, y)
+            optim.step()
+
+    if env.replica_rank() == 0:
+        gns_state = optim.state["gns"]
+        torch.save({
+            "weights": [p.detach().clone() for p in model.parameters()],
+            "sqr_avg": np.array(gns_state["sqr_avg"]),
+            "var_avg": np.array(gns_state["var_avg"]),
+            "stats": dict(stepper.stats) if stepper is not None else None,
+        }, out_path)
+    torch.distributed.destroy_process_group()
+    collective.teardown()
+    return 0
+
+
+def test_graphed_equals_eager_two_replicas(tmp_path):
+    eager_path = str(tmp_path / "eager2.pt")
+    graph_path = str(tmp_path / "graphed2.pt")
+    _train_two_replicas(False, eager_path)
+    _train_two_replicas(True, graph_path)
+    eager = torch.load(eager_path, weights_only=False)
+    graphed = torch.load(graph_path, weights_only=False)
+    for we, wg in zip(eager["weights"], graphed["weights"]):
+        assert torch.equal(we, wg), "weights diverged"
+    np.testing.assert_allclose(eager["sqr_avg"], graphed["sqr_avg"])
+    np.testing.assert_allclose(eager["var_avg"], graphed["var_avg"])
+    # 4 epochs x 3 microbatches/replica; mb1 untrained + one desync
+    # cycle (see test_graphed_equals_eager_with_accumulation).
+    assert graphed["stats"] == {"captures": 3, "replays": 3,
+                                "eager": 6, "fallbacks": 1}
