@@ -28,6 +28,8 @@ from adaptdl_amd.torch.accumulator import Accumulator  # noqa
 from adaptdl_amd.torch.iterator import AdaptiveBPTTIterator  # noqa
 from adaptdl_amd.torch.optim import (FusedSGD, FusedAdam,  # noqa
                                      FusedAdamW)
+from adaptdl_amd.torch.graph_step import (GraphedStepper,  # noqa
+                                          maybe_graphed_stepper)
 
 LOG = logging.getLogger(__name__)
 
@@ -47,6 +49,8 @@ __all__ = [
     "FusedSGD",
     "FusedAdam",
     "FusedAdamW",
+    "GraphedStepper",
+    "maybe_graphed_stepper",
 ]
 
 

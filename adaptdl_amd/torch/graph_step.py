@@ -1,9 +1,10 @@
 """hipGraph capture of the steady-state training microbatch cycle.
 
 EXPERIMENTAL, env-gated with ``ADAPTDL_HIPGRAPH=1`` (ROADMAP item 5):
-trace-union analysis of the flagship bench (profiles/bench_r08.json
-discussion) measured 4.3% of step time as GPU idle gaps between the many
-small kernel launches of forward+backward.  Capturing the pinned
+a trace-union analysis of the flagship bench measured ~4.3% of step
+time as GPU idle gaps between the many small kernel launches of
+forward+backward (earlier-session scratch trace; re-measured by the
+round-2 A/B in tools/ab_round2.sh).  Capturing the pinned
 steady-state microbatch into a hipGraph replays the whole kernel sequence
 with one launch, closing those gaps.
 

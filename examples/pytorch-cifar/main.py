@@ -77,6 +77,21 @@ def main():
                                     local_bsz_bounds=(32, 1024),
                                     gradient_accumulation=True)
 
+    def fwd_bwd(x, y):
+        optim.zero_grad()
+        if use_gpu:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                loss = F.cross_entropy(adp(x), y)
+        else:
+            loss = F.cross_entropy(adp(x), y)
+        loss.backward()
+        return loss
+
+    # Experimental: ADAPTDL_HIPGRAPH=1 captures the steady microbatch
+    # cycle into hipGraphs (see adaptdl_amd/torch/graph_step.py).
+    from adaptdl_amd.torch.graph_step import maybe_graphed_stepper
+    stepper = maybe_graphed_stepper(adp, optim, fwd_bwd)
+
     stats = adl.Accumulator()
     for epoch in adl.remaining_epochs_until(args.epochs):
         model.train()
@@ -86,16 +101,14 @@ def main():
             x, y = x.to(device), y.to(device)
             if use_gpu:
                 x = x.contiguous(memory_format=torch.channels_last)
-            optim.zero_grad()
-            if use_gpu:
-                with torch.autocast("cuda", dtype=torch.bfloat16):
-                    loss = F.cross_entropy(adp(x), y)
+            if stepper is not None:
+                loss = stepper.microbatch(x, y)
             else:
-                loss = F.cross_entropy(adp(x), y)
-            loss.backward()
+                loss = fwd_bwd(x, y)
             optim.step()
-            stats["loss_sum"] += loss.item() * len(y)
-            stats["count"] += len(y)
+            if loss is not None:
+                stats["loss_sum"] += loss.item() * len(y)
+                stats["count"] += len(y)
         sched.step()
         with stats.synchronized():
             if int(os.getenv("ADAPTDL_REPLICA_RANK", "0")) == 0:
