@@ -279,3 +279,90 @@ def test_graphed_equals_eager_two_replicas(tmp_path):
     # cycle (see test_graphed_equals_eager_with_accumulation).
     assert graphed["stats"] == {"captures": 3, "replays": 3,
                                 "eager": 6, "fallbacks": 1}
+
+
+@elastic_multiprocessing
+def _train_with_restart(graphed, out_path):
+    """Checkpoint-restart elasticity x graph stepping: rescale 1 -> 2
+    replicas at an epoch boundary; the fresh processes rebuild a fresh
+    stepper (warm + recapture) while model/optimizer/GNS state resumes
+    from the checkpoint.  Final weights must match the eager run with
+    the identical restart schedule."""
+    import adaptdl_amd.checkpoint as checkpoint
+    import adaptdl_amd.env as env
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+
+    collective.initialize()
+    torch.distributed.init_process_group(
+        "gloo", init_method="tcp://127.0.0.1:{}".format(
+            collective.broadcast(_free_port())),
+        world_size=env.num_replicas(), rank=env.replica_rank())
+
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.data import AdaptiveDataLoaderHelper
+
+    def fake_sync(self):
+        self._state.current_local_bsz = 8
+        self._state.accumulation_steps = 2
+        return 8
+
+    AdaptiveDataLoaderHelper._sync_local_bsz = fake_sync
+
+    torch.manual_seed(0)
+    xs = torch.randn(48, 8)
+    ys = torch.randint(0, 4, (48,))
+    dataset = torch.utils.data.TensorDataset(xs, ys)
+    torch.manual_seed(1 + env.replica_rank())
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                torch.nn.Linear(16, 4))
+    optim = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    loader = adl.AdaptiveDataLoader(dataset, batch_size=16)
+
+    def fwd_bwd(x, y):
+        optim.zero_grad()
+        loss = F.cross_entropy(adp(x), y)
+        loss.backward()
+        return loss
+
+    stepper = GraphedStepper(adp, optim, fwd_bwd,
+                             backend=EagerBackend(), warmup_cycles=1) \
+        if graphed else None
+
+    for epoch in adl.remaining_epochs_until(6):
+        for x, y in loader:
+            if stepper is not None:
+                stepper.microbatch(x, y)
+            else:
+                fwd_bwd(x, y)
+            optim.step()
+        if env.num_restarts() == 0 and epoch == 1:
+            checkpoint.save_all_states()
+            torch.distributed.destroy_process_group()
+            collective.teardown()
+            return 2  # rescale to two replicas
+
+    if env.replica_rank() == 0:
+        torch.save({
+            "weights": [p.detach().clone() for p in model.parameters()],
+            "restarts": env.num_restarts(),
+            "stats": dict(stepper.stats) if stepper is not None else None,
+        }, out_path)
+    torch.distributed.destroy_process_group()
+    collective.teardown()
+    return 0
+
+
+def test_graphed_restart_rescale_matches_eager(tmp_path):
+    eager_path = str(tmp_path / "re.pt")
+    graph_path = str(tmp_path / "rg.pt")
+    _train_with_restart(False, eager_path)
+    _train_with_restart(True, graph_path)
+    eager = torch.load(eager_path, weights_only=False)
+    graphed = torch.load(graph_path, weights_only=False)
+    assert eager["restarts"] == graphed["restarts"] == 1
+    for we, wg in zip(eager["weights"], graphed["weights"]):
+        assert torch.equal(we, wg), "weights diverged after rescale"
+    # The post-restart processes re-captured from scratch.
+    assert graphed["stats"]["captures"] == 3
+    assert graphed["stats"]["replays"] > 0
