@@ -55,7 +55,9 @@ Safety rules (all checked per cycle; violation falls back to eager):
   re-warms,
 - the observed sync position must match the expected one,
 - AMP GradScaler (mp_scaler) is not supported (its inf-check host logic
-  inspects gradients between backward and step); construction refuses,
+  inspects gradients between backward and step), and neither is the
+  Adam-preconditioned GNS (its statistic kernels take the per-step bias
+  correction as a kernel argument); construction refuses both,
 - any capture error permanently disables the stepper for the run (the
   current microbatch is re-run eagerly).
 
@@ -154,9 +156,20 @@ class GraphedStepper(object):
 
     def __init__(self, adp, optimizer, fwd_bwd, backend=None,
                  warmup_cycles=1):
+        from adaptdl_amd.torch.gradient_noise_scale import \
+            AdamGradientNoiseScale
         if getattr(adp.gns, "_mp_scaler", None) is not None:
             raise ValueError("GraphedStepper does not support mp_scaler "
                              "(GradScaler host logic is not capturable)")
+        if isinstance(adp.gns, AdamGradientNoiseScale):
+            # The preconditioned statistic kernels take the Adam step
+            # count as a kernel argument (bias correction changes every
+            # step) and read it with a host sync — a captured graph
+            # would bake a stale step in.  SGD-family only for now.
+            raise ValueError("GraphedStepper does not support Adam-"
+                             "preconditioned GNS (per-step bias "
+                             "correction is baked into captured "
+                             "statistic kernels)")
         self._adp = adp
         self._gns = adp.gns
         self._engine = adp.gns.engine

@@ -366,3 +366,38 @@ def test_graphed_restart_rescale_matches_eager(tmp_path):
     # The post-restart processes re-captured from scratch.
     assert graphed["stats"]["captures"] == 3
     assert graphed["stats"]["replays"] > 0
+
+
+def test_refuses_adam_preconditioned_gns(tmp_ckpt_env):
+    """Adam-preconditioned statistics bake the per-step bias correction
+    into captured kernels; construction must refuse (and the env-gated
+    factory must decline gracefully)."""
+    import pytest
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+    model = torch.nn.Linear(4, 2)
+    optim = torch.optim.Adam(model.parameters(), lr=1e-3)
+    adp = adl.AdaptiveDataParallel(model, optim, name="adam-graph-test")
+    with pytest.raises(ValueError, match="Adam"):
+        GraphedStepper(adp, optim, lambda *a: None,
+                       backend=EagerBackend())
+
+
+def test_refuses_mp_scaler(tmp_ckpt_env):
+    import pytest
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
+
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+    model = torch.nn.Linear(4, 2)
+    optim = torch.optim.SGD(model.parameters(), lr=0.1)
+    scaler = torch.amp.GradScaler("cuda", enabled=False)
+    adp = adl.AdaptiveDataParallel(model, optim, mp_scaler=scaler,
+                                   name="scaler-graph-test")
+    with pytest.raises(ValueError, match="mp_scaler"):
+        GraphedStepper(adp, optim, lambda *a: None,
+                       backend=EagerBackend())
