@@ -121,3 +121,31 @@ def test_tensorboard_command_requires_name_or_logdir():
     from adaptdl_amd import cli
     with pytest.raises(SystemExit):
         cli.main(["tensorboard"])
+
+
+def test_trace_gaps_union_math(tmp_path):
+    """tools/trace_gaps.py: union-busy / idle accounting on a synthetic
+    kernel trace (overlaps must not double-count)."""
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    import trace_gaps
+
+    # Span 0..1000ns; kernels: [0,100), [50,200) (overlap), [300,400),
+    # [900,1000).  Union busy = 200+100+100 = 400; idle = 600.
+    csv_path = tmp_path / "t_kernel_trace.csv"
+    csv_path.write_text(
+        "Kernel_Name,Start_Timestamp,End_Timestamp\n"
+        "a,0,100\nb,50,200\nc,300,400\nd,900,1000\n")
+    out = trace_gaps.analyze(
+        trace_gaps.load_intervals([str(csv_path)]), tail=1.0)
+    assert out["kernels"] == 4
+    assert abs(out["busy_ms"] - 400e-6) < 1e-12
+    assert abs(out["idle_ms"] - 600e-6) < 1e-12
+    assert abs(out["idle_pct"] - 60.0) < 1e-9
+    # Largest gap is 500ns (400 -> 900).
+    assert out["top_gaps_us"][0] == 0.5
+
+    # Tail window 0.5: cut at 500; only [900,1000) remains -> idle 400.
+    out = trace_gaps.analyze(
+        trace_gaps.load_intervals([str(csv_path)]), tail=0.5)
+    assert out["kernels"] == 1
+    assert abs(out["idle_pct"] - 80.0) < 1e-9

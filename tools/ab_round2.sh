@@ -31,3 +31,18 @@ for cfg in "" "ADAPTDL_S2_1X1=1" "ADAPTDL_S2_WRW=1" "ADAPTDL_S2_W8B=1" \
     grep '"metric"' "gpurun_out/ab_bench_${name}.log" | tail -1
 done
 # Keep only measured wins (the bench_r10 rule); see ROADMAP.md item 3.
+
+# 4. Idle-gap evidence for the hipGraph A/B: kernel traces of the
+# eager and graphed steady states, summarized with tools/trace_gaps.py
+# (copy the summaries into profiles/ when committing).
+cd /tmp && export TMPDIR=/tmp && cd - >/dev/null
+for cfg in "" "ADAPTDL_HIPGRAPH=1"; do
+    name=${cfg:-eager}; name=${name%%=*}
+    env $cfg timeout 300 rocprofv3 --kernel-trace \
+        -d "gpurun_out/trace_${name}" -- \
+        python bench.py --steps 15 --warmup 10 \
+        > "gpurun_out/trace_${name}.log" 2>&1
+    python tools/trace_gaps.py gpurun_out/trace_${name}/*/*kernel_trace*.csv \
+        > "gpurun_out/gaps_${name}.txt" 2>&1 || true
+    tail -5 "gpurun_out/gaps_${name}.txt"
+done
