@@ -376,8 +376,6 @@ def test_refuses_adam_preconditioned_gns(tmp_ckpt_env):
     import adaptdl_amd.torch as adl
     from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
 
-    if not collective.initialized():
-        collective.initialize(master_addr="127.0.0.1")
     model = torch.nn.Linear(4, 2)
     optim = torch.optim.Adam(model.parameters(), lr=1e-3)
     adp = adl.AdaptiveDataParallel(model, optim, name="adam-graph-test")
@@ -391,8 +389,6 @@ def test_refuses_mp_scaler(tmp_ckpt_env):
     import adaptdl_amd.torch as adl
     from adaptdl_amd.torch.graph_step import GraphedStepper, EagerBackend
 
-    if not collective.initialized():
-        collective.initialize(master_addr="127.0.0.1")
     model = torch.nn.Linear(4, 2)
     optim = torch.optim.SGD(model.parameters(), lr=0.1)
     scaler = torch.amp.GradScaler("cuda", enabled=False)
