@@ -397,3 +397,13 @@ def test_refuses_mp_scaler(tmp_ckpt_env):
     with pytest.raises(ValueError, match="mp_scaler"):
         GraphedStepper(adp, optim, lambda *a: None,
                        backend=EagerBackend())
+
+
+def test_env_gate_on_without_gpu_returns_none(monkeypatch):
+    """ADAPTDL_HIPGRAPH=1 on a CPU-only host declines gracefully."""
+    from adaptdl_amd.torch.graph_step import maybe_graphed_stepper
+    monkeypatch.setenv("ADAPTDL_HIPGRAPH", "1")
+    if torch.cuda.is_available():  # pragma: no cover - CPU CI only
+        import pytest
+        pytest.skip("CPU-only path")
+    assert maybe_graphed_stepper(None, None, None) is None
