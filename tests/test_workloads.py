@@ -125,3 +125,24 @@ def _run_mp_scaler_path():
 
 def test_mp_scaler_statistics_unscaled():
     _run_mp_scaler_path()
+
+
+def test_cifar_example_script_cpu_smoke(tmp_path):
+    """Run examples/pytorch-cifar/main.py end-to-end on CPU (1 epoch,
+    synthetic, no autoscale): guards the script itself — the underlying
+    paths have their own tests, but edits to the example (e.g. the
+    graph-stepper wiring) only showed up on GPU before this."""
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "examples", "pytorch-cifar",
+                                      "main.py"),
+         "--epochs", "1", "--samples", "128", "--bs", "64",
+         "--max-bs", "0"],
+        env=dict(os.environ, ADAPTDL_CHECKPOINT_PATH=str(tmp_path),
+                 OMP_NUM_THREADS="2", PYTHONPATH=repo),
+        cwd=repo, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "epoch 0" in out.stdout
