@@ -146,14 +146,17 @@ def test_hipgraph_stepper_matches_eager(tmp_ckpt_env):
         y = torch.randint(0, 10, (32,), device=device)
         x = x.contiguous(memory_format=torch.channels_last)
         results = {}
-        for graphed in (False, True):
+        # Three runs: two eager (their divergence calibrates the
+        # nondeterminism of the backward kernels) and one graphed.
+        for run_id, graphed in (("eager", False), ("eager2", False),
+                                ("graphed", True)):
             torch.manual_seed(7)
             model = ResNet18().to(device) \
                 .to(memory_format=torch.channels_last)
             optim = adl.FusedSGD(model.parameters(), lr=0.05,
                                  momentum=0.9)
             adp = adl.AdaptiveDataParallel(
-                model, optim, name="hipgraph-%s" % graphed)
+                model, optim, name="hipgraph-%s" % run_id)
             dataset = torch.utils.data.TensorDataset(torch.arange(384))
             loader = adl.AdaptiveDataLoader(dataset, batch_size=32)
 
@@ -174,23 +177,31 @@ def test_hipgraph_stepper_matches_eager(tmp_ckpt_env):
                         fwd_bwd(x, y)
                     optim.step()
             torch.cuda.synchronize()
-            results[graphed] = {
+            results[run_id] = {
                 "weights": [p.detach().float().cpu()
                             for p in model.parameters()],
                 "sqr": adp.gns.sqr_avg(), "var": adp.gns.var_avg(),
                 "stats": dict(stepper.stats) if stepper else None,
             }
 
-        stats = results[True]["stats"]
+        stats = results["graphed"]["stats"]
         assert stats["captures"] == 3, stats
         assert stats["replays"] >= 24, stats
         assert stats["fallbacks"] <= 1, stats  # first-pass desync only
-        for we, wg in zip(results[False]["weights"],
-                          results[True]["weights"]):
+        # Graphed-vs-eager divergence must be of the same order as
+        # eager-vs-eager (backward kernels may be nondeterministic, so
+        # an absolute tolerance would be arbitrary).
+        noise = max(
+            (we - w2).abs().max().item()
+            for we, w2 in zip(results["eager"]["weights"],
+                              results["eager2"]["weights"]))
+        budget = max(10.0 * noise, 5e-3)
+        for we, wg in zip(results["eager"]["weights"],
+                          results["graphed"]["weights"]):
             assert torch.isfinite(wg).all()
-            assert torch.allclose(we, wg, rtol=5e-2, atol=5e-3), \
-                (we - wg).abs().max()
-        assert np.isfinite(results[True]["sqr"])
-        assert np.isfinite(results[True]["var"])
+            diff = (we - wg).abs().max().item()
+            assert diff <= budget, (diff, noise)
+        assert np.isfinite(results["graphed"]["sqr"])
+        assert np.isfinite(results["graphed"]["var"])
     finally:
         AdaptiveDataLoaderHelper._sync_local_bsz = orig_sync
