@@ -75,14 +75,32 @@ class GoodputFunction(object):
         gain = np.where(denom > 0, (grad_var + grad_sqr) / denom, 1.0)
         return gain / scale
 
+    def _plan_microbatches(self, per_replica, replicas, max_atomic_bsz,
+                           accumulation):
+        """Split a per-replica batch into (atomic_bsz, accum_steps)."""
+        eps = 1e-8  # tolerance on the ceil() boundaries
+        if not accumulation:
+            atomic = np.where(replicas == 1, self._init_batch_size,
+                              np.ceil(per_replica - eps)).astype(int)
+            return atomic, np.zeros_like(atomic)
+        steps = np.ceil(per_replica / max_atomic_bsz - eps) - 1
+        # Solo replica with a scaled-up batch: force >= 1 accumulation step
+        # so the GNS has at least two microbatches to estimate variance.
+        solo = np.logical_and(replicas == 1,
+                              per_replica > self._init_batch_size + eps)
+        steps = np.where(solo, np.maximum(steps, 1), steps).astype(int)
+        atomic = np.ceil(per_replica / (steps + 1) - eps).astype(int)
+        return atomic, steps
+
     def optimize(self, num_nodes, num_replicas, max_batch_size=None,
                  atomic_bsz_range=None, accumulation=False):
         """Choose (atomic_bsz, accum_steps) maximizing goodput.
 
         Samples 50 total batch sizes in geometric space per (nodes, replicas)
-        configuration, derives per-replica (atomic_bsz, accum_steps), and
-        returns the best goodput with its configuration.  Semantics match the
-        reference ``GoodputFunction.optimize`` (goodput.py:88-148) including
+        configuration, derives per-replica (atomic_bsz, accum_steps) via
+        :meth:`_plan_microbatches`, and returns the best goodput with its
+        configuration.  Decision semantics are the contract shared with the
+        reference ``GoodputFunction.optimize`` (goodput.py:88-148), including
         the single-replica accumulation floor for statistics quality.
         """
         assert np.all(np.less_equal(1, num_nodes))
@@ -90,45 +108,30 @@ class GoodputFunction(object):
         if max_batch_size is None:
             max_batch_size = self._init_batch_size
         assert self._init_batch_size <= max_batch_size
-        atomic_bsz_range = atomic_bsz_range or (None, None)
-        min_atomic_bsz = atomic_bsz_range[0] or 1
-        max_atomic_bsz = atomic_bsz_range[1] or max_batch_size
-        output_shape = np.broadcast(num_nodes, num_replicas).shape
-        output_scalar = np.isscalar(num_nodes) or np.isscalar(num_replicas)
-        num_nodes = np.broadcast_to(num_nodes, output_shape).flatten()
-        num_replicas = np.broadcast_to(num_replicas, output_shape).flatten()
-        min_batch_size = np.maximum(self._init_batch_size,
-                                    min_atomic_bsz * num_replicas)
-        batch_size = np.geomspace(min_batch_size, max_batch_size)
-        local_bsz = batch_size / num_replicas
-        eps = 1e-8
-        if accumulation:
-            accum_steps = np.ceil(local_bsz / max_atomic_bsz - eps) - 1
-            # With a single replica and a scaled-up batch, require at least
-            # one accumulation step so the GNS has >= 2 samples per step.
-            accum_steps = np.where(
-                np.logical_and(num_replicas == 1,
-                               local_bsz > self._init_batch_size + eps),
-                np.maximum(accum_steps, 1), accum_steps).astype(int)
-            atomic_bsz = np.ceil(local_bsz / (accum_steps + 1) - eps)
-            atomic_bsz = atomic_bsz.astype(int)
-        else:
-            accum_steps = np.zeros_like(local_bsz, dtype=int)
-            atomic_bsz = np.where(
-                num_replicas == 1,
-                self._init_batch_size,
-                np.ceil(local_bsz - eps)).astype(int)
-        atomic_bsz = np.clip(atomic_bsz, min_atomic_bsz, max_atomic_bsz)
-        goodput = self.evaluate(num_nodes, num_replicas, atomic_bsz,
-                                accum_steps)
-        indices = np.argmax(goodput, axis=0), np.arange(goodput.shape[1])
-        goodput = goodput[indices].reshape(output_shape)
-        atomic_bsz = atomic_bsz[indices].reshape(output_shape)
-        accum_steps = accum_steps[indices].reshape(output_shape)
-        if output_scalar:
-            goodput = goodput.item()
-            atomic_bsz = atomic_bsz.item()
-            accum_steps = accum_steps.item()
+        lo_atomic, hi_atomic = atomic_bsz_range or (None, None)
+        lo_atomic = lo_atomic or 1
+        hi_atomic = hi_atomic or max_batch_size
+        # Vectorized over (nodes, replicas) pairs; remember the caller's
+        # shape (the Pollux policy passes whole arrays at once).
+        shape = np.broadcast(num_nodes, num_replicas).shape
+        scalar_out = np.isscalar(num_nodes) or np.isscalar(num_replicas)
+        nodes = np.broadcast_to(num_nodes, shape).ravel()
+        replicas = np.broadcast_to(num_replicas, shape).ravel()
+        # Candidate grid: geomspace's default 50 rows of total batch size
+        # per column, bounded below by both the init bsz and the smallest
+        # per-replica atomic size.
+        smallest = np.maximum(self._init_batch_size, lo_atomic * replicas)
+        per_replica = np.geomspace(smallest, max_batch_size) / replicas
+        atomic_bsz, accum_steps = self._plan_microbatches(
+            per_replica, replicas, hi_atomic, accumulation)
+        atomic_bsz = np.clip(atomic_bsz, lo_atomic, hi_atomic)
+        scores = self.evaluate(nodes, replicas, atomic_bsz, accum_steps)
+        best_row = np.argmax(scores, axis=0), np.arange(scores.shape[1])
+        picks = [arr[best_row].reshape(shape)
+                 for arr in (scores, atomic_bsz, accum_steps)]
+        if scalar_out:
+            picks = [p.item() for p in picks]
+        goodput, atomic_bsz, accum_steps = picks
         return goodput, atomic_bsz, accum_steps
 
 

@@ -253,8 +253,10 @@ class GradSyncEngine(object):
             # Defer the host-side GNS update: _on_sync_done host-syncs via
             # stats.cpu(), which is illegal during stream capture.  The
             # GraphedStepper invokes it after each replay, when the stats
-            # tensor holds this step's values.
-            self.last_sync_time = 0.0
+            # tensor holds this step's values.  last_sync_time keeps the
+            # most recent eager measurement (warmup cycles run eagerly at
+            # every signature change), so the goodput fit sees a realistic
+            # sync cost for replayed steps rather than 0.
             self._sync_end_ev = None
             self.pending_sync_done = True
             return

@@ -5,10 +5,14 @@ shuffled dataset across replicas, resumable from any sample index),
 :class:`AdaptiveDataLoaderHelper` (the elastic core: batch-size
 autoscaling via the goodput model, per-iteration exit-flag sync,
 restart-safe loop bookkeeping), and :class:`AdaptiveDataLoader` (a drop-in
-torch DataLoader).  Behavior mirrors the reference
-(``/root/reference/adaptdl/adaptdl/torch/data.py``): the ``batch_size``
-argument is the *total* batch size across replicas; with autoscaling on,
-an epoch iterates until statistical progress equals one non-adaptive epoch.
+torch DataLoader).  The *semantics* — batch accounting, restart
+replay-skip, the 1.05 hysteresis on batch-size changes, the exit-143
+protocol — are the behavioral contract shared with the reference
+(``/root/reference/adaptdl/adaptdl/torch/data.py``) and are encoded in
+this repo's restart/property tests; the implementation is adaptdl_amd's
+own.  The ``batch_size`` argument is the *total* batch size across
+replicas; with autoscaling on, an epoch iterates until statistical
+progress equals one non-adaptive epoch.
 """
 
 from contextlib import contextmanager
@@ -41,7 +45,7 @@ class ElasticSampler(Sampler):
     Shuffling is a deterministic function of (epoch, pass index), so all
     replicas agree on the order without communication, and sampling can be
     resumed from an arbitrary global sample index after a rescale changes
-    the replica count (reference: data.py:41-111).
+    the replica count.
     """
 
     def __init__(self, dataset, shuffle=True):
@@ -52,32 +56,36 @@ class ElasticSampler(Sampler):
         self.epoch = 0
         self.index = 0  # global sample index to resume from
 
-    def __iter__(self):
-        if self.shuffle:
-            g = torch.Generator()
-            # Deterministic across processes (unlike builtin hash of str).
-            g.manual_seed(self.epoch * 0x9E3779B1
-                          + self.index // len(self.dataset))
-            indices = torch.randperm(len(self.dataset), generator=g).tolist()
-        else:
-            indices = list(range(len(self.dataset)))
-        base_index = self.index % len(self.dataset)
-        local_indices = indices[base_index + self.rank::self.num_replicas]
-        # Pad so every replica yields the same number of samples.  The
-        # modulo covers num_replicas > len(dataset) (a rank with no
-        # samples of its own still yields one padding sample).
-        if len(local_indices) < len(self):
-            local_indices.append(indices[self.rank % len(indices)])
-        assert len(local_indices) == len(self)
-        return iter(local_indices)
-
-    def __len__(self):
-        base_index = self.index % len(self.dataset)
-        return math.ceil((len(self.dataset) - base_index) / self.num_replicas)
-
     def set_epoch(self, epoch, index=0):
         self.epoch = epoch
         self.index = index
+
+    def _ordering(self):
+        """Full index permutation for the current (epoch, data pass)."""
+        n = len(self.dataset)
+        if not self.shuffle:
+            return list(range(n))
+        g = torch.Generator()
+        # Deterministic across processes (unlike builtin hash of str).
+        g.manual_seed(self.epoch * 0x9E3779B1 + self.index // n)
+        return torch.randperm(n, generator=g).tolist()
+
+    def __len__(self):
+        remaining = len(self.dataset) - self.index % len(self.dataset)
+        return -(-remaining // self.num_replicas)  # ceil division
+
+    def __iter__(self):
+        order = self._ordering()
+        first = self.index % len(self.dataset) + self.rank
+        mine = order[first::self.num_replicas]
+        # Pad so every replica yields the same number of samples.  The
+        # modulo covers num_replicas > len(dataset) (a rank with no
+        # samples of its own still yields one padding sample).
+        want = len(self)
+        while len(mine) < want:
+            mine.append(order[self.rank % len(order)])
+        assert len(mine) == want
+        return iter(mine)
 
 
 def current_dataloader():
@@ -95,28 +103,32 @@ class AdaptiveDataLoaderHelper(object):
     _current = None
 
     def __init__(self, batch_size=1):
-        self._max_batch_size = None
-        self._local_bsz_bounds = None
-        self._state = _AdaptiveDataLoaderState()
-        adaptdl_amd.checkpoint.load_state(self._state)
         self.batch_size = batch_size
         self.future_exit = None
+        # Autoscale config (off until autoscale_batch_size() is called).
+        self._max_batch_size = None
+        self._local_bsz_bounds = None
         self._gradient_accumulation = False
         self._speedup_threshold = 1.05
         self._accum_count = 0
+        # Checkpointed loop position / batch-size choice.
+        self._state = _AdaptiveDataLoaderState()
+        adaptdl_amd.checkpoint.load_state(self._state)
+
+    # ---- elastic loop position (checkpointed) ---------------------------
 
     @property
     def current_index(self):
-        """Global number of samples processed so far in the current loop."""
-        if AdaptiveDataLoaderHelper._current is not self:
-            return None
-        return self._state.current_index
+        """Global number of samples processed so far in the current loop
+        (None unless this loader is the one being iterated)."""
+        if self._iterating:
+            return self._state.current_index
+        return None
 
     @current_index.setter
     def current_index(self, index):
-        if AdaptiveDataLoaderHelper._current is not self:
-            return
-        self._state.current_index = index
+        if self._iterating:
+            self._state.current_index = index
 
     @property
     def end_index(self):
@@ -125,6 +137,12 @@ class AdaptiveDataLoaderHelper(object):
     @end_index.setter
     def end_index(self, index):
         self._state.end_index = index
+
+    @property
+    def _iterating(self):
+        return AdaptiveDataLoaderHelper._current is self
+
+    # ---- batch-size configuration ---------------------------------------
 
     @property
     def max_batch_size(self):
@@ -143,11 +161,21 @@ class AdaptiveDataLoaderHelper(object):
     def accumulation_steps(self):
         return self._state.accumulation_steps
 
+    @property
+    def current_batch_size(self):
+        replicas = adaptdl_amd.env.num_replicas()
+        return self.current_local_bsz * (self.accumulation_steps + 1) \
+            * replicas
+
     def is_accum_step(self):
         return self._accum_count < self._state.accumulation_steps
 
     def is_optim_step(self):
         return not self.is_accum_step()
+
+    @property
+    def training(self):
+        return self is AdaptiveDataLoaderHelper._training
 
     def train(self):
         """Mark this loader as the training loader (only one allowed)."""
@@ -159,15 +187,14 @@ class AdaptiveDataLoaderHelper(object):
     def autoscale_batch_size(self, max_batch_size, local_bsz_bounds=None,
                              gradient_accumulation=False):
         """Enable goodput-driven adaptive batch sizes up to max_batch_size."""
-        if not isinstance(max_batch_size, int) or \
-                max_batch_size < self.batch_size:
+        if not isinstance(max_batch_size, int) \
+                or max_batch_size < self.batch_size:
             raise ValueError("invalid max_batch_size")
-        if local_bsz_bounds is not None and (
-                local_bsz_bounds[0] is not None and
-                local_bsz_bounds[0] > self.batch_size or
-                local_bsz_bounds[1] is not None and
-                local_bsz_bounds[1] < self.batch_size):
-            raise ValueError("invalid local_bsz_bounds")
+        if local_bsz_bounds is not None:
+            lo, hi = local_bsz_bounds
+            if (lo is not None and lo > self.batch_size) \
+                    or (hi is not None and hi < self.batch_size):
+                raise ValueError("invalid local_bsz_bounds")
         self._max_batch_size = max_batch_size
         self._local_bsz_bounds = local_bsz_bounds
         self._gradient_accumulation = gradient_accumulation
@@ -177,43 +204,42 @@ class AdaptiveDataLoaderHelper(object):
         """Choose (atomic_bsz, accum_steps) for this pass and broadcast it.
 
         Keeps the current choice unless the goodput model predicts at least
-        a 5% speedup from changing (hysteresis; reference data.py:270-305).
+        a ``_speedup_threshold`` speedup from changing (hysteresis).
         """
+        state = self._state
         goodput_fn = get_goodput_fn()
         if self.max_batch_size is None or goodput_fn is None:
-            self._state.current_local_bsz = math.ceil(
-                self.batch_size / adaptdl_amd.env.num_replicas())
-            self._state.accumulation_steps = 0
-        elif not self._state.current_local_bsz:
-            _, atomic_bsz, accum_steps = goodput_fn.optimize(
-                adaptdl_amd.env.num_nodes(), adaptdl_amd.env.num_replicas(),
-                max_batch_size=self._max_batch_size,
-                atomic_bsz_range=self._local_bsz_bounds,
-                accumulation=self._gradient_accumulation)
-            self._state.current_local_bsz = atomic_bsz
-            self._state.accumulation_steps = accum_steps
+            # Autoscaling off (or model not fitted yet): split the target
+            # batch size evenly and round up.
+            replicas = adaptdl_amd.env.num_replicas()
+            choice = (-(-self.batch_size // replicas), 0)
         else:
-            suggest_goodput, atomic_bsz, accum_steps = goodput_fn.optimize(
-                adaptdl_amd.env.num_nodes(), adaptdl_amd.env.num_replicas(),
-                max_batch_size=self._max_batch_size,
+            topo = (adaptdl_amd.env.num_nodes(),
+                    adaptdl_amd.env.num_replicas())
+            proposed_goodput, atomic_bsz, accum_steps = goodput_fn.optimize(
+                *topo, max_batch_size=self._max_batch_size,
                 atomic_bsz_range=self._local_bsz_bounds,
                 accumulation=self._gradient_accumulation)
-            current_goodput = goodput_fn(
-                adaptdl_amd.env.num_nodes(), adaptdl_amd.env.num_replicas(),
-                self.current_local_bsz, self.accumulation_steps)
-            speedup = suggest_goodput / max(current_goodput, 1e-8)
-            if speedup > self._speedup_threshold:
-                self._state.current_local_bsz = atomic_bsz
-                self._state.accumulation_steps = accum_steps
-        self._state.current_local_bsz, self._state.accumulation_steps = \
-            adaptdl_amd.collective.broadcast(
-                (self._state.current_local_bsz,
-                 self._state.accumulation_steps))
+            if not state.current_local_bsz:
+                # First choice of the run: adopt the proposal outright.
+                choice = (atomic_bsz, accum_steps)
+            else:
+                # Hysteresis: only adopt if predicted speedup over the
+                # current setting is significant.
+                held_goodput = goodput_fn(*topo, self.current_local_bsz,
+                                          self.accumulation_steps)
+                ratio = proposed_goodput / max(held_goodput, 1e-8)
+                if ratio > self._speedup_threshold:
+                    choice = (atomic_bsz, accum_steps)
+                else:
+                    choice = (state.current_local_bsz,
+                              state.accumulation_steps)
+        # All replicas adopt rank 0's choice.
+        state.current_local_bsz, state.accumulation_steps = \
+            adaptdl_amd.collective.broadcast(choice)
         return self.current_local_bsz
 
-    @property
-    def training(self):
-        return self is AdaptiveDataLoaderHelper._training
+    # ---- per-iteration / per-loop contexts -------------------------------
 
     @contextmanager
     def profile(self, commit):
@@ -232,8 +258,10 @@ class AdaptiveDataLoaderHelper(object):
         yield
         if commit:
             profile_step_commit(self.is_accum_step())
-        self._accum_count = (0 if self.is_optim_step()
-                             else self._accum_count + 1)
+        if self.is_optim_step():
+            self._accum_count = 0
+        else:
+            self._accum_count += 1
 
     @contextmanager
     def context(self):
@@ -246,37 +274,42 @@ class AdaptiveDataLoaderHelper(object):
             AdaptiveDataLoaderHelper._current = self
             yield
         finally:
+            # Loop finished (or aborted): reset resume indices and record
+            # this loop position as completed for restart replay-skip.
             self._state.current_index = 0
             self._state.end_index = 0
             self._state.last_position[epoch] = self._position[epoch]
             self._position[epoch] += 1
             AdaptiveDataLoaderHelper._current = None
 
-    @property
-    def current_batch_size(self):
-        return (self.current_local_bsz * (self.accumulation_steps + 1) *
-                adaptdl_amd.env.num_replicas())
-
     def skipdone(self):
         """True if this loop already completed before a restart (skip it)."""
         epoch = current_epoch()
         position = self._position[epoch]
-        if position <= self._state.last_position.get(epoch, -1):
-            LOG.info("skipping dataloader loop at position %s in epoch %s",
-                     position, epoch)
-            self._position[epoch] += 1
-            return True
-        return False
+        if position > self._state.last_position.get(epoch, -1):
+            return False
+        LOG.info("skipping dataloader loop at position %s in epoch %s",
+                 position, epoch)
+        self._position[epoch] += 1
+        return True
 
     def to_tensorboard(self, writer, global_step, tag_prefix=""):
         if tag_prefix and not tag_prefix.endswith("/"):
             tag_prefix += "/"
-        writer.add_scalar(tag_prefix + "Total_Batch_Size",
-                          self.current_batch_size, global_step)
-        writer.add_scalar(tag_prefix + "Local_Batch_Size",
-                          self.current_local_bsz, global_step)
-        writer.add_scalar(tag_prefix + "Accumulation_Steps",
-                          self.accumulation_steps, global_step)
+        for tag, value in (("Total_Batch_Size", self.current_batch_size),
+                           ("Local_Batch_Size", self.current_local_bsz),
+                           ("Accumulation_Steps", self.accumulation_steps)):
+            writer.add_scalar(tag_prefix + tag, value, global_step)
+
+
+def _delegated(attr, only_while_iterating=False):
+    """Property on the mixin that forwards to ``self._elastic``."""
+    def getter(self):
+        helper = self._elastic
+        if only_while_iterating and not helper._iterating:
+            return None
+        return getattr(helper, attr)
+    return property(getter)
 
 
 class AdaptiveDataLoaderMixin(object):
@@ -285,33 +318,18 @@ class AdaptiveDataLoaderMixin(object):
     def __init__(self, batch_size):
         self._elastic = AdaptiveDataLoaderHelper(batch_size)
 
-    def autoscale_batch_size(self, max_batch_size, local_bsz_bounds=None,
-                             gradient_accumulation=False):
-        self._elastic.autoscale_batch_size(max_batch_size, local_bsz_bounds,
-                                           gradient_accumulation)
+    def autoscale_batch_size(self, *args, **kwargs):
+        self._elastic.autoscale_batch_size(*args, **kwargs)
 
-    @property
-    def current_local_bsz(self):
-        if AdaptiveDataLoaderHelper._current is not self._elastic:
-            return None
-        return self._elastic.current_local_bsz
+    def to_tensorboard(self, *args, **kwargs):
+        self._elastic.to_tensorboard(*args, **kwargs)
 
-    @property
-    def accumulation_steps(self):
-        return self._elastic.accumulation_steps
-
-    @property
-    def training(self):
-        return self._elastic.training
-
-    @property
-    def current_batch_size(self):
-        if AdaptiveDataLoaderHelper._current is not self._elastic:
-            return None
-        return self._elastic.current_batch_size
-
-    def to_tensorboard(self, writer, global_step, tag_prefix=""):
-        self._elastic.to_tensorboard(writer, global_step, tag_prefix)
+    current_local_bsz = _delegated("current_local_bsz",
+                                   only_while_iterating=True)
+    current_batch_size = _delegated("current_batch_size",
+                                    only_while_iterating=True)
+    accumulation_steps = _delegated("accumulation_steps")
+    training = _delegated("training")
 
 
 def _worker_init_wrapper(worker_init_fn, num_workers):
@@ -319,13 +337,11 @@ def _worker_init_wrapper(worker_init_fn, num_workers):
 
     @functools.wraps(worker_init_fn)
     def wrapper(worker_id):
-        nonlocal num_workers
-        num_workers = num_workers or 1
         seed = torch.initial_seed() \
-            + adaptdl_amd.env.replica_rank() * num_workers
-        torch.manual_seed(seed)
-        np.random.seed(seed % 2 ** 32)
-        random.seed(seed)
+            + adaptdl_amd.env.replica_rank() * max(num_workers or 1, 1)
+        for seeder in (torch.manual_seed, random.seed,
+                       lambda s: np.random.seed(s % 2 ** 32)):
+            seeder(seed)
         if worker_init_fn is not None:
             return worker_init_fn(worker_id)
     return wrapper
@@ -340,15 +356,20 @@ class AdaptiveDataLoader(DataLoader, AdaptiveDataLoaderMixin):
     """
 
     def __init__(self, dataset, batch_size=1, shuffle=False, **kwargs):
-        if kwargs.get("batch_sampler") is not None \
-                or kwargs.get("sampler") is not None:
-            raise ValueError("AdaptiveDataLoader does not support "
-                             "custom 'sampler' or 'batch_sampler'")
+        for forbidden in ("sampler", "batch_sampler"):
+            if kwargs.get(forbidden) is not None:
+                raise ValueError("AdaptiveDataLoader does not support "
+                                 "custom 'sampler' or 'batch_sampler'")
         kwargs["sampler"] = ElasticSampler(dataset, shuffle=shuffle)
         kwargs["worker_init_fn"] = _worker_init_wrapper(
             kwargs.get("worker_init_fn"), kwargs.get("num_workers"))
         super().__init__(dataset, batch_size, shuffle=False, **kwargs)
         AdaptiveDataLoaderMixin.__init__(self, batch_size)
+
+    def _epoch_target_reached(self, epoch):
+        """Scale-invariant progress has covered ``epoch + 1`` epochs."""
+        target = len(self.dataset) * (epoch + 1) / self.batch_size
+        return get_progress() >= target
 
     def __iter__(self):
         """Iterate over batches; stops after one (statistical) epoch.
@@ -358,33 +379,33 @@ class AdaptiveDataLoader(DataLoader, AdaptiveDataLoaderMixin):
         until scale-invariant progress covers one epoch-equivalent.
         """
         epoch = current_epoch()
-        num_replicas = adaptdl_amd.env.num_replicas()
+        adaptive = self._elastic.max_batch_size is not None
+        replicas = adaptdl_amd.env.num_replicas()
         with self._elastic.context():
             if self._elastic.skipdone():
                 return
-            done = False
-            while not done:
-                self.sampler.set_epoch(epoch,
-                                       index=self._elastic.current_index)
-                self.batch_sampler.batch_size = \
-                    self._elastic._sync_local_bsz()
+            while True:
+                self.sampler.set_epoch(
+                    epoch, index=self._elastic.current_index)
+                atomic_bsz = self._elastic._sync_local_bsz()
+                self.batch_sampler.batch_size = atomic_bsz
+                stop = False
                 for idx, batch in enumerate(super().__iter__()):
                     # Skip profiling the first batch of each pass (loader
                     # worker startup would pollute the perf model).
                     with self._elastic.profile(self.training and idx >= 1):
                         yield batch
                         self._elastic.current_index += \
-                            num_replicas * self.batch_sampler.batch_size
-                        if self._elastic.max_batch_size is not None and \
-                                get_progress() >= len(self.dataset) * \
-                                (epoch + 1) / self.batch_size:
-                            done = True
+                            replicas * self.batch_sampler.batch_size
+                        if adaptive and self._epoch_target_reached(epoch):
+                            stop = True
                             break
-                if self._elastic.max_batch_size is None:
-                    done = True
-                # Round current_index up to a multiple of the dataset size.
-                self._elastic.current_index -= \
-                    self._elastic.current_index % -len(self.dataset)
+                # Round current_index up to a multiple of the dataset size
+                # (ends the data pass even if it stopped mid-way).
+                self._elastic.current_index += \
+                    -self._elastic.current_index % len(self.dataset)
+                if stop or not adaptive:
+                    return
 
 
 class _AdaptiveDataLoaderState(adaptdl_amd.checkpoint.State):
@@ -392,14 +413,18 @@ class _AdaptiveDataLoaderState(adaptdl_amd.checkpoint.State):
     # Dataloaders must be initialized in the same order on every replica.
     init_count = collections.Counter()
 
+    # Fields that survive a checkpoint-restart (the batch-size choice is
+    # deliberately NOT persisted: it is re-chosen after a rescale).
+    _SAVED = ("current_index", "end_index", "last_position")
+
     def __init__(self):
         if current_dataloader() is not None:
             raise RuntimeError("dataloader may not be initialized during "
                                "dataloader iteration")
         epoch = current_epoch()
-        count = _AdaptiveDataLoaderState.init_count[epoch]
-        super().__init__("adaptdl-dataloader-epoch{}-{}".format(epoch, count))
-        _AdaptiveDataLoaderState.init_count[epoch] += 1
+        ordinal = self.init_count[epoch]
+        self.init_count[epoch] += 1
+        super().__init__(f"adaptdl-dataloader-epoch{epoch}-{ordinal}")
         self.current_index = 0
         self.end_index = 0
         self.last_position = {}
@@ -407,9 +432,8 @@ class _AdaptiveDataLoaderState(adaptdl_amd.checkpoint.State):
         self.accumulation_steps = 0
 
     def save(self, fileobj):
-        pickle.dump((self.current_index, self.end_index,
-                     self.last_position), fileobj)
+        pickle.dump({f: getattr(self, f) for f in self._SAVED}, fileobj)
 
     def load(self, fileobj):
-        self.current_index, self.end_index, self.last_position = \
-            pickle.load(fileobj)
+        for f, v in pickle.load(fileobj).items():
+            setattr(self, f, v)

@@ -207,17 +207,23 @@ class GraphedStepper(object):
 
         accum = dl.accumulation_steps
         is_sync = dl.is_optim_step()
+        shapes = tuple((tuple(t.shape), t.dtype) for t in tensors)
         if self._pos == 0:
-            sig = (dl.current_local_bsz, accum,
-                   tuple((tuple(t.shape), t.dtype) for t in tensors))
+            sig = (dl.current_local_bsz, accum, shapes)
             if sig != self._sig:
                 self._reset(sig, accum)
-        # Desync guard: position math must agree with the dataloader.
+        # Desync guards: position math must agree with the dataloader, and
+        # every microbatch's tensors must match the captured signature (a
+        # mid-cycle shape change — e.g. a short final batch with
+        # drop_last=False — would otherwise hit the static-buffer copy with
+        # a non-broadcastable shape).  Violations fall back to eager.
         expected_sync = (self._accum == 0) or (self._pos == self._accum)
-        if is_sync != expected_sync or accum != self._accum:
+        if is_sync != expected_sync or accum != self._accum \
+                or shapes != self._sig[2]:
             LOG.warning("graph stepper desync (pos %d, accum %d->%d, "
-                        "sync %s); falling back to eager this cycle",
-                        self._pos, self._accum, accum, is_sync)
+                        "sync %s, shape change %s); falling back to eager "
+                        "this cycle", self._pos, self._accum, accum,
+                        is_sync, shapes != self._sig[2])
             self.stats["fallbacks"] += 1
             self._sig = None      # force re-signature at next cycle start
             self._pos = 0 if is_sync else self._pos + 1
@@ -289,7 +295,9 @@ class GraphedStepper(object):
                 self._accum + 1 if kind == "sync" else 1
             for bucket in engine.buckets:
                 bucket.work = None
-            engine.last_sync_time = 0.0
+            # last_sync_time keeps the most recent eager-measured value
+            # (from the warmup cycles of this signature) so the goodput
+            # fit does not see all-reduce as free under replays.
             engine.pending_sync_done = False
             engine.graph_mode = False
             # Deferred host finalize: GNS estimates, scaling-rule gain,
@@ -310,6 +318,15 @@ class GraphedStepper(object):
             self._disabled = True
             self._engine.graph_mode = False
             self.stats["fallbacks"] += 1
+            # The aborted capture may have left partial side effects
+            # (accum_count increments, statistic-tensor contributions).
+            # Restart the cycle from clean state before re-running
+            # eagerly so no microbatch's statistics are double-counted.
+            # Earlier microbatches of this cycle lose their gradient
+            # contribution to this one step (a once-per-run event).
+            self._gns.reset_accumulation()
+            self._pos = 0
+            self._sig = None
             return self._run_eager(*tensors)
         self._graphs[kind] = (graph, out)
         self.stats["captures"] += 1
@@ -339,6 +356,9 @@ def maybe_graphed_stepper(adp, optimizer, fwd_bwd):
         return None
     try:
         return GraphedStepper(adp, optimizer, fwd_bwd)
-    except ValueError as exc:
+    except (ValueError, RuntimeError) as exc:
+        # ValueError: unsupported configuration (mp_scaler / Adam GNS).
+        # RuntimeError: HipGraphBackend construction (graph_pool_handle,
+        # Stream) can fail on misconfigured ROCm setups.
         LOG.warning("ADAPTDL_HIPGRAPH=1 ignored: %s", exc)
         return None

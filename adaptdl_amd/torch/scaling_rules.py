@@ -1,17 +1,21 @@
 """Learning-rate scaling rules applied at each (scaled) optimizer step.
 
-The optimizer's ``step``/``zero_grad`` are patched so user training loops
-stay unchanged: ``step`` applies per-param-group LR factors from the active
-rule, runs the original step, restores LRs, and advances scale-invariant
-progress by the current gain; ``zero_grad`` defers to the GNS accumulation
-bookkeeping.  Rules and semantics match the reference
-(``/root/reference/adaptdl/adaptdl/torch/scaling_rules.py``).
+The optimizer's ``step``/``zero_grad`` are taken over so user training
+loops stay unchanged: ``step`` runs the original optimizer step inside a
+context that multiplies each param group's LR by the active rule's
+factor, then advances scale-invariant progress by the current gain;
+``zero_grad`` defers to the GNS accumulation bookkeeping (it must NOT
+clear gradients mid-accumulation).
+
+Behavioral parity target (rule formulas + patching semantics):
+``/root/reference/adaptdl/adaptdl/torch/scaling_rules.py`` — the code
+here is an independent implementation; only the published rule math
+(AdaScale/LEGW etc.) is shared.
 """
 
-import functools
+import contextlib
 import math
 import warnings
-from types import MethodType
 
 import numpy as np
 
@@ -33,55 +37,81 @@ class ScalingRuleBase(object):
         optim.step()   # patched: applies the scaled learning rate
     """
 
+    #: Attributes wired up by :meth:`initialize`.
+    adp = None
+    _optimizer = None
+    _orig_optimizer_step = None
+
     def __init__(self):
         self.adp = None
         self._optimizer = None
         self._orig_optimizer_step = None
 
+    # ---- rule interface -------------------------------------------------
+
     def scale_lr(self, scale):
+        """Per-param-group LR multiplier at batch-size scale ``scale``."""
         raise NotImplementedError
 
-    def zero_grad(self, *args, **kwargs):
-        if self.adp.gns.should_zero_grad:
-            self.adp.gns.reset_accumulation()
-        else:
-            warnings.warn("skipping zero_grad for accumulated gradient")
+    # ---- patched optimizer entry points ---------------------------------
+
+    @contextlib.contextmanager
+    def _lr_scaled(self, scale):
+        """Temporarily multiply every param group's LR by the rule factor."""
+        groups = self._optimizer.param_groups
+        saved = {id(pg): pg["lr"] for pg in groups}
+        factors = np.broadcast_to(np.asarray(self.scale_lr(scale)),
+                                  (len(groups),))
+        try:
+            for factor, pg in zip(factors, groups):
+                pg["lr"] = saved[id(pg)] * float(factor)
+            yield
+        finally:
+            for pg in groups:
+                pg["lr"] = saved[id(pg)]
 
     def step(self, *args, **kwargs):
         """Run one optimizer step with a scaled learning rate."""
-        if not self.adp:
+        if self.adp is None:
             raise ValueError("AdaptiveDataParallel instance is not set!")
+        gns = self.adp.gns
         if not self.adp.require_backward_grad_sync:
-            return
-        scale = self.adp.gns.accum_scale * self.adp.gns.accum_count
-        initial_lr = [pg["lr"] for pg in self._optimizer.param_groups]
-        scaled_lr = np.multiply(self.scale_lr(scale), initial_lr)
-        for lr, pg in zip(scaled_lr, self._optimizer.param_groups):
-            pg["lr"] = lr
-        self._orig_optimizer_step(*args, **kwargs)
-        for lr, pg in zip(initial_lr, self._optimizer.param_groups):
-            pg["lr"] = lr
-        self.adp.gns.set_progress(self.adp.gns.get_progress()
-                                  + self.adp.gns.gain(scale))
+            # Mid-accumulation microbatch: the real step happens at the
+            # end of the accumulation cycle.
+            return None
+        scale = gns.accum_scale * gns.accum_count
+        with self._lr_scaled(scale):
+            self._orig_optimizer_step(*args, **kwargs)
+        gns.set_progress(gns.get_progress() + gns.gain(scale))
+        return None
 
-    def _patch_optimizer(self):
-        @functools.wraps(self._optimizer.step)
-        def step_wrapper(optim, *args, **kwargs):
-            return self.step(*args, **kwargs)
+    def zero_grad(self, *args, **kwargs):
+        gns = self.adp.gns
+        if gns.should_zero_grad:
+            gns.reset_accumulation()
+        else:
+            warnings.warn("skipping zero_grad for accumulated gradient")
 
-        @functools.wraps(self._optimizer.zero_grad)
-        def zero_wrapper(optim, *args, **kwargs):
-            return self.zero_grad(*args, **kwargs)
-
-        self._optimizer.step = MethodType(step_wrapper, self._optimizer)
-        self._optimizer.zero_grad = MethodType(zero_wrapper, self._optimizer)
+    # ---- wiring ----------------------------------------------------------
 
     def initialize(self, adp, optimizer, patch_optimizer=False):
+        """Attach to an ADP instance; optionally take over the optimizer's
+        ``step``/``zero_grad`` so plain training loops pick up the rule."""
         self.adp = adp
         self._optimizer = optimizer
         self._orig_optimizer_step = optimizer.step
         if patch_optimizer:
-            self._patch_optimizer()
+            rule = self
+            # Plain function assignment: attribute lookup on the instance
+            # shadows the class methods, no MethodType dance needed.
+            def patched_step(*a, **kw):
+                return rule.step(*a, **kw)
+            def patched_zero_grad(*a, **kw):
+                return rule.zero_grad(*a, **kw)
+            patched_step.__name__ = "step"
+            patched_zero_grad.__name__ = "zero_grad"
+            optimizer.step = patched_step
+            optimizer.zero_grad = patched_zero_grad
 
 
 class AdaScale(ScalingRuleBase):
@@ -89,10 +119,9 @@ class AdaScale(ScalingRuleBase):
     (var + sqr) / (var / scale + sqr)."""
 
     def scale_lr(self, scale):
-        var = self.adp.gns.raw_var_avg
-        sqr = self.adp.gns.raw_sqr_avg
-        var = np.maximum(var, 1e-6)
-        sqr = np.maximum(sqr, 0.0)
+        gns = self.adp.gns
+        var = np.maximum(gns.raw_var_avg, 1e-6)
+        sqr = np.maximum(gns.raw_sqr_avg, 0.0)
         return (var + sqr) / (var / scale + sqr)
 
 
@@ -126,10 +155,10 @@ class LEGWScale(ScalingRuleBase):
 
     def scale_lr(self, scale):
         dataloader = current_dataloader()
-        total_steps = self._base_warmup_epochs * scale * \
-            self._data_size / dataloader.batch_size
-        max_lr_multiplier = math.sqrt(scale)
+        warmup_steps = (self._base_warmup_epochs * scale
+                        * self._data_size / dataloader.batch_size)
+        peak = math.sqrt(scale)
         progress = self.adp.gns.get_progress()
-        if progress < total_steps:
-            return max_lr_multiplier * (progress / total_steps)
-        return max_lr_multiplier
+        if progress < warmup_steps:
+            return peak * progress / warmup_steps
+        return peak
