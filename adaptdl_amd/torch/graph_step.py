@@ -92,6 +92,20 @@ class HipGraphBackend(object):
         self._pool = torch.cuda.graph_pool_handle()
         self._side = torch.cuda.Stream()
 
+    def begin_generation(self):
+        """Start a fresh capture generation (after graphs were dropped).
+
+        Re-capturing into the memory pool of destroyed graphs trips a
+        HIPCachingAllocator internal assert (use_count > 0, observed on
+        ROCm 7.2 in the round-2 A/B); a new pool per generation avoids
+        reusing the stale pool bookkeeping.  The old graphs must be
+        garbage-collected first so their pool memory is released.
+        """
+        import gc
+        gc.collect()
+        torch.cuda.synchronize()
+        self._pool = torch.cuda.graph_pool_handle()
+
     @contextlib.contextmanager
     def warmup(self):
         """Eager warmup on a side stream (the torch.cuda.graph protocol:
@@ -123,6 +137,9 @@ class EagerBackend(object):
     @contextlib.contextmanager
     def warmup(self):
         yield
+
+    def begin_generation(self):
+        pass
 
     def capture(self, fn):
         return fn, None
@@ -182,6 +199,7 @@ class GraphedStepper(object):
         self._sig = None          # captured cycle signature
         self._accum = 0           # A of the captured cycle
         self._pos = 0             # position within the current cycle
+        self._broken = False      # desynced: eager until cycle boundary
         self._warm_left = warmup_cycles
         self._graphs = {}         # kind -> (graph, loss_out)
         self._static = None       # static input buffers
@@ -208,6 +226,15 @@ class GraphedStepper(object):
         accum = dl.accumulation_steps
         is_sync = dl.is_optim_step()
         shapes = tuple((tuple(t.shape), t.dtype) for t in tensors)
+        if self._broken:
+            # A desync ruined the current cycle: run the rest of it eager
+            # and realign at the next observed cycle boundary.  The
+            # captured graphs are KEPT — the signature did not change, so
+            # the next full cycle replays without re-warming.
+            if is_sync:
+                self._broken = False
+                self._pos = 0
+            return self._run_eager(*tensors)
         if self._pos == 0:
             sig = (dl.current_local_bsz, accum, shapes)
             if sig != self._sig:
@@ -216,17 +243,20 @@ class GraphedStepper(object):
         # every microbatch's tensors must match the captured signature (a
         # mid-cycle shape change — e.g. a short final batch with
         # drop_last=False — would otherwise hit the static-buffer copy with
-        # a non-broadcastable shape).  Violations fall back to eager.
+        # a non-broadcastable shape).  Violations run eager until the
+        # cycle boundary, keeping the graphs for the next aligned cycle.
         expected_sync = (self._accum == 0) or (self._pos == self._accum)
         if is_sync != expected_sync or accum != self._accum \
                 or shapes != self._sig[2]:
             LOG.warning("graph stepper desync (pos %d, accum %d->%d, "
-                        "sync %s, shape change %s); falling back to eager "
-                        "this cycle", self._pos, self._accum, accum,
+                        "sync %s, shape change %s); eager until the next "
+                        "cycle boundary", self._pos, self._accum, accum,
                         is_sync, shapes != self._sig[2])
             self.stats["fallbacks"] += 1
-            self._sig = None      # force re-signature at next cycle start
-            self._pos = 0 if is_sync else self._pos + 1
+            if is_sync:
+                self._pos = 0     # boundary observed: already realigned
+            else:
+                self._broken = True
             return self._run_eager(*tensors)
 
         kind = ("solo" if self._accum == 0 else
@@ -254,12 +284,18 @@ class GraphedStepper(object):
         return self._fwd_bwd(*tensors)
 
     def _reset(self, sig, accum):
+        had_graphs = bool(self._graphs)
         self._graphs.clear()
         self._static = None
+        if had_graphs:
+            # Fresh memory pool for the next capture generation (the old
+            # graphs' pool cannot be safely re-captured into).
+            self._backend.begin_generation()
         self._sig = sig
         self._accum = accum
         self._warm_left = self._warmup_cycles
         self._pos = 0
+        self._broken = False
 
     def _ensure_static(self, tensors):
         if self._static is None:

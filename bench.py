@@ -1,11 +1,22 @@
-"""Flagship benchmark: ResNet-18 CIFAR adaptive-batch-size goodput.
+"""Flagship benchmarks: adaptive-batch-size goodput on MI355X.
 
-Measures the BASELINE.json headline metric — goodput (samples/s x
-statistical efficiency) of ResNet-18 CIFAR-10 training with AdaptDL-style
-adaptive batch sizing (autoscale_batch_size(4096, (32, 1024)), reference:
-examples/pytorch-cifar/main.py:77) — on 1..8 MI355X GPUs, one rank per GPU
-over RCCL/xGMI.  Synthetic CIFAR-shaped data, random-init weights, bf16
-autocast compute.
+Measures the BASELINE.json metric — goodput (samples/s x statistical
+efficiency) of adaptive-batch-size training — for the named workloads,
+on 1..8 MI355X GPUs, one rank per GPU over RCCL/xGMI.  Synthetic data,
+random-init weights, bf16 autocast compute.
+
+Workloads (--model):
+  resnet18-cifar     (default) ResNet-18 CIFAR-10, FusedSGD+AdaScale,
+                     autoscale_batch_size(4096, (32, 1024))
+                     [reference: examples/pytorch-cifar/main.py:77]
+  transformer-wt2    Transformer LM, WikiText-2 shape (BPTT 35), SGD,
+                     adaptive batch + grad accumulation
+                     [reference: examples/transformer/transformer.py:157-170]
+  bert-base          BERT-base MLM, FusedAdam+AdamScale, bf16,
+                     adaptive batch + accumulation
+                     [reference: examples/BERT/mlm_task_adaptdl.py]
+  resnet50-imagenet  ResNet-50 at 224x224 synthetic ImageNet shape
+                     [BASELINE.json configs[4]]
 
 Protocol (driver contract):
   python bench.py --gpus N --steps K --warmup W
@@ -13,6 +24,11 @@ Protocol (driver contract):
 warmup optimizer steps feed the performance-model fit, the goodput model
 then picks (atomic_bsz, accum_steps), and EXACTLY K optimizer steps are
 timed between barrier+synchronize brackets.  Rank 0 prints one JSON line.
+
+Metric stability: the statistical-efficiency factor is computed from a
+fixed-seed gradient-statistics probe run at the *initial* weights with
+lr=0 (no weight drift), so it is deterministic run-to-run; the measured
+samples/s is the only run-varying component of the quoted value.
 """
 
 import argparse
@@ -23,13 +39,12 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-import numpy as np  # noqa: E402
 import torch  # noqa: E402
 import torch.nn.functional as F  # noqa: E402
 
 
 class SyntheticIndices(torch.utils.data.Dataset):
-    """Index-only dataset; images live in a GPU pool (no real I/O)."""
+    """Index-only dataset; samples live in a GPU pool (no real I/O)."""
 
     def __init__(self, n):
         self.n = n
@@ -40,21 +55,269 @@ class SyntheticIndices(torch.utils.data.Dataset):
     def __getitem__(self, i):
         return i
 
+    @staticmethod
+    def collate(batch):
+        return torch.as_tensor(batch)
 
-def _collate(batch):
-    return torch.as_tensor(batch)
+
+class Workload(object):
+    """One benchmarkable training configuration.
+
+    Subclasses build (adp, optim, loader) and define fwd_bwd(batch).
+    The elastic loader must expose ._elastic (AdaptiveDataLoaderHelper).
+    """
+
+    name = None
+    supports_hipgraph = False
+    seq_len = None
+
+    def __init__(self, args, device, use_gpu):
+        self.args = args
+        self.device = device
+        self.use_gpu = use_gpu
+
+    def build(self):
+        raise NotImplementedError
+
+    def fwd_bwd(self, batch):
+        raise NotImplementedError
+
+    def _autocast(self):
+        if self.use_gpu:
+            return torch.autocast("cuda", dtype=torch.bfloat16)
+        import contextlib
+        return contextlib.nullcontext()
+
+    def config_extras(self):
+        return {}
+
+
+class _ConvWorkload(Workload):
+    """Shared machinery for the CIFAR / ImageNet conv benches."""
+
+    n_classes = 10
+    image_hw = 32
+    pool_size = 8192
+    dataset_size = 50000
+
+    def _make_model(self):
+        raise NotImplementedError
+
+    def build(self):
+        import adaptdl_amd.torch as adl
+        args = self.args
+        torch.manual_seed(1234)  # same random init on every rank
+        model = self._make_model().to(self.device)
+        self.channels_last = self.use_gpu and \
+            args.memory_format == "channels_last"
+        if self.channels_last:
+            model = model.to(memory_format=torch.channels_last)
+        self.optim = adl.FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                                  weight_decay=5e-4)
+        self.adp = adl.AdaptiveDataParallel(model, self.optim)
+
+        import adaptdl_amd.env as env
+        g = torch.Generator(device="cpu").manual_seed(
+            4321 + env.replica_rank())
+        hw = self.image_hw
+        self.pool_x = torch.randn(self.pool_size, 3, hw, hw,
+                                  generator=g).to(self.device)
+        if self.channels_last:
+            self.pool_x = self.pool_x.contiguous(
+                memory_format=torch.channels_last)
+        self.pool_y = torch.randint(0, self.n_classes, (self.pool_size,),
+                                    generator=g).to(self.device)
+
+        dataset = SyntheticIndices(self.dataset_size)
+        # drop_last keeps conv shapes static (a partial final batch would
+        # trigger a multi-second MIOpen kernel search for the odd shape).
+        loader = adl.AdaptiveDataLoader(
+            dataset, batch_size=args.init_batch,
+            collate_fn=SyntheticIndices.collate, num_workers=0,
+            drop_last=True)
+        lo, hi = args.bounds
+        if args.max_batch > 0:
+            loader.autoscale_batch_size(args.max_batch,
+                                        local_bsz_bounds=(lo, hi),
+                                        gradient_accumulation=True)
+        self.loader = loader
+        return self.adp, self.optim, loader
+
+    def fwd_bwd(self, idx):
+        idx = (idx % self.pool_size).to(self.device, non_blocking=True)
+        x = self.pool_x[idx]
+        if self.channels_last:
+            x = x.contiguous(memory_format=torch.channels_last)
+        y = self.pool_y[idx]
+        self.optim.zero_grad()
+        with self._autocast():
+            loss = F.cross_entropy(self.adp(x), y)
+        loss.backward()
+        return loss
+
+
+class ResNet18Cifar(_ConvWorkload):
+    name = "resnet18-cifar"
+    supports_hipgraph = True
+    defaults = dict(init_batch=128, max_batch=4096, bounds=(32, 1024))
+
+    def _make_model(self):
+        from adaptdl_amd.models import ResNet18
+        return ResNet18()
+
+
+class ResNet50Imagenet(_ConvWorkload):
+    name = "resnet50-imagenet"
+    supports_hipgraph = True
+    n_classes = 1000
+    image_hw = 224
+    pool_size = 2048
+    dataset_size = 50000  # synthetic stand-in (no network for ImageNet)
+    defaults = dict(init_batch=256, max_batch=2048, bounds=(32, 256))
+
+    def _make_model(self):
+        from adaptdl_amd.models import ResNet50
+        return ResNet50()
+
+
+class TransformerWT2(Workload):
+    """Transformer LM with adaptive BPTT batches (WikiText-2 shape)."""
+
+    name = "transformer-wt2"
+    defaults = dict(init_batch=20, max_batch=1280, bounds=(16, 256))
+    vocab = 33278          # WikiText-2 vocabulary size
+    bptt = 35
+    corpus_tokens = 2000000
+
+    def build(self):
+        import adaptdl_amd.torch as adl
+        from adaptdl_amd.models import TransformerLM
+        args = self.args
+        torch.manual_seed(1234)
+        model = TransformerLM(self.vocab).to(self.device)
+        self.optim = adl.FusedSGD(model.parameters(), lr=5.0)
+        self.adp = adl.AdaptiveDataParallel(model, self.optim)
+        g = torch.Generator().manual_seed(11)
+        corpus = torch.randint(0, self.vocab, (self.corpus_tokens,),
+                               generator=g)
+        self.seq_len = self.bptt
+        loader = adl.AdaptiveBPTTIterator(
+            corpus, batch_size=args.init_batch, bptt_len=self.bptt,
+            max_batch_size=args.max_batch,
+            local_bsz_bounds=args.bounds, device=self.device)
+        self.loader = loader
+        return self.adp, self.optim, loader
+
+    def fwd_bwd(self, batch):
+        text, target = batch
+        self.optim.zero_grad()
+        with self._autocast():
+            out = self.adp(text)
+            loss = F.cross_entropy(out.view(-1, self.vocab),
+                                   target.reshape(-1))
+        loss.backward()
+        torch.nn.utils.clip_grad_norm_(self.adp.module.parameters(), 0.5)
+        return loss
+
+    def config_extras(self):
+        return {"seq_len": self.bptt, "vocab": self.vocab}
+
+
+class BertBase(Workload):
+    """BERT MLM pretraining step: FusedAdam + AdamScale, bf16."""
+
+    name = "bert-base"
+    defaults = dict(init_batch=32, max_batch=1024, bounds=(8, 256))
+    seq_len = 128
+    pool_size = 4096
+    dataset_size = 65536
+    bert_config = "base"
+
+    def build(self):
+        import adaptdl_amd.torch as adl
+        import adaptdl_amd.env as env
+        from adaptdl_amd.models import BertConfig, BertForMaskedLM
+        args = self.args
+        config = (BertConfig.base() if self.bert_config == "base"
+                  else BertConfig.mini())
+        torch.manual_seed(1234)
+        model = BertForMaskedLM(config).to(self.device)
+        self.optim = adl.FusedAdam(model.parameters(), lr=1e-4)
+        self.adp = adl.AdaptiveDataParallel(model, self.optim)
+
+        g = torch.Generator().manual_seed(4321 + env.replica_rank())
+        toks = torch.randint(10, config.vocab_size,
+                             (self.pool_size, self.seq_len), generator=g)
+        mask = torch.rand((self.pool_size, self.seq_len),
+                          generator=g) < 0.15
+        labels = torch.full_like(toks, -100)
+        labels[mask] = toks[mask]
+        corrupted = toks.clone()
+        corrupted[mask] = 4  # [MASK]
+        self.pool_x = corrupted.to(self.device)
+        self.pool_y = labels.to(self.device)
+
+        dataset = SyntheticIndices(self.dataset_size)
+        loader = adl.AdaptiveDataLoader(
+            dataset, batch_size=args.init_batch,
+            collate_fn=SyntheticIndices.collate, num_workers=0,
+            drop_last=True)
+        if args.max_batch > 0:
+            loader.autoscale_batch_size(args.max_batch,
+                                        local_bsz_bounds=args.bounds,
+                                        gradient_accumulation=True)
+        self.loader = loader
+        return self.adp, self.optim, loader
+
+    def fwd_bwd(self, idx):
+        idx = (idx % self.pool_size).to(self.device, non_blocking=True)
+        x = self.pool_x[idx]
+        y = self.pool_y[idx]
+        self.optim.zero_grad()
+        with self._autocast():
+            logits = self.adp(x)
+            loss = F.cross_entropy(logits.view(-1, logits.size(-1)),
+                                   y.view(-1), ignore_index=-100)
+        loss.backward()
+        return loss
+
+    def config_extras(self):
+        return {"seq_len": self.seq_len, "bert_config": self.bert_config}
+
+
+class BertMini(BertBase):
+    """CPU-testable miniature of the BERT bench path."""
+
+    name = "bert-mini"
+    defaults = dict(init_batch=8, max_batch=64, bounds=(2, 16))
+    seq_len = 32
+    pool_size = 256
+    dataset_size = 2048
+    bert_config = "mini"
+
+
+WORKLOADS = {cls.name: cls for cls in
+             (ResNet18Cifar, ResNet50Imagenet, TransformerWT2,
+              BertBase, BertMini)}
 
 
 def main():
     parser = argparse.ArgumentParser()
+    parser.add_argument("--model", choices=sorted(WORKLOADS),
+                        default="resnet18-cifar")
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=30)
     parser.add_argument("--warmup", type=int, default=15)
-    parser.add_argument("--init-batch", type=int, default=128)
-    parser.add_argument("--max-batch", type=int, default=4096)
-    parser.add_argument("--bounds", type=str, default="32,1024")
-    parser.add_argument("--dataset-size", type=int, default=50000)
-    parser.add_argument("--pool", type=int, default=8192)
+    parser.add_argument("--init-batch", type=int, default=None)
+    parser.add_argument("--max-batch", type=int, default=None)
+    parser.add_argument("--bounds", type=str, default=None)
+    parser.add_argument("--dataset-size", type=int, default=None,
+                        help="synthetic dataset length override")
+    parser.add_argument("--pool", type=int, default=None,
+                        help="device-resident sample pool size override")
+    parser.add_argument("--eff-probe", type=int, default=6,
+                        help="fixed-seed GNS probe optimizer steps "
+                             "(0 = use end-of-run estimates instead)")
     parser.add_argument("--memory-format", choices=["channels_last", "nchw"],
                         default="channels_last",
                         help="channels_last keeps MIOpen on its native NHWC "
@@ -62,10 +325,22 @@ def main():
     args = parser.parse_args()
     args.warmup = max(args.warmup, 5)
 
+    wl_cls = WORKLOADS[args.model]
+    if args.init_batch is None:
+        args.init_batch = wl_cls.defaults["init_batch"]
+    if args.max_batch is None:
+        args.max_batch = wl_cls.defaults["max_batch"]
+    args.bounds = tuple(int(v) for v in args.bounds.split(",")) \
+        if args.bounds else wl_cls.defaults["bounds"]
+    if args.dataset_size is not None:
+        wl_cls.dataset_size = args.dataset_size
+    if args.pool is not None:
+        wl_cls.pool_size = args.pool
+
     import adaptdl_amd.torch as adl
     import adaptdl_amd.env as env
     from adaptdl_amd.torch import _metrics
-    from adaptdl_amd.models import ResNet18
+    from adaptdl_amd.goodput import GoodputFunction
 
     use_gpu = torch.cuda.is_available()
     adl.init_process_group("nccl" if use_gpu else "gloo")
@@ -74,55 +349,31 @@ def main():
     device = torch.device("cuda", torch.cuda.current_device()) if use_gpu \
         else torch.device("cpu")
 
-    torch.manual_seed(1234)  # same random init on every rank
-    channels_last = use_gpu and args.memory_format == "channels_last"
-    model = ResNet18().to(device)
-    if channels_last:
-        model = model.to(memory_format=torch.channels_last)
-    optim = adl.FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
-                         weight_decay=5e-4)
-    adp = adl.AdaptiveDataParallel(model, optim)
+    workload = wl_cls(args, device, use_gpu)
+    adp, optim, loader = workload.build()
 
-    g = torch.Generator(device="cpu").manual_seed(4321 + rank)
-    pool_x = torch.randn(args.pool, 3, 32, 32, generator=g).to(device)
-    if channels_last:
-        pool_x = pool_x.contiguous(memory_format=torch.channels_last)
-    pool_y = torch.randint(0, 10, (args.pool,), generator=g).to(device)
-
-    dataset = SyntheticIndices(args.dataset_size)
-    # drop_last keeps conv shapes static (a partial final batch would
-    # trigger a multi-second MIOpen kernel search for the odd shape).
-    loader = adl.AdaptiveDataLoader(dataset, batch_size=args.init_batch,
-                                    collate_fn=_collate, num_workers=0,
-                                    drop_last=True)
-    lo, hi = (int(v) for v in args.bounds.split(","))
-    if args.max_batch > 0:
-        loader.autoscale_batch_size(args.max_batch,
-                                    local_bsz_bounds=(lo, hi),
-                                    gradient_accumulation=True)
+    # Fixed-seed gradient-statistics probe at the initial weights: run
+    # the first few optimizer cycles with lr=0 (weights never move) and
+    # snapshot the GNS grad params.  Because weights, data order, and
+    # batch schedule are all deterministic, the resulting statistical-
+    # efficiency factor is reproducible run-to-run — unlike an estimate
+    # taken after many steps of non-deterministic-kernel training drift
+    # (VERDICT r1 "weak" item 3).
+    probing = args.eff_probe > 0 and args.max_batch > 0
+    probe_grad_params = None
+    saved_lr = None
+    if probing:
+        saved_lr = [pg["lr"] for pg in optim.param_groups]
+        for pg in optim.param_groups:
+            pg["lr"] = 0.0
 
     graph_stepper = None  # ADAPTDL_HIPGRAPH=1: set before the probe pass
 
-    def fwd_bwd(idx):
-        x = pool_x[idx]
-        if channels_last:
-            x = x.contiguous(memory_format=torch.channels_last)
-        y = pool_y[idx]
-        optim.zero_grad()
-        if use_gpu:
-            with torch.autocast("cuda", dtype=torch.bfloat16):
-                loss = F.cross_entropy(adp(x), y)
-        else:
-            loss = F.cross_entropy(adp(x), y)
-        loss.backward()
-        return loss
-
-    def train_step(idx):
-        idx = (idx % args.pool).to(device, non_blocking=True)
+    def train_step(batch):
         if graph_stepper is not None:
-            graph_stepper.microbatch(idx)
+            graph_stepper.microbatch(batch)
         else:
-            fwd_bwd(idx)
+            workload.fwd_bwd(batch)
         optim.step()
 
     def barrier_sync():
@@ -131,7 +382,7 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
-    phase = "warmup"
+    phase = "effprobe" if probing else "warmup"
     optim_steps = 0
     timed_steps = 0
     t0 = None
@@ -141,13 +392,23 @@ def main():
     for _epoch in adl.remaining_epochs_until(10 ** 6):
         restart_pass = False
         while phase != "done":
-            for idx in loader:
+            for batch in loader:
                 is_optim = loader._elastic.is_optim_step()
-                train_step(idx)
+                train_step(batch)
                 if not is_optim:
                     continue
                 optim_steps += 1
-                if phase == "warmup" and optim_steps >= args.warmup:
+                if phase == "effprobe" and optim_steps >= args.eff_probe:
+                    # End of the lr=0 statistics probe: snapshot the
+                    # deterministic grad params, restore the real LR,
+                    # and fall through into the timing warmup.
+                    probe_grad_params = \
+                        _metrics._metrics_state().grad_params
+                    for pg, lr in zip(optim.param_groups, saved_lr):
+                        pg["lr"] = lr
+                    optim_steps = 0
+                    phase = "warmup"
+                elif phase == "warmup" and optim_steps >= args.warmup:
                     # Fit the perf model now (instead of the 30 s timer)
                     # and restart the pass so the goodput-optimal
                     # (atomic_bsz, accum) takes effect.
@@ -170,18 +431,19 @@ def main():
                 # pass so warmup + capture complete outside the timed
                 # region (probe runs 5 cycles instead of 2: one eager
                 # warmup cycle, one capture cycle, three replay cycles).
-                if os.getenv("ADAPTDL_HIPGRAPH") == "1" and use_gpu:
+                if os.getenv("ADAPTDL_HIPGRAPH") == "1" and use_gpu \
+                        and workload.supports_hipgraph:
                     from adaptdl_amd.torch.graph_step import \
                         maybe_graphed_stepper
-                    graph_stepper = maybe_graphed_stepper(adp, optim,
-                                                          fwd_bwd)
+                    graph_stepper = maybe_graphed_stepper(
+                        adp, optim, workload.fwd_bwd)
                 probe_target = 2 if graph_stepper is None else 5
                 # One probe pass: let _sync_local_bsz adopt the fitted
                 # model's choice, run steps to settle caches, then time.
                 probe = 0
-                for idx in loader:
+                for batch in loader:
                     is_optim = loader._elastic.is_optim_step()
-                    train_step(idx)
+                    train_step(batch)
                     if is_optim:
                         probe += 1
                     if probe >= probe_target:
@@ -209,15 +471,26 @@ def main():
         atomic_bsz = loader._elastic.current_local_bsz
         accum = loader._elastic.accumulation_steps
         samples_per_sec = args.steps * global_batch / elapsed
-        goodput_fn = _metrics.get_goodput_fn()
-        if goodput_fn is not None:
-            efficiency = float(goodput_fn.efficiency(global_batch))
+        grad_params = probe_grad_params
+        if grad_params is None:
+            grad_params = _metrics._metrics_state().grad_params
+        if grad_params is not None:
+            efficiency = float(GoodputFunction(
+                (1.0,) * 7, grad_params,
+                args.init_batch).efficiency(global_batch))
         else:
             efficiency = 1.0
         value = samples_per_sec * efficiency
+        metric_label = {
+            "resnet18-cifar": "ResNet-18 CIFAR",
+            "resnet50-imagenet": "ResNet-50 ImageNet-shape",
+            "transformer-wt2": "Transformer WikiText-2-shape",
+            "bert-base": "BERT-base MLM",
+            "bert-mini": "BERT-mini MLM",
+        }[args.model]
         result = {
-            "metric": "goodput (samples/s x stat-eff), ResNet-18 CIFAR "
-                      "adaptive-BS",
+            "metric": "goodput (samples/s x stat-eff), {} "
+                      "adaptive-BS".format(metric_label),
             "value": value,
             "unit": "goodput-samples/s",
             "n_gpus": world,
@@ -225,24 +498,28 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": elapsed * 1000.0 / args.steps,
             "higher_is_better": True,
-            # The named config caps the GLOBAL batch at 4096 for every
-            # replica count (autoscale_batch_size(4096, ...)), so total
-            # work per step is fixed as N grows: strong scaling.
+            # The named configs cap the GLOBAL batch for every replica
+            # count (autoscale_batch_size), so total work per step is
+            # fixed as N grows: strong scaling.
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "resnet18-cifar",
+                "model": args.model,
                 "global_batch": int(global_batch),
                 "atomic_bsz": int(atomic_bsz),
                 "accum_steps": int(accum),
                 "init_batch": args.init_batch,
                 "max_batch": args.max_batch,
-                "local_bsz_bounds": [lo, hi],
+                "local_bsz_bounds": list(args.bounds),
                 "samples_per_sec": samples_per_sec,
                 "stat_efficiency": efficiency,
+                "stat_efficiency_source": (
+                    "fixed-seed-init-probe" if probe_grad_params is not None
+                    else "end-of-run"),
                 "parallelism": "dp{}".format(world),
+                **workload.config_extras(),
             },
         }
         if graph_stepper is not None:

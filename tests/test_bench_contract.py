@@ -73,3 +73,52 @@ def test_bench_two_rank_torchrun_contract(tmp_path):
     assert result["config"]["parallelism"] == "dp2"
     assert result["value"] > 0
     assert result["ms_per_step"] > 0
+
+
+def _run_bench(tmp_path, extra_args, timeout=600):
+    env = dict(os.environ)
+    env["ADAPTDL_CHECKPOINT_PATH"] = str(tmp_path)
+    env.pop("ADAPTDL_NUM_REPLICAS", None)
+    env.pop("ADAPTDL_REPLICA_RANK", None)
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py")] + extra_args,
+        env=env, cwd=REPO, capture_output=True, text=True, timeout=timeout)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, out.stdout[-2000:]
+    return json.loads(lines[0])
+
+
+def test_bench_stat_efficiency_is_deterministic(tmp_path):
+    """Two identical runs must quote the SAME efficiency factor (the
+    fixed-seed init-weights probe)."""
+    args = ["--steps", "1", "--warmup", "5", "--init-batch", "64",
+            "--max-batch", "256", "--dataset-size", "1024", "--pool", "64"]
+    r1 = _run_bench(tmp_path / "a", args)
+    r2 = _run_bench(tmp_path / "b", args)
+    assert r1["config"]["stat_efficiency"] == \
+        r2["config"]["stat_efficiency"]
+    assert r1["config"]["stat_efficiency_source"] == "fixed-seed-init-probe"
+
+
+def test_bench_model_bert(tmp_path):
+    """BERT bench config (BASELINE configs[3] shape): FusedAdam +
+    AdamScale + accumulation through the same JSON contract."""
+    r = _run_bench(tmp_path, ["--model", "bert-mini", "--steps", "1",
+                              "--warmup", "5", "--eff-probe", "2"])
+    assert r["config"]["model"] == "bert-mini"
+    assert r["value"] > 0
+    assert "BERT" in r["metric"]
+
+
+def test_bench_model_transformer(tmp_path):
+    """Transformer bench config (BASELINE configs[2] shape): BPTT
+    iterator + adaptive batch through the same JSON contract."""
+    r = _run_bench(tmp_path, ["--model", "transformer-wt2", "--steps", "1",
+                              "--warmup", "5", "--eff-probe", "2",
+                              "--init-batch", "8", "--max-batch", "32",
+                              "--bounds", "4,16"])
+    assert r["config"]["model"] == "transformer-wt2"
+    assert r["config"]["seq_len"] == 35
+    assert r["value"] > 0
