@@ -38,11 +38,12 @@ done
 cd /tmp && export TMPDIR=/tmp && cd - >/dev/null
 for cfg in "" "ADAPTDL_HIPGRAPH=1"; do
     name=${cfg:-eager}; name=${name%%=*}
-    env $cfg timeout 300 rocprofv3 --kernel-trace \
+    env $cfg timeout 300 rocprofv3 --kernel-trace -f csv \
         -d "gpurun_out/trace_${name}" -- \
         python bench.py --steps 15 --warmup 10 \
         > "gpurun_out/trace_${name}.log" 2>&1
-    python tools/trace_gaps.py gpurun_out/trace_${name}/*/*kernel_trace*.csv \
+    python tools/trace_gaps.py \
+        $(find gpurun_out/trace_${name} -name '*kernel_trace*.csv') \
         > "gpurun_out/gaps_${name}.txt" 2>&1 || true
     tail -5 "gpurun_out/gaps_${name}.txt"
 done
