@@ -55,9 +55,19 @@ def _staging_dir(root):
     return d
 
 
-def save_all_states():
-    """Save every registered State; returns the checkpoint root on rank 0."""
-    root = env.checkpoint_path()
+def save_all_states(cold=False):
+    """Save every registered State; returns the checkpoint root on rank 0.
+
+    The rescale hot path writes to the RAM-backed warm root when the
+    controller provisioned one (``ADAPTDL_WARM_CHECKPOINT_PATH``,
+    normally tmpfs): model/optimizer/GNS state then round-trips through
+    host memory, never disk, on an elastic rescale.  ``cold=True`` (or
+    no warm root) writes to the on-disk ``ADAPTDL_CHECKPOINT_PATH`` in
+    the reference-compatible checkpoint-<K> format for crash recovery.
+    """
+    root = None if cold else env.warm_checkpoint_path()
+    if root is None:
+        root = env.checkpoint_path()
     for state in list(_REGISTRY.values()):
         save_state(state, root)
     if env.replica_rank() == 0 and root is not None:
@@ -83,8 +93,11 @@ def save_state(state, root, sync=True):
 
 
 def _latest_ckpt_dir(root):
+    """(dir, K) of the highest checkpoint-K under root, or (None, -1)."""
     best = None
     best_k = -1
+    if root is None or not os.path.isdir(root):
+        return best, best_k
     for name in os.listdir(root):
         if name.startswith(CKPT_DIR_PREFIX):
             try:
@@ -93,18 +106,22 @@ def _latest_ckpt_dir(root):
                 continue
             if k > best_k:
                 best_k, best = k, os.path.join(root, name)
-    return best
+    return best, best_k
 
 
 def load_state(state):
     """Load a State from the most recent checkpoint dir, if one exists.
 
+    Both roots are considered — the RAM-backed warm root (rescale hot
+    path) and the on-disk cold root — and the highest restart number
+    wins, the warm root breaking ties (it is at least as recent).
+
     Returns True iff the state file was found and ``state.load`` invoked.
     """
-    root = env.checkpoint_path()
-    if root is None or not os.path.isdir(root):
-        return False
-    ckpt_dir = _latest_ckpt_dir(root)
+    warm_dir, warm_k = _latest_ckpt_dir(env.warm_checkpoint_path())
+    cold_dir, cold_k = _latest_ckpt_dir(env.checkpoint_path())
+    ckpt_dir = warm_dir if warm_k >= cold_k and warm_dir is not None \
+        else cold_dir
     if ckpt_dir is None:
         return False
     path = os.path.join(ckpt_dir, state.name)

@@ -18,6 +18,14 @@ def checkpoint_path():
     return os.getenv("ADAPTDL_CHECKPOINT_PATH")
 
 
+def warm_checkpoint_path():
+    """RAM-backed checkpoint root for elastic rescales (tmpfs, e.g. a
+    /dev/shm directory provisioned by the controller).  When set, the
+    SIGTERM-rescale checkpoint round-trips through memory instead of
+    disk; ``checkpoint_path`` remains the cold (crash-recovery) path."""
+    return os.getenv("ADAPTDL_WARM_CHECKPOINT_PATH")
+
+
 def share_path():
     """Directory shared between all replicas (e.g. for dataset caches)."""
     return os.getenv("ADAPTDL_SHARE_PATH")

@@ -17,6 +17,7 @@ State machine per job: PENDING -> STARTING -> RUNNING -> STOPPING ->
 
 import logging
 import os
+import shutil
 import signal
 import socket
 import subprocess
@@ -128,6 +129,7 @@ class _Job(object):
         self.num_restarts = 0
         self.failures = 0
         self.completion = None    # set when SUCCEEDED/FAILED
+        self.warm_dir = None      # RAM-backed rescale checkpoint root
 
     @property
     def num_replicas(self):
@@ -223,7 +225,14 @@ class LocalController(object):
             jobs = list(self._jobs.values())
         for job in jobs:
             self._signal_group(job, signal.SIGKILL)
+            self._drop_warm_dir(job)
         self.supervisor.stop()
+
+    @staticmethod
+    def _drop_warm_dir(job):
+        if job.warm_dir is not None:
+            shutil.rmtree(job.warm_dir, ignore_errors=True)
+            job.warm_dir = None
 
     def log_paths(self, name):
         job = self._jobs[name]
@@ -324,11 +333,23 @@ class LocalController(object):
                     busy.extend(job.gpus)
         return busy
 
+    @staticmethod
+    def _make_warm_dir(name):
+        """RAM-backed (tmpfs) per-job dir for the in-memory rescale
+        checkpoint; /dev/shm on Linux, with a tempdir fallback."""
+        import tempfile
+        base = "/dev/shm" if os.access("/dev/shm", os.W_OK) \
+            else tempfile.gettempdir()
+        return tempfile.mkdtemp(prefix="adaptdl-warm-{}-".format(name),
+                                dir=base)
+
     def _start_group(self, job):
         spec = job.spec
         n = len(job.allocation)
         job.state = STARTING
         master_port = _free_port()
+        if job.warm_dir is None:
+            job.warm_dir = self._make_warm_dir(spec.name)
         gpus = []
         if spec.gpus_per_replica > 0:
             busy = self._assigned_gpus()
@@ -349,6 +370,7 @@ class LocalController(object):
             env.update({
                 "ADAPTDL_JOB_ID": spec.name,
                 "ADAPTDL_CHECKPOINT_PATH": spec.job_dir,
+                "ADAPTDL_WARM_CHECKPOINT_PATH": job.warm_dir,
                 "ADAPTDL_MASTER_ADDR": "127.0.0.1",
                 "ADAPTDL_MASTER_PORT": str(master_port),
                 "MASTER_ADDR": "127.0.0.1",
@@ -418,6 +440,7 @@ class LocalController(object):
         if all(c == 0 for c in codes):
             job.state = SUCCEEDED
             job.completion = time.time()
+            self._drop_warm_dir(job)
             _metric("succeeded")
             LOG.info("job %s succeeded", job.spec.name)
         elif all(c in (0, GRACEFUL_EXIT) for c in codes) or \
@@ -437,6 +460,7 @@ class LocalController(object):
             if job.failures > job.spec.restart_limit:
                 job.state = FAILED
                 job.completion = time.time()
+                self._drop_warm_dir(job)
                 _metric("failed")
                 LOG.warning("job %s failed (codes=%s)", job.spec.name,
                             codes)

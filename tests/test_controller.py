@@ -200,3 +200,49 @@ def test_jobspec_admission_validation(tmp_path):
     with pytest.raises(ValueError):
         JobSpec(["x"], name="bad", job_dir=str(tmp_path),
                 max_replicas=0)
+
+
+def test_warm_rescale_no_disk_checkpoint(tmp_path, controller):
+    """In-memory (tmpfs) rescale handoff: a 1 -> 2 -> 1 elastic drill
+    must resume correctly from the controller-provisioned warm root
+    with NO checkpoint-K directory ever written to the on-disk job_dir
+    (the cold path stays reference-format for crash recovery).
+    North star: 'elastic rescale via in-HBM checkpoint and rejoin'."""
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="warm-job",
+                   job_dir=job_dir, min_replicas=1, max_replicas=2,
+                   gpus_per_replica=0)
+    controller.submit(spec)
+
+    deadline = time.time() + 60
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("warm-job")
+        time.sleep(0.1)
+    controller.rescale("warm-job", 2)
+    # Wait for the first restart to land, then scale back down.
+    while controller.status("warm-job")["restarts"] < 1:
+        assert time.time() < deadline, controller.status("warm-job")
+        time.sleep(0.1)
+    while not [ln for ln in open(trace_path)
+               if json.loads(ln)["restarts"] >= 1]:
+        assert time.time() < deadline
+        time.sleep(0.1)
+    controller.rescale("warm-job", 1)
+
+    state = controller.wait("warm-job", timeout=180)
+    assert state == "Succeeded", controller.status("warm-job")
+
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(30))
+    assert {t["replicas"] for t in trace} == {1, 2}
+    assert max(t["restarts"] for t in trace) >= 2
+    # THE warm-path property: the elastic rescales round-tripped through
+    # RAM; nothing wrote the on-disk checkpoint-K format.
+    assert not any(n.startswith("checkpoint-")
+                   for n in os.listdir(job_dir)), os.listdir(job_dir)
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
