@@ -122,6 +122,11 @@ class _AdminServer(object):
                         self._reply(404)
                 except KeyError as e:
                     self._reply(400, {"error": str(e)})
+                except ValueError as e:
+                    # Admission rejection (invalid spec, or an attempt
+                    # to re-submit/mutate an existing job's spec —
+                    # reference validator.py:103-113 semantics).
+                    self._reply(422, {"error": str(e)})
 
         self.state_dir = None
         self._server = ThreadingHTTPServer((host, port), Handler)

@@ -115,6 +115,27 @@ class JobSpec(object):
         self.workdir = workdir
         self.restart_limit = restart_limit
         self.preemptible = preemptible
+        self._frozen = False
+
+    def freeze(self):
+        """Make the spec immutable (called at submission).
+
+        Mirrors the reference's update validator (sched/adaptdl_sched/
+        validator.py:103-113): updates to a running job's spec are
+        forbidden — restarts must always relaunch the admitted spec.
+        """
+        import types
+        object.__setattr__(self, "argv", tuple(self.argv))
+        object.__setattr__(self, "env",
+                           types.MappingProxyType(dict(self.env)))
+        object.__setattr__(self, "_frozen", True)
+
+    def __setattr__(self, key, value):
+        if getattr(self, "_frozen", False):
+            raise AttributeError(
+                "updates to a submitted JobSpec are forbidden "
+                "(field {!r}); submit a new job instead".format(key))
+        object.__setattr__(self, key, value)
 
 
 class _Job(object):
@@ -163,8 +184,11 @@ class LocalController(object):
     def submit(self, spec):
         with self._lock:
             if spec.name in self._jobs:
-                raise ValueError("job {} already exists".format(spec.name))
+                raise ValueError("job {} already exists; updates to a "
+                                 "running job's spec are forbidden"
+                                 .format(spec.name))
             os.makedirs(os.path.join(spec.job_dir, "logs"), exist_ok=True)
+            spec.freeze()  # admitted specs are immutable from here on
             job = _Job(spec, time.time())
             self._jobs[spec.name] = job
         _metric("submitted")

@@ -246,3 +246,66 @@ def test_warm_rescale_no_disk_checkpoint(tmp_path, controller):
                    for n in os.listdir(job_dir)), os.listdir(job_dir)
     final = json.load(open(os.path.join(job_dir, "final.json")))
     assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
+
+
+def test_submitted_spec_is_immutable(tmp_path, controller):
+    """Admitted specs are frozen (reference validator.py:103-113:
+    'updates to job spec are forbidden'): daemon-side mutation of a
+    running job's spec must raise, and re-submitting the same name is
+    rejected."""
+    script = tmp_path / "sleep.py"
+    script.write_text("import time; time.sleep(30)\n")
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="frozen-job",
+                   job_dir=job_dir, gpus_per_replica=0, max_replicas=1)
+    spec.max_replicas = 2            # mutable before admission
+    controller.submit(spec)
+    with pytest.raises(AttributeError, match="forbidden"):
+        spec.max_replicas = 4        # frozen after admission
+    with pytest.raises(AttributeError, match="forbidden"):
+        spec.argv = ["true"]
+    with pytest.raises(TypeError):
+        spec.env["INJECTED"] = "1"   # env is a read-only mapping
+    with pytest.raises(ValueError, match="forbidden"):
+        controller.submit(JobSpec(["true"], name="frozen-job",
+                                  job_dir=job_dir, gpus_per_replica=0))
+
+
+def test_external_sigterm_preemption(tmp_path, controller):
+    """Spot-instance-style external preemption: SIGTERM the running
+    replica group from OUTSIDE the controller (as a cloud provider
+    would); workers must checkpoint and exit(143), and the controller
+    must treat it as a graceful preemption and restart the group, with
+    training completing correctly (reference analog:
+    ray/adaptdl_ray/aws/worker.py:34-70 spot reclamation)."""
+    import signal as _signal
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="spot-job",
+                   job_dir=job_dir, min_replicas=1, max_replicas=2,
+                   gpus_per_replica=0)
+    controller.submit(spec)
+
+    deadline = time.time() + 60
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("spot-job")
+        time.sleep(0.1)
+    # External reclamation: signal the worker process groups directly.
+    with controller._lock:
+        pids = [p.pid for p in controller._jobs["spot-job"].procs]
+    assert pids
+    for pid in pids:
+        os.killpg(pid, _signal.SIGTERM)
+
+    state = controller.wait("spot-job", timeout=180)
+    assert state == "Succeeded", controller.status("spot-job")
+    assert controller.status("spot-job")["restarts"] >= 1
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(30))
+    assert max(t["restarts"] for t in trace) >= 1
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
