@@ -175,19 +175,95 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_avg(
     block_reduce_atomic(acc, out);
 }
 
+// Shared body for the Adam-preconditioned sum-of-squares:
+//   acc = sum((g / pinv)^2), pinv = sqrt(v) * inv_corr_sqrt + eps
+// float4-vectorized when g and v are co-aligned (bucket-flat segments
+// share offsets, so they always are in practice).
+__device__ __forceinline__ double precond_sqsum_body(
+        const float* __restrict__ g, const float* __restrict__ v, long n,
+        float ics, float eps) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    if (((size_t)g & 15) != ((size_t)v & 15)) {  // mis-coaligned: scalar
+        for (long j = i; j < n; j += stride) {
+            float pinv = sqrtf(v[j]) * ics + eps;
+            double q = (double)(g[j] / pinv);
+            acc += q * q;
+        }
+        return acc;
+    }
+    long head = (16 - ((size_t)g & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        float pinv = sqrtf(v[j]) * ics + eps;
+        double q = (double)(g[j] / pinv);
+        acc += q * q;
+    }
+    const float4* gv = (const float4*)(g + head);
+    const float4* vv = (const float4*)(v + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 gg = gv[j];
+        float4 s = vv[j];
+        double q0 = (double)(gg.x / (sqrtf(s.x) * ics + eps));
+        double q1 = (double)(gg.y / (sqrtf(s.y) * ics + eps));
+        double q2 = (double)(gg.z / (sqrtf(s.z) * ics + eps));
+        double q3 = (double)(gg.w / (sqrtf(s.w) * ics + eps));
+        acc += q0 * q0 + q1 * q1 + q2 * q2 + q3 * q3;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        float pinv = sqrtf(v[j]) * ics + eps;
+        double q = (double)(g[j] / pinv);
+        acc += q * q;
+    }
+    return acc;
+}
+
+__device__ __forceinline__ double plain_sqsum_body(
+        const float* __restrict__ x, long n) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = (16 - ((size_t)x & 15)) / 4 & 3;
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        double q = (double)x[j];
+        acc += q * q;
+    }
+    const float4* xv = (const float4*)(x + head);
+    const long nv = (n - head) / 4;
+    for (long j = i; j < nv; j += stride) {
+        float4 c = xv[j];
+        acc += (double)c.x * c.x + (double)c.y * c.y
+             + (double)c.z * c.z + (double)c.w * c.w;
+    }
+    for (long j = head + nv * 4 + i; j < n; j += stride) {
+        double q = (double)x[j];
+        acc += q * q;
+    }
+    return acc;
+}
+
 // out += sum((g / pinv)^2), pinv = sqrt(v / corr) + eps (Adam precondition;
 // reference AdamGradientNoiseScale._calculate_preconditioner)
 extern "C" __global__ __launch_bounds__(BLOCK) void k_precond_sqsum(
         const float* __restrict__ g, const float* __restrict__ v, long n,
         float inv_corr_sqrt, float eps, double* __restrict__ out) {
-    double acc = 0.0;
-    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
-    const long stride = (long)gridDim.x * BLOCK;
-    for (long j = i; j < n; j += stride) {
-        float pinv = sqrtf(v[j]) * inv_corr_sqrt + eps;
-        double q = (double)(g[j] / pinv);
-        acc += q * q;
-    }
+    double acc = precond_sqsum_body(g, v, n, inv_corr_sqrt, eps);
+    block_reduce_atomic(acc, out);
+}
+
+// hipGraph-capturable variant: the bias-correction scalars live in
+// DEVICE memory (pc = {inv_corr_sqrt, eps, use_precond, unused}) so a
+// captured graph picks up each step's values instead of baking them in.
+// use_precond == 0 -> identity preconditioner (warmup steps).
+extern "C" __global__ __launch_bounds__(BLOCK) void k_precond_sqsum_dev(
+        const float* __restrict__ g, const float* __restrict__ v, long n,
+        const float* __restrict__ pc, double* __restrict__ out) {
+    double acc = (pc[2] != 0.f)
+        ? precond_sqsum_body(g, v, n, pc[0], pc[1])
+        : plain_sqsum_body(g, n);
     block_reduce_atomic(acc, out);
 }
 
@@ -277,6 +353,13 @@ extern "C" void launch_precond_sqsum(const float* g, const float* v, long n,
                                      double* out, hipStream_t s) {
     hipLaunchKernelGGL(k_precond_sqsum, grid_for(n), dim3(BLOCK), 0, s, g, v,
                        n, inv_corr_sqrt, eps, out);
+}
+
+extern "C" void launch_precond_sqsum_dev(const float* g, const float* v,
+                                         long n, const float* pc,
+                                         double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_precond_sqsum_dev, grid_for(n), dim3(BLOCK), 0, s,
+                       g, v, n, pc, out);
 }
 
 extern "C" void launch_fused_sgd(float* p, const float* g, float* m, long n,

@@ -61,6 +61,18 @@ class FusedSGD(torch.optim.SGD):
         return bucket.sgd_momentum
 
     @torch.no_grad()
+    def reset_state(self):
+        """Zero the momentum buffers (bench statistics-probe isolation)."""
+        if self._engine is None:
+            for st in self.state.values():
+                if st.get("momentum_buffer") is not None:
+                    st["momentum_buffer"].zero_()
+            return
+        for bucket in self._engine.buckets:
+            if bucket.sgd_momentum is not None:
+                bucket.sgd_momentum.zero_()
+
+    @torch.no_grad()
     def step(self, closure=None):
         if self._engine is None:
             return super().step(closure)
@@ -146,6 +158,25 @@ class _FusedAdamBase(torch.optim.AdamW):
             for p, _, _ in bucket.segments:
                 self.state[p]["step"].fill_(state["step"])
         return loss
+
+    @torch.no_grad()
+    def reset_state(self):
+        """Zero Adam moments and step counts (bench probe isolation)."""
+        if self._engine is not None:
+            for bucket in self._engine.buckets:
+                if bucket.adam_state is not None:
+                    bucket.adam_state["exp_avg"].zero_()
+                    bucket.adam_state["exp_avg_sq"].zero_()
+                    bucket.adam_state["step"] = 0
+        for st in self.state.values():
+            for key in ("exp_avg", "exp_avg_sq"):
+                if st.get(key) is not None:
+                    st[key].zero_()
+            step = st.get("step")
+            if isinstance(step, torch.Tensor):
+                step.zero_()
+            elif step is not None:
+                st["step"] = 0
 
     def load_state_dict(self, state_dict):
         super().load_state_dict(state_dict)

@@ -79,6 +79,14 @@ class Workload(object):
     def build(self):
         raise NotImplementedError
 
+    def prep(self, batch):
+        """Host-side batch preparation (H2D transfer etc.).
+
+        Runs OUTSIDE the hipGraph capture: pageable H2D copies are
+        illegal inside stream capture, so fwd_bwd must only consume
+        device tensors produced here."""
+        return batch
+
     def fwd_bwd(self, batch):
         raise NotImplementedError
 
@@ -143,8 +151,10 @@ class _ConvWorkload(Workload):
         self.loader = loader
         return self.adp, self.optim, loader
 
+    def prep(self, idx):
+        return (idx % self.pool_size).to(self.device, non_blocking=True)
+
     def fwd_bwd(self, idx):
-        idx = (idx % self.pool_size).to(self.device, non_blocking=True)
         x = self.pool_x[idx]
         if self.channels_last:
             x = x.contiguous(memory_format=torch.channels_last)
@@ -269,8 +279,10 @@ class BertBase(Workload):
         self.loader = loader
         return self.adp, self.optim, loader
 
+    def prep(self, idx):
+        return (idx % self.pool_size).to(self.device, non_blocking=True)
+
     def fwd_bwd(self, idx):
-        idx = (idx % self.pool_size).to(self.device, non_blocking=True)
         x = self.pool_x[idx]
         y = self.pool_y[idx]
         self.optim.zero_grad()
@@ -362,14 +374,22 @@ def main():
     probing = args.eff_probe > 0 and args.max_batch > 0
     probe_grad_params = None
     saved_lr = None
+    gns_backup = None
     if probing:
         saved_lr = [pg["lr"] for pg in optim.param_groups]
         for pg in optim.param_groups:
             pg["lr"] = 0.0
+        # Snapshot the GNS estimator state so the probe leaves the
+        # training trajectory (and the goodput model's own batch-size
+        # choice) untouched — the probe only produces the deterministic
+        # grad params used for the final efficiency quote.
+        import copy
+        gns_backup = copy.deepcopy(dict(optim.state["gns"]))
 
     graph_stepper = None  # ADAPTDL_HIPGRAPH=1: set before the probe pass
 
     def train_step(batch):
+        batch = workload.prep(batch)  # H2D etc., outside any capture
         if graph_stepper is not None:
             graph_stepper.microbatch(batch)
         else:
@@ -400,12 +420,23 @@ def main():
                 optim_steps += 1
                 if phase == "effprobe" and optim_steps >= args.eff_probe:
                     # End of the lr=0 statistics probe: snapshot the
-                    # deterministic grad params, restore the real LR,
-                    # and fall through into the timing warmup.
+                    # deterministic grad params, then restore LR, GNS
+                    # estimator state, optimizer state (momentum/Adam
+                    # moments charged at lr=0), and the reported grad
+                    # params, so warmup training proceeds exactly as it
+                    # would have without the probe.
                     probe_grad_params = \
                         _metrics._metrics_state().grad_params
                     for pg, lr in zip(optim.param_groups, saved_lr):
                         pg["lr"] = lr
+                    optim.state["gns"].clear()
+                    optim.state["gns"].update(gns_backup)
+                    adp.gns._prev_total_sqr = None
+                    if hasattr(optim, "reset_state"):
+                        optim.reset_state()
+                    _metrics._metrics_state().grad_params = None
+                    _metrics.update_progress(
+                        optim.state["gns"]["progress"])
                     optim_steps = 0
                     phase = "warmup"
                 elif phase == "warmup" and optim_steps >= args.warmup:

@@ -150,6 +150,13 @@ def test_hipgraph_stepper_matches_eager(tmp_ckpt_env):
         # nondeterminism of the backward kernels) and one graphed.
         for run_id, graphed in (("eager", False), ("eager2", False),
                                 ("graphed", True)):
+            # Fresh elastic context per run: the previous run's loader
+            # would otherwise stay marked as THE training loader and the
+            # new run would never engage profiling or the stepper (this
+            # was the r1/r2 A/B failure: 0 captures, all-eager).
+            AdaptiveDataLoaderHelper._training = None
+            AdaptiveDataLoaderHelper._current = None
+            AdaptiveDataLoaderHelper._position.clear()
             torch.manual_seed(7)
             model = ResNet18().to(device) \
                 .to(memory_format=torch.channels_last)
