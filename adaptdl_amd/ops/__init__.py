@@ -176,3 +176,37 @@ def precond_sqsum(grad, exp_avg_sq, beta2, eps, step, out):
         corr = 1.0 - beta2 ** step
         pinv = (_flat(exp_avg_sq).double() / corr).sqrt().add_(eps)
         out.add_((_flat(grad).double() / pinv).pow(2).sum())
+
+
+def set_precond_scalars(pc, beta2, eps, step, min_steps=5):
+    """Write the device-resident preconditioner scalars for
+    :func:`precond_sqsum_dev`: pc = {inv_corr_sqrt, eps, use_precond, 0}.
+
+    Identity preconditioning (use_precond=0) below ``min_steps`` — the
+    same warmup gate the per-segment path applies host-side.  Called
+    from the (eager) fused-Adam step, so hipGraph replays of the
+    statistics kernels pick up each step's bias correction.
+    """
+    import math
+    if step >= min_steps:
+        vals = [1.0 / math.sqrt(1.0 - beta2 ** step), eps, 1.0, 0.0]
+    else:
+        vals = [1.0, 0.0, 0.0, 0.0]
+    pc.copy_(torch.tensor(vals, dtype=torch.float32))
+
+
+def precond_sqsum_dev(grad, exp_avg_sq, pc, out):
+    """out += sum((grad / pinv)**2) with the preconditioner scalars read
+    from device tensor ``pc`` (see :func:`set_precond_scalars`) — one
+    launch per flat bucket, safe inside hipGraph capture."""
+    ext = _require_ext(grad)
+    if ext is not None:
+        ext.precond_sqsum_dev(_flat(grad), _flat(exp_avg_sq), pc, out)
+    else:
+        ics, eps, use_p = (float(pc[0]), float(pc[1]), float(pc[2]))
+        g = _flat(grad).double()
+        if use_p == 0.0:
+            out.add_(g.pow(2).sum())
+        else:
+            pinv = _flat(exp_avg_sq).double().sqrt().mul_(ics).add_(eps)
+            out.add_((g / pinv).pow(2).sum())

@@ -19,6 +19,8 @@ void launch_sqsum_avg(const float*, const float*, long, double*,
                       hipStream_t);
 void launch_precond_sqsum(const float*, const float*, long, float, float,
                           double*, hipStream_t);
+void launch_precond_sqsum_dev(const float*, const float*, long,
+                              const float*, double*, hipStream_t);
 void launch_fused_sgd(float*, const float*, float*, long, float, float,
                       float, float, int, int, hipStream_t);
 void launch_fused_adamw(float*, const float*, float*, float*, long, float,
@@ -133,6 +135,22 @@ void precond_sqsum(torch::Tensor g, torch::Tensor v, double beta2,
     launch_precond_sqsum(g.data_ptr<float>(), v.data_ptr<float>(), n,
                          inv_corr_sqrt, (float)eps, out.data_ptr<double>(),
                          stream());
+}
+
+void precond_sqsum_dev(torch::Tensor g, torch::Tensor v, torch::Tensor pc,
+                       torch::Tensor out) {
+    // hipGraph-capturable variant: bias-correction scalars pc =
+    // {inv_corr_sqrt, eps, use_precond, unused} live in device memory
+    // (updated by the eager fused-Adam step between replays).
+    check_f32(g, "g"); check_f32(v, "v"); check_out(out);
+    check_f32(pc, "pc");
+    TORCH_CHECK(g.numel() == v.numel(), "g/v size mismatch");
+    TORCH_CHECK(pc.numel() >= 3, "pc must hold >= 3 scalars");
+    long n = g.numel();
+    if (n == 0) return;
+    launch_precond_sqsum_dev(g.data_ptr<float>(), v.data_ptr<float>(), n,
+                             pc.data_ptr<float>(), out.data_ptr<double>(),
+                             stream());
 }
 
 void fused_sgd(torch::Tensor p, torch::Tensor g, torch::Tensor m, double lr,
@@ -459,6 +477,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     mod.def("sqsum_avg", &sqsum_avg, "out += sum(((cur+prev)/2)^2)");
     mod.def("precond_sqsum", &precond_sqsum,
             "out += sum((g/pinv)^2), Adam preconditioner");
+    mod.def("precond_sqsum_dev", &precond_sqsum_dev,
+            "precond_sqsum with device-resident scalars (graph-safe)");
     mod.def("fused_sgd", &fused_sgd, "fused flat-bucket SGD step");
     mod.def("fused_adamw", &fused_adamw, "fused flat-bucket Adam(W) step");
     mod.def("bn_fwd", &bn_fwd, "fused NHWC bf16 BatchNorm(+ReLU) forward");

@@ -55,9 +55,11 @@ Safety rules (all checked per cycle; violation falls back to eager):
   re-warms,
 - the observed sync position must match the expected one,
 - AMP GradScaler (mp_scaler) is not supported (its inf-check host logic
-  inspects gradients between backward and step), and neither is the
-  Adam-preconditioned GNS (its statistic kernels take the per-step bias
-  correction as a kernel argument); construction refuses both,
+  inspects gradients between backward and step); Adam-preconditioned
+  GNS is supported with the fused optimizers (whose statistic kernels
+  read the bias-correction scalars from device memory) but refused for
+  stock torch.optim.Adam (per-step scalars would be baked into the
+  capture),
 - any capture error permanently disables the stepper for the run (the
   current microbatch is re-run eagerly).
 
@@ -178,15 +180,20 @@ class GraphedStepper(object):
         if getattr(adp.gns, "_mp_scaler", None) is not None:
             raise ValueError("GraphedStepper does not support mp_scaler "
                              "(GradScaler host logic is not capturable)")
-        if isinstance(adp.gns, AdamGradientNoiseScale):
-            # The preconditioned statistic kernels take the Adam step
-            # count as a kernel argument (bias correction changes every
-            # step) and read it with a host sync — a captured graph
-            # would bake a stale step in.  SGD-family only for now.
-            raise ValueError("GraphedStepper does not support Adam-"
-                             "preconditioned GNS (per-step bias "
-                             "correction is baked into captured "
-                             "statistic kernels)")
+        if isinstance(adp.gns, AdamGradientNoiseScale) and \
+                not hasattr(optimizer, "attach_engine"):
+            # With the FUSED Adam optimizers the preconditioned
+            # statistics run one graph-safe kernel per bucket whose
+            # bias-correction scalars live in device memory (updated by
+            # the eager optimizer step between replays), so capture is
+            # supported.  The stock torch.optim.Adam path bakes the
+            # host-read step count into per-segment kernels and stays
+            # refused.
+            raise ValueError("GraphedStepper supports Adam-"
+                             "preconditioned GNS only with the fused "
+                             "optimizers (FusedAdam/FusedAdamW); stock "
+                             "torch.optim.Adam bakes per-step bias "
+                             "correction into captured kernels")
         self._adp = adp
         self._gns = adp.gns
         self._engine = adp.gns.engine

@@ -127,8 +127,16 @@ class _FusedAdamBase(torch.optim.AdamW):
         if bucket.adam_state is None:
             exp_avg = torch.zeros_like(bucket.flat)
             exp_avg_sq = torch.zeros_like(bucket.flat)
+            # Device-resident preconditioner scalars for the one-launch-
+            # per-bucket GNS statistics (identity until step >= 5).
+            precond = torch.zeros(4, dtype=torch.float32,
+                                  device=bucket.flat.device)
+            ops.set_precond_scalars(
+                precond, self.param_groups[bucket.group_idx]["betas"][1],
+                self.param_groups[bucket.group_idx]["eps"], 0)
             bucket.adam_state = {"exp_avg": exp_avg,
-                                 "exp_avg_sq": exp_avg_sq, "step": 0}
+                                 "exp_avg_sq": exp_avg_sq, "step": 0,
+                                 "precond": precond}
             for p, off, n in bucket.segments:
                 self.state[p]["step"] = torch.tensor(0.0)
                 self.state[p]["exp_avg"] = _segment_view(
@@ -155,6 +163,10 @@ class _FusedAdamBase(torch.optim.AdamW):
                 state["exp_avg_sq"], group["lr"], beta1, beta2,
                 group["eps"], group["weight_decay"], state["step"],
                 self._adam_mode)
+            # Refresh the device-side preconditioner scalars so graphed
+            # statistic replays see the new step's bias correction.
+            ops.set_precond_scalars(state["precond"], beta2,
+                                    group["eps"], state["step"])
             for p, _, _ in bucket.segments:
                 self.state[p]["step"].fill_(state["step"])
         return loss
@@ -168,6 +180,10 @@ class _FusedAdamBase(torch.optim.AdamW):
                     bucket.adam_state["exp_avg"].zero_()
                     bucket.adam_state["exp_avg_sq"].zero_()
                     bucket.adam_state["step"] = 0
+                    group = self.param_groups[bucket.group_idx]
+                    ops.set_precond_scalars(bucket.adam_state["precond"],
+                                            group["betas"][1],
+                                            group["eps"], 0)
         for st in self.state.values():
             for key in ("exp_avg", "exp_avg_sq"):
                 if st.get(key) is not None:
@@ -187,7 +203,10 @@ class _FusedAdamBase(torch.optim.AdamW):
                 bucket.adam_state = {
                     "exp_avg": torch.zeros_like(bucket.flat),
                     "exp_avg_sq": torch.zeros_like(bucket.flat),
-                    "step": 0}
+                    "step": 0,
+                    "precond": torch.zeros(
+                        4, dtype=torch.float32,
+                        device=bucket.flat.device)}
             for p, off, n in bucket.segments:
                 st = self.state.get(p)
                 if not st or "exp_avg" not in st:
@@ -201,6 +220,10 @@ class _FusedAdamBase(torch.optim.AdamW):
                 bucket.adam_state["step"] = int(
                     step.item() if isinstance(step, torch.Tensor)
                     else step)
+            group = self.param_groups[bucket.group_idx]
+            ops.set_precond_scalars(bucket.adam_state["precond"],
+                                    group["betas"][1], group["eps"],
+                                    bucket.adam_state["step"])
 
 
 class FusedAdamW(_FusedAdamBase):

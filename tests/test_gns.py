@@ -155,3 +155,70 @@ def test_nan_gradient_skipped():
     x = torch.tensor([[float("nan"), 1.0, 1.0, 1.0]])
     model(x).sum().backward()
     assert np.allclose(gns._state["sqr_avg"], sqr_before)
+
+
+def test_precond_sqsum_dev_cpu_fallback_matches():
+    """Device-scalar preconditioned sqsum: CPU fallback math matches the
+    host-scalar variant in both modes."""
+    from adaptdl_amd import ops as _ops
+    torch.manual_seed(3)
+    g = torch.randn(1001)
+    v = torch.rand(1001)
+    beta2, eps, step = 0.999, 1e-8, 17
+    pc = torch.zeros(4)
+    _ops.set_precond_scalars(pc, beta2, eps, step)
+    out_dev = torch.zeros((), dtype=torch.float64)
+    _ops.precond_sqsum_dev(g, v, pc, out_dev)
+    out_ref = torch.zeros((), dtype=torch.float64)
+    _ops.precond_sqsum(g, v, beta2, eps, step, out_ref)
+    assert torch.allclose(out_dev, out_ref, rtol=1e-6)
+    # Warmup gate -> identity preconditioner -> plain sqsum.
+    _ops.set_precond_scalars(pc, beta2, eps, 1)
+    out_id = torch.zeros((), dtype=torch.float64)
+    _ops.precond_sqsum_dev(g, v, pc, out_id)
+    assert torch.allclose(out_id, g.double().pow(2).sum(), rtol=1e-12)
+
+
+def test_adam_precond_stats_one_launch_per_bucket(tmp_ckpt_env):
+    """With FusedAdam, the Adam-preconditioned statistics pass must use
+    the whole-bucket device-scalar kernel (O(buckets) launches), not the
+    per-segment fallback (VERDICT r1 task 7)."""
+    from unittest import mock
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd import ops as _ops
+
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                torch.nn.Linear(16, 4))  # 4 params
+    optim = adl.FusedAdam(model.parameters(), lr=1e-3)
+    adp = adl.AdaptiveDataParallel(model, optim, name="adam-batched")
+    loader = adl.AdaptiveDataLoader(
+        torch.utils.data.TensorDataset(torch.randn(64, 8),
+                                       torch.randint(0, 4, (64,))),
+        batch_size=16)
+
+    dev_calls = []
+    seg_calls = []
+    real_dev = _ops.precond_sqsum_dev
+    real_seg = _ops.precond_sqsum
+    with mock.patch.object(_ops, "precond_sqsum_dev",
+                           side_effect=lambda *a: (dev_calls.append(1),
+                                                   real_dev(*a))), \
+         mock.patch.object(_ops, "precond_sqsum",
+                           side_effect=lambda *a: (seg_calls.append(1),
+                                                   real_seg(*a))):
+        # Re-import the names inside gradient_noise_scale? Not needed:
+        # it calls through the ops module attribute.
+        for _epoch in adl.remaining_epochs_until(2):
+            for x, y in loader:
+                optim.zero_grad()
+                torch.nn.functional.cross_entropy(adp(x), y).backward()
+                optim.step()
+    n_buckets = len(adp.gns.engine.buckets)
+    assert n_buckets >= 1
+    # After the first optimizer step the flat Adam state exists, so all
+    # statistics launches take the one-per-bucket fast path.
+    assert len(dev_calls) > 0
+    assert not seg_calls, "per-segment fallback used with FusedAdam"

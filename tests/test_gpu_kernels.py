@@ -94,6 +94,31 @@ def test_precond_sqsum(n):
     assert torch.allclose(out, ref, rtol=1e-5)
 
 
+@pytest.mark.parametrize("n,off", [(1023, 1), (1 << 18, 0),
+                                   ((1 << 18) + 3, 2)])
+def test_precond_sqsum_dev(n, off):
+    """Device-scalar (hipGraph-safe) variant vs the fp64 reference, in
+    both preconditioning and identity (warmup) modes, incl. unaligned
+    heads of the float4 path."""
+    torch.manual_seed(n + 9)
+    g = _rand(n, off)
+    v = torch.rand(n + off, device="cuda")[off:]
+    beta2, eps, step = 0.999, 1e-8, 42
+    pc = torch.zeros(4, dtype=torch.float32, device="cuda")
+    ops.set_precond_scalars(pc, beta2, eps, step)
+    out = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.precond_sqsum_dev(g, v, pc, out)
+    corr = 1 - beta2 ** step
+    pinv = (v.double() / corr).sqrt() + eps
+    ref = (g.double() / pinv).pow(2).sum()
+    assert torch.allclose(out, ref, rtol=1e-5)
+    # Identity mode (step below the warmup gate) == plain sqsum.
+    ops.set_precond_scalars(pc, beta2, eps, 2)
+    out2 = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.precond_sqsum_dev(g, v, pc, out2)
+    assert torch.allclose(out2, g.double().pow(2).sum(), rtol=1e-10)
+
+
 @pytest.mark.parametrize("momentum,nesterov,wd", [
     (0.0, False, 0.0), (0.9, False, 5e-4), (0.9, True, 5e-4)])
 def test_fused_sgd(momentum, nesterov, wd):
