@@ -73,9 +73,14 @@ struct WrwRegs {
     V16 vx[2][2];
 };
 
+// Wp is the padded line width (next multiple of 8 >= W): pixels with
+// w in [W, Wp) and tail-chunk lines with h >= H stage ZERO dy, so they
+// contribute nothing to the accumulation — this is what lets the
+// ImageNet-resolution widths (56/28/14/7) run on the same MFMA tiling
+// (legacy W in {8,16,32} has Wp == W and the guards are always true).
 __device__ __forceinline__ void wrw_issue(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-        int n, int h0, int H, int W, int C, int K, int kt, int ct,
+        int n, int h0, int H, int W, int Wp, int C, int K, int kt, int ct,
         int CP, int P, int t, WrwRegs& r) {
 #ifdef WRW_PROBE_NOLOAD
     (void)x; (void)dy; (void)n; (void)h0;
@@ -87,39 +92,49 @@ __device__ __forceinline__ void wrw_issue(
         const int i = t + it * 512;
         if (i < dyn8) {
             const int p = (i >> 3) * 2;
+            const int h = h0 + p / Wp;
+            const int w = p % Wp;
             const ushort_t* g = dy +
-                (((size_t)n * H + (h0 + p / W)) * W + p % W) * K
+                (((size_t)n * H + h) * W + w) * K
                 + (size_t)kt * 64 + (i & 7) * 8;
-            r.vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
-            r.vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
+            if (h < H && w < W)
+                r.vdy[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+            else
+                r.vdy[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
+            if (h < H && w + 1 < W)
+                r.vdy[it][1].u4 = *reinterpret_cast<const u32x4*>(g + K);
+            else
+                r.vdy[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
         }
     }
-    const int xn8 = (P + 2) * W * 4;
+    const int xn8 = (P + 2) * Wp * 4;
     #pragma unroll
     for (int it = 0; it < 2; ++it) {
         const int i = t + it * 512;
         if (i < xn8) {
-            const int j = i / (W * 4);
-            const int rem = i % (W * 4);
+            const int j = i / (Wp * 4);
+            const int rem = i % (Wp * 4);
             const int w = (rem >> 3) * 2;
             const int h = h0 - 1 + j;
-            if (h < 0 || h >= H) {
-                r.vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
-                r.vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
-            } else {
-                const ushort_t* g = x +
-                    (((size_t)n * H + h) * W + w) * C
-                    + (size_t)ct * 64 + (rem & 7) * 8;
+            const ushort_t* g = x +
+                (((size_t)n * H + h) * W + w) * C
+                + (size_t)ct * 64 + (rem & 7) * 8;
+            const bool hok = h >= 0 && h < H;
+            if (hok && w < W)
                 r.vx[it][0].u4 = *reinterpret_cast<const u32x4*>(g);
+            else
+                r.vx[it][0].u4 = u32x4{0u, 0u, 0u, 0u};
+            if (hok && w + 1 < W)
                 r.vx[it][1].u4 = *reinterpret_cast<const u32x4*>(g + C);
-            }
+            else
+                r.vx[it][1].u4 = u32x4{0u, 0u, 0u, 0u};
         }
     }
 }
 
 __device__ __forceinline__ void wrw_write(
         short* __restrict__ dy_t, short* __restrict__ x_t,
-        int W, int CP, int P, int LS, int t, WrwRegs& r) {
+        int Wp, int CP, int P, int LS, int t, WrwRegs& r) {
     const int dyn8 = CP * 4;
     #pragma unroll
     for (int it = 0; it < 1; ++it) {
@@ -139,13 +154,13 @@ __device__ __forceinline__ void wrw_write(
             }
         }
     }
-    const int xn8 = (P + 2) * W * 4;
+    const int xn8 = (P + 2) * Wp * 4;
     #pragma unroll
     for (int it = 0; it < 2; ++it) {
         const int i = t + it * 512;
         if (i < xn8) {
-            const int j = i / (W * 4);
-            const int rem = i % (W * 4);
+            const int j = i / (Wp * 4);
+            const int rem = i % (Wp * 4);
             const int w = (rem >> 3) * 2;
             const int cg = (rem & 7) * 8;
             #pragma unroll
@@ -178,12 +193,12 @@ struct WrwTaus {
 template <int WT2>
 __device__ __forceinline__ void wrw_mfma_phase(
         const short* __restrict__ dy_t, const short* __restrict__ x_t,
-        int kchunks, int W, int LS, int wk, int wc, int row16, int slot8,
-        f32x4 (&acc)[2][2][5]) {
+        int kchunks, int wshift, int LS, int wk, int wc, int row16,
+        int slot8, f32x4 (&acc)[2][2][5]) {
     for (int kc = 0; kc < kchunks; ++kc) {
         const int p0 = kc * 32 + slot8;
-        const int li = p0 / W;
-        const int w0 = p0 % W;
+        const int li = p0 >> wshift;           // Wp is a power of two
+        const int w0 = p0 & ((1 << wshift) - 1);
         bf16x8 afrag[2];
         #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
@@ -257,13 +272,14 @@ __device__ __forceinline__ void wrw_mfma_phase(
 // at 1 wave/SIMD; staging only ~35 us).
 extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
         const ushort_t* __restrict__ x, const ushort_t* __restrict__ dy,
-        float* __restrict__ ws, int N, int H, int W, int C, int K,
+        float* __restrict__ ws, int N, int H, int W, int Wp, int C, int K,
         int P, int nsplit) {
     __shared__ short lds[2 * WRW_BUF];
 
-    const int LS = W + 8;            // padded line stride (mult of 8)
-    const int CP = P * W;            // chunk pixels (mult of 32)
+    const int LS = Wp + 8;           // padded line stride (mult of 8)
+    const int CP = P * Wp;           // chunk pixels (mult of 32)
     const int kchunks = CP / 32;
+    const int wshift = 31 - __clz(Wp);
     const int nc = C / 64;
     const int nsplit_t = nsplit;
     const int tile = blockIdx.x / nsplit_t;
@@ -291,7 +307,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
                 for (int r = 0; r < 4; ++r)
                     acc[mf][nf][ti][r] = 0.f;
 
-    const int lines_per_img = H / P;
+    const int lines_per_img = (H + P - 1) / P;  // tail chunks zero-fill
     const long chunks_total = (long)N * lines_per_img;
 
     // Zero the pad columns of BOTH x_t buffers once.
@@ -302,7 +318,7 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
             const int rem = i % ((P + 2) * 8);
             const int j = rem / 8;
             const int pp = rem % 8;
-            const int q = pp < 4 ? pp : (4 + W + (pp - 4));
+            const int q = pp < 4 ? pp : (4 + Wp + (pp - 4));
             x_t[cc * XT_STRIDE + DSWZ(cc, j * LS + q)] = 0;
         }
     }
@@ -315,13 +331,13 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
     // Prologue: stage chunk 0 into buffer 0, issue chunk 1's loads.
     if (split < chunks_total) {
         WRW_NH(split)
-        wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t, regs);
-        wrw_write(lds, lds + 64 * DY_STRIDE, W, CP, P, LS, t, regs);
+        wrw_issue(x, dy, n_, h0_, H, W, Wp, C, K, kt, ct, CP, P, t, regs);
+        wrw_write(lds, lds + 64 * DY_STRIDE, Wp, CP, P, LS, t, regs);
     }
     __syncthreads();
     if (split + nsplit_t < chunks_total) {
         WRW_NH(split + nsplit_t)
-        wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t, regs);
+        wrw_issue(x, dy, n_, h0_, H, W, Wp, C, K, kt, ct, CP, P, t, regs);
     }
 
     int cur = 0;
@@ -331,19 +347,20 @@ extern "C" __global__ __launch_bounds__(512, 2) void k_conv3x3_wrw(
 
         // ---- MFMA phase (chunk q, buffer cur) ----
         if (wt2 == 0)
-            wrw_mfma_phase<0>(dy_t, x_t, kchunks, W, LS, wk, wc, row16,
-                              slot8, acc);
+            wrw_mfma_phase<0>(dy_t, x_t, kchunks, wshift, LS, wk, wc,
+                              row16, slot8, acc);
         else
-            wrw_mfma_phase<1>(dy_t, x_t, kchunks, W, LS, wk, wc, row16,
-                              slot8, acc);
+            wrw_mfma_phase<1>(dy_t, x_t, kchunks, wshift, LS, wk, wc,
+                              row16, slot8, acc);
 
         // ---- stage chunk q+1 into the other buffer; issue q+2 ----
         if (q + nsplit_t < chunks_total) {
             short* ndy = lds + (cur ^ 1) * WRW_BUF;
-            wrw_write(ndy, ndy + 64 * DY_STRIDE, W, CP, P, LS, t, regs);
+            wrw_write(ndy, ndy + 64 * DY_STRIDE, Wp, CP, P, LS, t,
+                      regs);
             if (q + 2 * nsplit_t < chunks_total) {
                 WRW_NH(q + 2 * nsplit_t)
-                wrw_issue(x, dy, n_, h0_, H, W, C, K, kt, ct, CP, P, t,
+                wrw_issue(x, dy, n_, h0_, H, W, Wp, C, K, kt, ct, CP, P, t,
                           regs);
             }
         }
@@ -416,18 +433,37 @@ extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
 
 // ---- host-side launchers ---------------------------------------------
 
-extern "C" int conv3x3_wrw_supported(int H, int W, int C, int K) {
-    if (C % 64 || K % 64) return 0;
-    if (W != 8 && W != 16 && W != 32) return 0;
-    int P = (W == 32) ? 4 : 8;
-    if (H % P) return 0;
+// Per-shape chunking: Wp = padded line width (power of two >= W), P =
+// lines per chunk; CP = P * Wp <= 128 must be a multiple of 32.  The
+// CIFAR widths use full-width chunks (Wp == W); the ImageNet-resolution
+// widths (ResNet-50 at 224: 56/28/14/7) zero-pad the line to Wp and
+// zero-fill tail-chunk lines, trading <= 24% staged-pixel waste for the
+// same aligned-b128 LDS tiling.
+extern "C" int conv3x3_wrw_params(int H, int W, int* P, int* Wp) {
+    switch (W) {
+    case 8:  *Wp = 8;  *P = 8; break;
+    case 16: *Wp = 16; *P = 8; break;
+    case 32: *Wp = 32; *P = 4; break;
+    case 56: *Wp = 64; *P = 1; break;
+    case 28: *Wp = 32; *P = 4; break;
+    case 14: *Wp = 16; *P = 2; break;
+    case 7:  *Wp = 8;  *P = 4; break;
+    default: return 0;
+    }
     if (H < 2) return 0;
     return 1;
 }
 
+extern "C" int conv3x3_wrw_supported(int H, int W, int C, int K) {
+    if (C % 64 || K % 64) return 0;
+    int P, Wp;
+    return conv3x3_wrw_params(H, W, &P, &Wp);
+}
+
 extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
-    int P = (W == 32) ? 4 : 8;
-    long chunks = (long)N * (H / P);
+    int P = 8, Wp = 8;
+    conv3x3_wrw_params(H, W, &P, &Wp);
+    long chunks = (long)N * ((H + P - 1) / P);
     int tiles = (K / 64) * (C / 64);
     long target = 256 / tiles;  // one residency round (1 block/CU at 98 KB LDS)
     if (target < 1) target = 1;
@@ -438,11 +474,12 @@ extern "C" int conv3x3_wrw_nsplit(int N, int H, int W, int C, int K) {
 extern "C" void launch_conv3x3_wrw(
         const ushort_t* x, const ushort_t* dy, float* ws, float* dw,
         int N, int H, int W, int C, int K, hipStream_t s) {
-    const int P = (W == 32) ? 4 : 8;
+    int P = 8, Wp = 8;
+    conv3x3_wrw_params(H, W, &P, &Wp);
     const int nsplit = conv3x3_wrw_nsplit(N, H, W, C, K);
     const int tiles = (K / 64) * (C / 64);
     hipLaunchKernelGGL(k_conv3x3_wrw, dim3(tiles * nsplit), dim3(512), 0,
-                       s, x, dy, ws, N, H, W, C, K, P, nsplit);
+                       s, x, dy, ws, N, H, W, Wp, C, K, P, nsplit);
     const long n = (long)K * 9 * C;
     // 16-slab groups: enough blocks to fill the chip (the 64-slab
     // version ran 144 blocks on layer1 and measured 5x off bandwidth).

@@ -389,10 +389,12 @@ class GraphedStepper(object):
         return out
 
 
-def maybe_graphed_stepper(adp, optimizer, fwd_bwd):
-    """Build a GraphedStepper iff ``ADAPTDL_HIPGRAPH=1`` and a GPU is
-    available; returns None (caller keeps the eager path) otherwise."""
-    if os.getenv("ADAPTDL_HIPGRAPH") != "1":
+def maybe_graphed_stepper(adp, optimizer, fwd_bwd, default_on=False):
+    """Build a GraphedStepper iff ``ADAPTDL_HIPGRAPH=1`` (or unset with
+    ``default_on``) and a GPU is available; returns None (caller keeps
+    the eager path) otherwise."""
+    flag = os.getenv("ADAPTDL_HIPGRAPH", "1" if default_on else "0")
+    if flag != "1":
         return None
     if not torch.cuda.is_available():
         LOG.warning("ADAPTDL_HIPGRAPH=1 ignored: no GPU available")

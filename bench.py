@@ -457,17 +457,23 @@ def main():
                     break
             if restart_pass:
                 restart_pass = False
-                # Experimental hipGraph capture of the steady microbatch
-                # cycle (ADAPTDL_HIPGRAPH=1): created before the probe
-                # pass so warmup + capture complete outside the timed
-                # region (probe runs 5 cycles instead of 2: one eager
-                # warmup cycle, one capture cycle, three replay cycles).
-                if os.getenv("ADAPTDL_HIPGRAPH") == "1" and use_gpu \
-                        and workload.supports_hipgraph:
+                # hipGraph capture of the steady microbatch cycle:
+                # created before the probe pass so warmup + capture
+                # complete outside the timed region (probe runs 5
+                # cycles instead of 2: one eager warmup cycle, one
+                # capture cycle, three replay cycles).  Measured win on
+                # MI355X at N=1: 42.55 vs 43.1-43.4 ms/step with steady
+                # -state GPU idle 19% -> 1.7% (profiles/, round 2), so
+                # it is ON by default for single-GPU runs; replay with
+                # RCCL collectives captured (N>1) is unvalidated on
+                # hardware and stays opt-in.  ADAPTDL_HIPGRAPH=0/1
+                # overrides.  Any capture failure falls back to eager.
+                if use_gpu and workload.supports_hipgraph:
                     from adaptdl_amd.torch.graph_step import \
                         maybe_graphed_stepper
                     graph_stepper = maybe_graphed_stepper(
-                        adp, optim, workload.fwd_bwd)
+                        adp, optim, workload.fwd_bwd,
+                        default_on=(world == 1))
                 probe_target = 2 if graph_stepper is None else 5
                 # One probe pass: let _sync_local_bsz adopt the fitted
                 # model's choice, run steps to settle caches, then time.

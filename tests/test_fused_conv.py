@@ -17,6 +17,11 @@ from adaptdl_amd.torch.layers import FusedConv2d
     (2, 64, 32, 32, 64),      # layer1-like
     (2, 64, 16, 16, 128),     # channel growth
     (3, 128, 16, 16, 128),    # layer2-like
+    # ImageNet-resolution widths (ResNet-50 at 224): padded-width chunks
+    (2, 64, 56, 56, 64),      # r50 layer1 conv2
+    (2, 128, 28, 28, 128),    # r50 layer2 conv2
+    (2, 256, 14, 14, 256),    # r50 layer3 conv2
+    (2, 512, 7, 7, 512),      # r50 layer4 conv2 (odd W + tail lines)
 ])
 def test_wrw_kernel_matches_fp32(shape):
     from adaptdl_amd import ops

@@ -147,3 +147,63 @@ def test_w8b_wave_mapping():
             s = (ph * P2 + (hl >> 1)) * Wo + (wi >> 1)
             ph2, rem = divmod(s, P2 * Wo)
             assert (ph2, rem // Wo, rem % Wo) == (ph, hl >> 1, wi >> 1)
+
+
+def _wrw_params(W):
+    """Mirror of conv3x3_wrw_params (conv_kernels.hip)."""
+    table = {8: (8, 8), 16: (8, 16), 32: (4, 32),
+             56: (1, 64), 28: (4, 32), 14: (2, 16), 7: (4, 8)}
+    P, Wp = table[W]
+    return P, Wp
+
+
+def _simulate_wrw(x, dy, P, Wp):
+    """CPU mirror of k_conv3x3_wrw's padded-width chunking: dy pixels
+    with w >= W or h >= H stage zero; the tap read for dy pixel (li, w)
+    and offset (dh, dw) is x_pad[li + dh][3 + w + dw] in the LS-padded
+    line layout (4 left zeros)."""
+    N, H, W, C = x.shape
+    K = dy.shape[3]
+    LS = Wp + 8
+    lines = -(-H // P)
+    dw_acc = torch.zeros(K, 9, C)
+    for n in range(N):
+        for chunk in range(lines):
+            h0 = chunk * P
+            x_pad = torch.zeros(P + 2, LS, C)
+            for j in range(P + 2):
+                h = h0 - 1 + j
+                if 0 <= h < H:
+                    x_pad[j, 4:4 + W] = x[n, h]
+            dy_pad = torch.zeros(P, Wp, K)
+            for li in range(P):
+                h = h0 + li
+                if h < H:
+                    dy_pad[li, :W] = dy[n, h]
+            for dh in range(3):
+                for dwo in range(3):
+                    tau = dh * 3 + dwo
+                    # x cols for w = 0..Wp-1 at this tap
+                    cols = torch.arange(Wp) + 3 + dwo
+                    xt = x_pad[dh:dh + P, cols]     # (P, Wp, C)
+                    dw_acc[:, tau] += torch.einsum(
+                        "pwk,pwc->kc", dy_pad, xt)
+    return dw_acc
+
+
+def test_wrw_padded_width_chunking():
+    """k_conv3x3_wrw generalized chunking vs torch conv2d_weight for the
+    ImageNet-resolution widths (ResNet-50 at 224) and a legacy width."""
+    torch.manual_seed(1)
+    for (H, W, C, K, N) in [(7, 7, 64, 64, 2), (14, 14, 64, 64, 2),
+                            (28, 28, 64, 64, 1), (56, 56, 64, 64, 1),
+                            (16, 16, 64, 64, 2)]:
+        P, Wp = _wrw_params(W)
+        x = torch.randn(N, H, W, C)
+        dy = torch.randn(N, H, W, K)
+        got = _simulate_wrw(x, dy, P, Wp)
+        ref = torch.nn.grad.conv2d_weight(
+            x.permute(0, 3, 1, 2), (K, C, 3, 3),
+            dy.permute(0, 3, 1, 2), stride=1, padding=1)
+        ref = ref.permute(0, 2, 3, 1).reshape(K, 9, C)
+        assert torch.allclose(got, ref, rtol=1e-4, atol=1e-3), (H, W)

@@ -3,7 +3,10 @@ sys.path.insert(0, __import__("os").path.dirname(__import__("os").path.dirname(_
 from adaptdl_amd import ops
 ext = ops._load_extension()
 dev = torch.device("cuda")
-shapes = [(1024, 64, 32, 32, 64), (1024, 128, 16, 16, 128), (1024, 256, 8, 8, 256)]
+shapes = [(1024, 64, 32, 32, 64), (1024, 128, 16, 16, 128), (1024, 256, 8, 8, 256),
+          # ImageNet-resolution (ResNet-50 at 224) padded-width shapes
+          (256, 64, 56, 56, 64), (256, 128, 28, 28, 128),
+          (256, 256, 14, 14, 256), (256, 512, 7, 7, 512)]
 for n, c, h, w, k in shapes:
     x = torch.randn(n, c, h, w, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
     dy = torch.randn(n, k, h, w, device=dev).to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
