@@ -21,6 +21,13 @@ void launch_precond_sqsum(const float*, const float*, long, float, float,
                           double*, hipStream_t);
 void launch_precond_sqsum_dev(const float*, const float*, long,
                               const float*, double*, hipStream_t);
+void launch_sqsum_bf16(const unsigned short*, long, double*, hipStream_t);
+void launch_scale_sqsum_bf16(unsigned short*, long, float, double*,
+                             hipStream_t);
+void launch_sqsum_diff_update_bf16(const unsigned short*, unsigned short*,
+                                   long, double*, hipStream_t);
+void launch_sqsum_avg_bf16(const unsigned short*, const unsigned short*,
+                           long, double*, hipStream_t);
 void launch_fused_sgd(float*, const float*, float*, long, float, float,
                       float, float, int, int, hipStream_t);
 void launch_fused_adamw(float*, const float*, float*, float*, long, float,
@@ -90,38 +97,80 @@ hipStream_t stream() {
     return at::hip::getCurrentHIPStream().stream();
 }
 
+// The statistics entry points accept float32 (master-grad training)
+// and bfloat16 (true-bf16-parameter models) buckets.
+bool is_bf16(const torch::Tensor& t) {
+    return t.scalar_type() == torch::kBFloat16;
+}
+
+void check_stat_in(const torch::Tensor& t, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
+    TORCH_CHECK(t.scalar_type() == torch::kFloat32 ||
+                t.scalar_type() == torch::kBFloat16,
+                name, " must be float32 or bfloat16");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+unsigned short* bf16_ptr(torch::Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+
 void sqsum(torch::Tensor x, torch::Tensor out) {
-    check_f32(x, "x"); check_out(out);
+    check_stat_in(x, "x"); check_out(out);
     long n = x.numel();
     if (n == 0) return;
-    launch_sqsum(x.data_ptr<float>(), n, out.data_ptr<double>(), stream());
+    if (is_bf16(x))
+        launch_sqsum_bf16(bf16_ptr(x), n, out.data_ptr<double>(),
+                          stream());
+    else
+        launch_sqsum(x.data_ptr<float>(), n, out.data_ptr<double>(),
+                     stream());
 }
 
 void scale_and_sqsum(torch::Tensor x, double scale, torch::Tensor out) {
-    check_f32(x, "x"); check_out(out);
+    check_stat_in(x, "x"); check_out(out);
     long n = x.numel();
     if (n == 0) return;
-    launch_scale_sqsum(x.data_ptr<float>(), n, (float)scale,
-                       out.data_ptr<double>(), stream());
+    if (is_bf16(x))
+        launch_scale_sqsum_bf16(bf16_ptr(x), n, (float)scale,
+                                out.data_ptr<double>(), stream());
+    else
+        launch_scale_sqsum(x.data_ptr<float>(), n, (float)scale,
+                           out.data_ptr<double>(), stream());
 }
 
 void sqsum_diff_update(torch::Tensor cur, torch::Tensor prev,
                        torch::Tensor out) {
-    check_f32(cur, "cur"); check_f32(prev, "prev"); check_out(out);
+    check_stat_in(cur, "cur"); check_stat_in(prev, "prev");
+    check_out(out);
     TORCH_CHECK(cur.numel() == prev.numel(), "cur/prev size mismatch");
+    TORCH_CHECK(cur.scalar_type() == prev.scalar_type(),
+                "cur/prev dtype mismatch");
     long n = cur.numel();
     if (n == 0) return;
-    launch_sqsum_diff_update(cur.data_ptr<float>(), prev.data_ptr<float>(),
-                             n, out.data_ptr<double>(), stream());
+    if (is_bf16(cur))
+        launch_sqsum_diff_update_bf16(bf16_ptr(cur), bf16_ptr(prev), n,
+                                      out.data_ptr<double>(), stream());
+    else
+        launch_sqsum_diff_update(cur.data_ptr<float>(),
+                                 prev.data_ptr<float>(), n,
+                                 out.data_ptr<double>(), stream());
 }
 
 void sqsum_avg(torch::Tensor cur, torch::Tensor prev, torch::Tensor out) {
-    check_f32(cur, "cur"); check_f32(prev, "prev"); check_out(out);
+    check_stat_in(cur, "cur"); check_stat_in(prev, "prev");
+    check_out(out);
     TORCH_CHECK(cur.numel() == prev.numel(), "cur/prev size mismatch");
+    TORCH_CHECK(cur.scalar_type() == prev.scalar_type(),
+                "cur/prev dtype mismatch");
     long n = cur.numel();
     if (n == 0) return;
-    launch_sqsum_avg(cur.data_ptr<float>(), prev.data_ptr<float>(), n,
-                     out.data_ptr<double>(), stream());
+    if (is_bf16(cur))
+        launch_sqsum_avg_bf16(bf16_ptr(cur), bf16_ptr(prev), n,
+                              out.data_ptr<double>(), stream());
+    else
+        launch_sqsum_avg(cur.data_ptr<float>(), prev.data_ptr<float>(), n,
+                         out.data_ptr<double>(), stream());
 }
 
 void precond_sqsum(torch::Tensor g, torch::Tensor v, double beta2,

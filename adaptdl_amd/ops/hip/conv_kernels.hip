@@ -32,6 +32,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <cstdlib>
 
 #define WAVE 64
 
@@ -440,14 +441,19 @@ extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
 // zero-fill tail-chunk lines, trading <= 24% staged-pixel waste for the
 // same aligned-b128 LDS tiling.
 extern "C" int conv3x3_wrw_params(int H, int W, int* P, int* Wp) {
+    // W=14/7 A/B (r2 pass d): the exact-cover small chunks (P=2/P=4,
+    // CP=32) lost ~20% to MIOpen - mostly barrier rounds per pixel
+    // (kchunks=1).  ADAPTDL_WRW_BIGCHUNK=1 trades utilization (87->77%)
+    // for 4x larger chunks; keep whichever wins the next timing pass.
+    const bool big = getenv("ADAPTDL_WRW_BIGCHUNK") != nullptr;
     switch (W) {
     case 8:  *Wp = 8;  *P = 8; break;
     case 16: *Wp = 16; *P = 8; break;
     case 32: *Wp = 32; *P = 4; break;
     case 56: *Wp = 64; *P = 1; break;
     case 28: *Wp = 32; *P = 4; break;
-    case 14: *Wp = 16; *P = 2; break;
-    case 7:  *Wp = 8;  *P = 4; break;
+    case 14: *Wp = 16; *P = big ? 8 : 2; break;
+    case 7:  *Wp = 8;  *P = big ? 8 : 4; break;
     default: return 0;
     }
     if (H < 2) return 0;

@@ -19,6 +19,7 @@
 //   gradient bytes are touched exactly once more than strictly necessary
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 #define WAVE 64
 #define BLOCK 256
@@ -170,6 +171,158 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_avg(
     }
     for (long j = head + nv * 4 + i; j < n; j += stride) {
         double a = 0.5 * ((double)cur[j] + (double)prev[j]);
+        acc += a * a;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+// ---- bf16 variants (true-bf16-parameter models).
+// Same structure as the f32 kernels: grid-stride with a 16-byte-aligned
+// 8-wide vector body (8 bf16 = 16 B), fp64 accumulation; the in-place
+// variants round back to bf16 with RNE.
+
+typedef __attribute__((ext_vector_type(8))) unsigned short us16x8;
+
+__device__ __forceinline__ float bf2f(unsigned short h) {
+    union { float f; unsigned u; } c;
+    c.u = ((unsigned)h) << 16;
+    return c.f;
+}
+
+__device__ __forceinline__ unsigned short f2b(float f) {
+    __hip_bfloat16 h = __float2bfloat16(f);  // RNE
+    return *reinterpret_cast<unsigned short*>(&h);
+}
+
+#define BF16_HEAD(ptr) ((16 - ((size_t)(ptr) & 15)) / 2 & 7)
+
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_bf16(
+        const unsigned short* __restrict__ x, long n,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = BF16_HEAD(x);
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        double v = (double)bf2f(x[j]);
+        acc += v * v;
+    }
+    const us16x8* xv = (const us16x8*)(x + head);
+    const long nv = (n - head) / 8;
+    for (long j = i; j < nv; j += stride) {
+        us16x8 v = xv[j];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            double q = (double)bf2f(v[k]);
+            acc += q * q;
+        }
+    }
+    for (long j = head + nv * 8 + i; j < n; j += stride) {
+        double v = (double)bf2f(x[j]);
+        acc += v * v;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK) void k_scale_sqsum_bf16(
+        unsigned short* __restrict__ x, long n, float scale,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = BF16_HEAD(x);
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        float v = bf2f(x[j]) * scale;
+        x[j] = f2b(v);
+        acc += (double)v * v;
+    }
+    us16x8* xv = (us16x8*)(x + head);
+    const long nv = (n - head) / 8;
+    for (long j = i; j < nv; j += stride) {
+        us16x8 v = xv[j];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            float q = bf2f(v[k]) * scale;
+            v[k] = f2b(q);
+            acc += (double)q * q;
+        }
+        xv[j] = v;
+    }
+    for (long j = head + nv * 8 + i; j < n; j += stride) {
+        float v = bf2f(x[j]) * scale;
+        x[j] = f2b(v);
+        acc += (double)v * v;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK)
+void k_sqsum_diff_update_bf16(
+        const unsigned short* __restrict__ cur,
+        unsigned short* __restrict__ prev, long n,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = BF16_HEAD(cur);
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        unsigned short c = cur[j];
+        double d = (double)bf2f(c) - (double)bf2f(prev[j]);
+        prev[j] = c;
+        acc += d * d;
+    }
+    const us16x8* cv = (const us16x8*)(cur + head);
+    us16x8* pv = (us16x8*)(prev + head);
+    const long nv = (n - head) / 8;
+    for (long j = i; j < nv; j += stride) {
+        us16x8 c = cv[j];
+        us16x8 p = pv[j];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            double d = (double)bf2f(c[k]) - (double)bf2f(p[k]);
+            acc += d * d;
+        }
+        pv[j] = c;
+    }
+    for (long j = head + nv * 8 + i; j < n; j += stride) {
+        unsigned short c = cur[j];
+        double d = (double)bf2f(c) - (double)bf2f(prev[j]);
+        prev[j] = c;
+        acc += d * d;
+    }
+    block_reduce_atomic(acc, out);
+}
+
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_avg_bf16(
+        const unsigned short* __restrict__ cur,
+        const unsigned short* __restrict__ prev, long n,
+        double* __restrict__ out) {
+    double acc = 0.0;
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long stride = (long)gridDim.x * BLOCK;
+    long head = BF16_HEAD(cur);
+    if (head > n) head = n;
+    for (long j = i; j < head; j += stride) {
+        double a = 0.5 * ((double)bf2f(cur[j]) + (double)bf2f(prev[j]));
+        acc += a * a;
+    }
+    const us16x8* cv = (const us16x8*)(cur + head);
+    const us16x8* pv = (const us16x8*)(prev + head);
+    const long nv = (n - head) / 8;
+    for (long j = i; j < nv; j += stride) {
+        us16x8 c = cv[j];
+        us16x8 p = pv[j];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            double a = 0.5 * ((double)bf2f(c[k]) + (double)bf2f(p[k]));
+            acc += a * a;
+        }
+    }
+    for (long j = head + nv * 8 + i; j < n; j += stride) {
+        double a = 0.5 * ((double)bf2f(cur[j]) + (double)bf2f(prev[j]));
         acc += a * a;
     }
     block_reduce_atomic(acc, out);
@@ -353,6 +506,33 @@ extern "C" void launch_precond_sqsum(const float* g, const float* v, long n,
                                      double* out, hipStream_t s) {
     hipLaunchKernelGGL(k_precond_sqsum, grid_for(n), dim3(BLOCK), 0, s, g, v,
                        n, inv_corr_sqrt, eps, out);
+}
+
+extern "C" void launch_sqsum_bf16(const unsigned short* x, long n,
+                                  double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_bf16, grid_for(n), dim3(BLOCK), 0, s, x, n,
+                       out);
+}
+
+extern "C" void launch_scale_sqsum_bf16(unsigned short* x, long n,
+                                        float scale, double* out,
+                                        hipStream_t s) {
+    hipLaunchKernelGGL(k_scale_sqsum_bf16, grid_for(n), dim3(BLOCK), 0, s,
+                       x, n, scale, out);
+}
+
+extern "C" void launch_sqsum_diff_update_bf16(const unsigned short* cur,
+                                              unsigned short* prev, long n,
+                                              double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_diff_update_bf16, grid_for(n), dim3(BLOCK),
+                       0, s, cur, prev, n, out);
+}
+
+extern "C" void launch_sqsum_avg_bf16(const unsigned short* cur,
+                                      const unsigned short* prev, long n,
+                                      double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_avg_bf16, grid_for(n), dim3(BLOCK), 0, s,
+                       cur, prev, n, out);
 }
 
 extern "C" void launch_precond_sqsum_dev(const float* g, const float* v,

@@ -28,6 +28,16 @@ __all__ = ["FusedSGD", "FusedAdam", "FusedAdamW"]
 def _flatten_engine_params(engine):
     """Move every bucket's parameters into a flat buffer (views back)."""
     for bucket in engine.buckets:
+        if bucket.flat.dtype != torch.float32:
+            # The fused update kernels are fp32 (master-precision
+            # training).  True-bf16-parameter models train through the
+            # stock torch optimizers (the engine's statistics kernels
+            # DO handle bf16 buckets).
+            raise ValueError(
+                "Fused optimizers require float32 parameters/gradients; "
+                "got a {} bucket.  Use torch.optim.SGD/Adam(W) for "
+                "low-precision-parameter models.".format(
+                    bucket.flat.dtype))
         if getattr(bucket, "param_flat", None) is not None:
             continue
         flat = torch.empty_like(bucket.flat)

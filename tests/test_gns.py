@@ -222,3 +222,41 @@ def test_adam_precond_stats_one_launch_per_bucket(tmp_ckpt_env):
     # statistics launches take the one-per-bucket fast path.
     assert len(dev_calls) > 0
     assert not seg_calls, "per-segment fallback used with FusedAdam"
+
+
+def test_bf16_parameter_model_trains(tmp_ckpt_env):
+    """True-bf16-parameter model through the engine: statistics run on
+    bf16 buckets; fused optimizers refuse clearly."""
+    import pytest
+    import adaptdl_amd.torch as adl
+
+    if not collective.initialized():
+        collective.initialize(master_addr="127.0.0.1")
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
+                                torch.nn.Linear(16, 4)).to(torch.bfloat16)
+    optim = torch.optim.SGD(model.parameters(), lr=0.05)
+    adp = adl.AdaptiveDataParallel(model, optim, name="bf16-model")
+    assert all(b.flat.dtype == torch.bfloat16
+               for b in adp.gns.engine.buckets)
+    xs = torch.randn(64, 8).to(torch.bfloat16)
+    ys = torch.randint(0, 4, (64,))
+    loader = adl.AdaptiveDataLoader(
+        torch.utils.data.TensorDataset(xs, ys), batch_size=16)
+    for _epoch in adl.remaining_epochs_until(2):
+        for x, y in loader:
+            optim.zero_grad()
+            loss = torch.nn.functional.cross_entropy(
+                adp(x).float(), y)
+            loss.backward()
+            optim.step()
+    assert np.isfinite(adp.gns.sqr_avg())
+    assert np.isfinite(adp.gns.var_avg())
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
+
+    # Fused optimizers state their fp32 requirement up front.
+    model2 = torch.nn.Linear(4, 2).to(torch.bfloat16)
+    optim2 = adl.FusedSGD(model2.parameters(), lr=0.1)
+    with pytest.raises(ValueError, match="float32"):
+        adl.AdaptiveDataParallel(model2, optim2, name="bf16-fused")

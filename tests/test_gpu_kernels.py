@@ -200,3 +200,37 @@ def test_gns_gpu_matches_cpu_oracle():
     assert np.allclose(w.grad.cpu().numpy().ravel(), mean, atol=1e-6)
     assert np.isclose(gns._state["sqr_avg"][0], grad_sqr, rtol=1e-4)
     assert np.isclose(gns._state["var_avg"][0], grad_var, rtol=1e-4)
+
+
+@pytest.mark.parametrize("n,off", [(17, 1), (4097, 2), ((1 << 20) + 5, 3)])
+def test_bf16_stat_kernels(n, off):
+    """bf16 variants of the four statistics kernels vs fp64 references
+    (true-bf16-parameter models)."""
+    torch.manual_seed(n + 31)
+    base = torch.randn(n + off, device="cuda").to(torch.bfloat16)
+    x = base[off:].clone()
+    prev = (torch.randn(n + off, device="cuda")
+            .to(torch.bfloat16))[off:].clone()
+
+    out = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.sqsum(x, out)
+    assert torch.allclose(out, x.double().pow(2).sum(), rtol=1e-10)
+
+    x2 = x.clone()
+    out = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.scale_and_sqsum(x2, 0.25, out)
+    scaled = (x.float() * 0.25).to(torch.bfloat16)
+    assert torch.equal(x2, scaled)  # RNE round-trip must match torch
+    assert torch.allclose(out, scaled.double().pow(2).sum(), rtol=1e-10)
+
+    cur, pr = x.clone(), prev.clone()
+    out = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.sqsum_diff_update(cur, pr, out)
+    ref = (x.double() - prev.double()).pow(2).sum()
+    assert torch.allclose(out, ref, rtol=1e-10)
+    assert torch.equal(pr, x)  # prev updated to cur
+
+    out = torch.zeros((), dtype=torch.float64, device="cuda")
+    ops.sqsum_avg(x, prev, out)
+    ref = ((x.double() + prev.double()) / 2).pow(2).sum()
+    assert torch.allclose(out, ref, rtol=1e-10)
