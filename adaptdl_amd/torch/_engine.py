@@ -21,7 +21,11 @@ engine that owns the gradient memory:
 
 Unused-parameter handling is correct by construction: every bucket is
 all-reduced on every synchronized step (never-computed gradients stay zero),
-so replicas can never disagree on the reduction schedule.
+so replicas can never disagree on the reduction schedule.  For models whose
+parameter usage is identical across replicas but varies step-to-step
+(e.g. multiple ADP instances trained alternately), ``ADAPTDL_SKIP_UNUSED_
+BUCKETS=1`` skips the all-reduce and statistics launches of buckets that
+produced no gradient in the current cycle.
 """
 
 import logging
@@ -60,7 +64,8 @@ def _segment_view(seg, p):
 
 class Bucket(object):
     __slots__ = ["group_idx", "flat", "segments", "ready", "work", "prev",
-                 "param_flat", "sgd_momentum", "adam_state"]
+                 "param_flat", "sgd_momentum", "adam_state", "used",
+                 "prev_cycle_valid"]
 
     def __init__(self, group_idx, flat, segments):
         self.group_idx = group_idx
@@ -72,6 +77,8 @@ class Bucket(object):
         self.param_flat = None    # fused-optimizer flat parameter buffer
         self.sgd_momentum = None  # fused SGD momentum (FusedSGD)
         self.adam_state = None    # fused Adam moments (FusedAdam/W)
+        self.used = False         # any gradient produced this cycle
+        self.prev_cycle_valid = False  # prev holds this cycle's last mb
 
     def ensure_prev(self):
         if self.prev is None:
@@ -110,6 +117,15 @@ class GradSyncEngine(object):
         cap_mb = bucket_cap_mb or float(os.getenv("ADAPTDL_BUCKET_CAP_MB",
                                                   _DEFAULT_BUCKET_CAP_MB))
         cap_bytes = int(cap_mb * 1024 * 1024)
+        # Opt-in: skip the all-reduce (and statistics launches) of
+        # buckets in which NO gradient was produced this optimizer
+        # cycle.  Only valid when parameter usage is IDENTICAL across
+        # replicas (e.g. multi-ADP setups alternating between models) -
+        # with asymmetric usage the per-communicator collective order
+        # would diverge between ranks.  Default keeps the
+        # always-all-reduce correct-by-construction behavior.
+        self._skip_unused = \
+            os.getenv("ADAPTDL_SKIP_UNUSED_BUCKETS") == "1"
         self.require_sync = True
         self.accum_count = 0       # completed (un-synced) microbatches
         # hipGraph-capture mode (set by torch.graph_step.GraphedStepper):
@@ -189,6 +205,7 @@ class GradSyncEngine(object):
             self._callback_queued = True
         bucket = self._param_to_bucket[param]
         bucket.ready += 1
+        bucket.used = True
         if bucket.ready == len(bucket.segments):
             self._flush_bucket(bucket)
 
@@ -211,7 +228,8 @@ class GradSyncEngine(object):
             # accumulating gradients in the buckets.
             self.accum_count += 1
             for bucket in self.buckets:
-                self._owner._accum_stat(bucket)
+                if not (self._skip_unused and not bucket.used):
+                    self._owner._accum_stat(bucket)
                 bucket.ready = 0
             self._owner._on_accum_done()
             return
@@ -219,7 +237,8 @@ class GradSyncEngine(object):
         # accum_count is incremented only after: _flush_bucket's local
         # statistics read it as "microbatches completed BEFORE this one".
         for bucket in self.buckets:
-            self._flush_bucket(bucket)
+            if not (self._skip_unused and not bucket.used):
+                self._flush_bucket(bucket)
             bucket.ready = 0
         self.accum_count += 1
         if self._world > 1:
@@ -243,6 +262,8 @@ class GradSyncEngine(object):
             if bucket.work is not None:
                 bucket.work.wait()
                 bucket.work = None
+            if self._skip_unused and not bucket.used:
+                continue  # all-zero bucket: contributes nothing
             # Scale to the mean gradient and accumulate the total
             # sum-of-squares statistic (owner may precondition it).
             self._owner._total_stat(bucket, scale)
@@ -293,6 +314,8 @@ class GradSyncEngine(object):
             bucket.flat.zero_()
             bucket.ready = 0
             bucket.work = None
+            bucket.used = False
+            bucket.prev_cycle_valid = False
         self.stats.zero_()
         self.accum_count = 0
 

@@ -153,7 +153,11 @@ class GradientNoiseScale(object):
         if self._engine.world_size == 1 and accum_so_far == 0:
             return  # single-sample step: differenced estimator used instead
         out = self._engine.stats[0, bucket.group_idx]
-        if accum_so_far == 0:
+        if accum_so_far == 0 or not bucket.prev_cycle_valid:
+            # Whole-cycle content IS this microbatch's gradient (either a
+            # no-accumulation step, or the bucket's first gradient of the
+            # cycle arrived at the sync microbatch - possible with
+            # skip-unused-buckets and conditionally-used parameters).
             self._bucket_sqsum(bucket, bucket.flat, out)
         else:
             self._bucket_sqsum_diff(bucket, out)
@@ -162,11 +166,14 @@ class GradientNoiseScale(object):
     def _accum_stat(self, bucket):
         """Local sum-of-squares of a non-final (accumulation) microbatch."""
         out = self._engine.stats[0, bucket.group_idx]
-        if self._engine.accum_count == 1:
-            # First microbatch: bucket content IS the microbatch gradient
-            # (prev may hold stale data from a previous regime).
+        if not bucket.prev_cycle_valid:
+            # Bucket's first gradient-carrying microbatch of this cycle
+            # (== accum_count 1 unless skip-unused deferred it): content
+            # IS the microbatch gradient (prev may hold stale data from
+            # a previous regime).
             self._bucket_sqsum(bucket, bucket.flat, out)
             bucket.ensure_prev().copy_(bucket.flat)
+            bucket.prev_cycle_valid = True
         else:
             self._bucket_sqsum_diff(bucket, out)
 

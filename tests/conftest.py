@@ -55,7 +55,13 @@ def elastic_multiprocessing(func):
                             ._REGISTRY.clear()
                     if "adaptdl_amd.torch._metrics" in _sys.modules:
                         _sys.modules["adaptdl_amd.torch._metrics"] \
-                            ._METRICS_STATE = None
+                            ._reset_for_tests()
+                    if "adaptdl_amd.collective" in _sys.modules:
+                        # A main-process test may have initialized the
+                        # coordinator; the fork must start fresh (its
+                        # socket fds are the parent's).
+                        _sys.modules["adaptdl_amd.collective"] \
+                            ._COORD = None
                     if "adaptdl_amd.torch.epoch" in _sys.modules:
                         _sys.modules["adaptdl_amd.torch.epoch"] \
                             ._EPOCH_STATE = None

@@ -222,6 +222,7 @@ def test_adam_precond_stats_one_launch_per_bucket(tmp_ckpt_env):
     # statistics launches take the one-per-bucket fast path.
     assert len(dev_calls) > 0
     assert not seg_calls, "per-segment fallback used with FusedAdam"
+    collective.teardown()
 
 
 def test_bf16_parameter_model_trains(tmp_ckpt_env):
@@ -260,3 +261,4 @@ def test_bf16_parameter_model_trains(tmp_ckpt_env):
     optim2 = adl.FusedSGD(model2.parameters(), lr=0.1)
     with pytest.raises(ValueError, match="float32"):
         adl.AdaptiveDataParallel(model2, optim2, name="bf16-fused")
+    collective.teardown()

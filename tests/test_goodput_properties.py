@@ -96,3 +96,80 @@ def test_optimize_beats_grid(params, grad):
                 continue
             alt = fn.evaluate(1, 4, bsz, acc)
             assert best >= alt * 0.9
+
+
+# ---- fit_perf_params optimistic-freeze properties:
+# the degenerate single-config profiles are exactly where a bad
+# extrapolation would steer the 8-GPU replica choice, so pin the
+# freeze semantics down.
+
+from adaptdl_amd.goodput import fit_perf_params  # noqa: E402
+
+
+def _times(params, nodes, replicas, bsz):
+    """Synthesize observed (accum, optim) step times from a true model."""
+    fn = _goodput_fn(params, (1e-3, 1e-3))
+    accum = params[0] + params[1] * bsz
+    total = bsz * replicas / fn.throughput(nodes, replicas, bsz, 0)
+    return accum, total
+
+
+@settings(max_examples=25, deadline=None)
+@given(alpha=st.floats(1e-3, 0.2), beta=st.floats(1e-5, 1e-3),
+       bsz=st.integers(32, 1024))
+def test_fit_single_bsz_freezes_optimistically(alpha, beta, bsz):
+    """One observed atomic size at 1x1: alpha_c pins to half the mean
+    accum time, so the model stays optimistic about raising the batch
+    size (per-sample time falls) instead of extrapolating wildly."""
+    t = alpha + beta * bsz
+    fitted = fit_perf_params([1, 1], [1, 1], [bsz, bsz],
+                             [t, t], [t * 1.05, t * 1.05])
+    assert np.isclose(fitted.alpha_c, t / 2, rtol=1e-6)
+    # Optimism: doubling the batch must cost < 2x the step time.
+    assert fitted.alpha_c + fitted.beta_c * 2 * bsz < 2 * t + 1e-9
+    assert np.all(np.isfinite(np.asarray(fitted)))
+
+
+@settings(max_examples=25, deadline=None)
+@given(alpha=st.floats(1e-3, 0.2), beta=st.floats(1e-5, 1e-3))
+def test_fit_no_multireplica_data_predicts_scaleup_speedup(alpha, beta):
+    """Profiles from a single replica only: the network terms freeze at
+    ~zero, so the fitted model must predict a real speedup from adding
+    replicas (the optimistic direction that makes the allocator try)."""
+    bszs = np.array([64, 128, 256, 512])
+    accum = alpha + beta * bszs
+    optim = accum * 1.1
+    fitted = fit_perf_params(np.ones(4), np.ones(4), bszs, accum, optim)
+    fn = GoodputFunction(fitted, GradParams(1e-3, 1e-3), 128)
+    t1 = fn.throughput(1, 1, 128, 0)
+    t2 = fn.throughput(1, 2, 128, 0)
+    assert t2 > 1.5 * t1
+    # Inter-node prior: at least 10% worse than intra-node.
+    assert fitted.alpha_n >= fitted.alpha_r * 1.1 - 1e-12
+    assert fitted.beta_n >= fitted.beta_r * 1.1 - 1e-12
+
+
+def test_fit_no_retrogression_data_freezes_beta():
+    """No profile with replicas > 2: the per-replica retrogression
+    slopes stay pinned (no invented penalty for scaling out)."""
+    nodes = np.array([1, 1, 1])
+    replicas = np.array([1, 2, 2])
+    bszs = np.array([128, 128, 256])
+    accum = 0.05 + 1e-4 * bszs
+    optim = accum + 0.01
+    fitted = fit_perf_params(nodes, replicas, bszs, accum, optim)
+    assert fitted.beta_n <= 2e-8 * 1.1 + 1e-12
+    assert fitted.beta_r <= 1e-8 + 1e-12
+    assert np.all(np.isfinite(np.asarray(fitted)))
+
+
+def test_fit_degenerate_identical_rows():
+    """All-identical observations must not produce NaNs or a model that
+    predicts slower-with-more-replicas at the observed point."""
+    n = 4
+    fitted = fit_perf_params(np.ones(n), np.ones(n),
+                             np.full(n, 128), np.full(n, 0.05),
+                             np.full(n, 0.06))
+    assert np.all(np.isfinite(np.asarray(fitted)))
+    fn = GoodputFunction(fitted, GradParams(1e-3, 1e-3), 128)
+    assert fn.throughput(1, 2, 128, 0) >= fn.throughput(1, 1, 128, 0)
