@@ -441,19 +441,22 @@ extern "C" __global__ __launch_bounds__(256) void k_wrw_reduce(
 // zero-fill tail-chunk lines, trading <= 24% staged-pixel waste for the
 // same aligned-b128 LDS tiling.
 extern "C" int conv3x3_wrw_params(int H, int W, int* P, int* Wp) {
-    // W=14/7 A/B (r2 pass d): the exact-cover small chunks (P=2/P=4,
-    // CP=32) lost ~20% to MIOpen - mostly barrier rounds per pixel
-    // (kchunks=1).  ADAPTDL_WRW_BIGCHUNK=1 trades utilization (87->77%)
-    // for 4x larger chunks; keep whichever wins the next timing pass.
-    const bool big = getenv("ADAPTDL_WRW_BIGCHUNK") != nullptr;
+    // W=14/7 chunking A/B (r2 pass e, N256 C=K=256/512): the
+    // exact-cover small chunks (P=2/P=4, CP=32, 87% pixel utilization)
+    // ran 163/183 us - barrier-round-bound at kchunks=1 - while P=8
+    // (CP=64/128, 77% utilization via zero-filled tail lines) runs
+    // 120/127 us and beats MIOpen's 135/143 us.  Large chunks are the
+    // default; ADAPTDL_WRW_SMALLCHUNK restores the exact-cover variant
+    // for re-measurement.
+    const bool small = getenv("ADAPTDL_WRW_SMALLCHUNK") != nullptr;
     switch (W) {
     case 8:  *Wp = 8;  *P = 8; break;
     case 16: *Wp = 16; *P = 8; break;
     case 32: *Wp = 32; *P = 4; break;
     case 56: *Wp = 64; *P = 1; break;
     case 28: *Wp = 32; *P = 4; break;
-    case 14: *Wp = 16; *P = big ? 8 : 2; break;
-    case 7:  *Wp = 8;  *P = big ? 8 : 4; break;
+    case 14: *Wp = 16; *P = small ? 2 : 8; break;
+    case 7:  *Wp = 8;  *P = small ? 4 : 8; break;
     default: return 0;
     }
     if (H < 2) return 0;

@@ -152,7 +152,7 @@ def test_w8b_wave_mapping():
 def _wrw_params(W):
     """Mirror of conv3x3_wrw_params (conv_kernels.hip)."""
     table = {8: (8, 8), 16: (8, 16), 32: (4, 32),
-             56: (1, 64), 28: (4, 32), 14: (2, 16), 7: (4, 8)}
+             56: (1, 64), 28: (4, 32), 14: (8, 16), 7: (8, 8)}
     P, Wp = table[W]
     return P, Wp
 
