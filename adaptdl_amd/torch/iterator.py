@@ -34,13 +34,20 @@ class AdaptiveBPTTIterator(AdaptiveDataLoaderMixin):
         batch_first (bool): yield (batch, seq) instead of (seq, batch).
         pad_token (int): id used to pad the corpus to a multiple of the
             batch size.
-        max_batch_size / local_bsz_bounds: enable autoscaling (same
-            contract as AdaptiveDataLoader.autoscale_batch_size).
+        max_batch_size / local_bsz_bounds / gradient_accumulation:
+            enable autoscaling (same contract as
+            AdaptiveDataLoader.autoscale_batch_size).  Accumulation is
+            an extension over the reference iterator (iterator.py:46,
+            which never passes it): without it a single replica can
+            never scale its batch (the non-accumulation planner pins
+            atomic_bsz to the initial size at num_replicas == 1), so
+            BASELINE config 3's "adaptive batch + grad accumulation"
+            only engaged at high replica counts.
     """
 
     def __init__(self, data, batch_size, bptt_len, batch_first=False,
                  pad_token=0, max_batch_size=None, local_bsz_bounds=None,
-                 device=None):
+                 gradient_accumulation=False, device=None):
         AdaptiveDataLoaderMixin.__init__(self, batch_size)
         if data.dim() != 1:
             raise ValueError("data must be a flat 1-D token tensor")
@@ -50,8 +57,9 @@ class AdaptiveBPTTIterator(AdaptiveDataLoaderMixin):
         self.pad_token = pad_token
         self.device = device
         if max_batch_size:
-            self._elastic.autoscale_batch_size(max_batch_size,
-                                               local_bsz_bounds)
+            self._elastic.autoscale_batch_size(
+                max_batch_size, local_bsz_bounds,
+                gradient_accumulation=gradient_accumulation)
 
     @staticmethod
     def _recompute_start(prev_curr, prev_end, curr_end):

@@ -214,7 +214,8 @@ class TransformerWT2(Workload):
         loader = adl.AdaptiveBPTTIterator(
             corpus, batch_size=args.init_batch, bptt_len=self.bptt,
             max_batch_size=args.max_batch,
-            local_bsz_bounds=args.bounds, device=self.device)
+            local_bsz_bounds=args.bounds,
+            gradient_accumulation=True, device=self.device)
         self.loader = loader
         return self.adp, self.optim, loader
 

@@ -67,3 +67,21 @@ def _run_bptt_restart():
 
 def test_bptt_restart_shapes():
     _run_bptt_restart()
+
+
+def test_bptt_gradient_accumulation_flag(tmp_ckpt_env):
+    """BPTT autoscaling can enable accumulation (extension over the
+    reference iterator, which never passes the flag and therefore can
+    never scale a single replica's batch)."""
+    import adaptdl_amd.torch as adl
+    import torch as _torch
+    import adaptdl_amd.collective as _collective
+    if not _collective.initialized():
+        _collective.initialize(master_addr="127.0.0.1")
+    it = adl.AdaptiveBPTTIterator(_torch.arange(1000), batch_size=4,
+                                  bptt_len=10, max_batch_size=64,
+                                  local_bsz_bounds=(2, 8),
+                                  gradient_accumulation=True)
+    assert it._elastic._gradient_accumulation is True
+    assert it._elastic.max_batch_size == 64
+    _collective.teardown()
