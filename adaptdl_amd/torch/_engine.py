@@ -40,6 +40,15 @@ LOG = logging.getLogger(__name__)
 
 _DEFAULT_BUCKET_CAP_MB = 32
 
+# Monotonic optimizer-cycle counter (bumped by every engine's
+# zero_grad): layers cache per-cycle derived tensors (e.g. the bf16
+# weight cast) against it, since weights only change between cycles.
+_cycle_serial = 0
+
+
+def cycle_serial():
+    return _cycle_serial
+
 
 def _segment_view(seg, p):
     """View a flat bucket segment with the same memory layout as ``p``.
@@ -307,6 +316,8 @@ class GradSyncEngine(object):
 
     def zero_grad(self):
         """Zero all gradient buckets (and statistics) for the next step."""
+        global _cycle_serial
+        _cycle_serial += 1
         # Note: bucket.prev is NOT zeroed — it must persist across steps for
         # the differenced single-sample GNS estimator; the accumulation path
         # re-initializes it on the first microbatch instead.

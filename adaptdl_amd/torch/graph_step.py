@@ -288,6 +288,12 @@ class GraphedStepper(object):
     def _run_eager(self, *tensors):
         self._engine.graph_mode = False
         self.stats["eager"] += 1
+        # Replays run no Python, so cycle-keyed layer caches (e.g. the
+        # bf16 weight cast) were last refreshed at capture time; bump
+        # the serial so this eager microbatch re-derives them from the
+        # CURRENT weights instead of a stale snapshot.
+        from adaptdl_amd.torch import _engine as _engine_mod
+        _engine_mod._cycle_serial += 1
         return self._fwd_bwd(*tensors)
 
     def _reset(self, sig, accum):

@@ -134,10 +134,8 @@ class _FusedConvFunction(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x, weight, stride, padding, dilation, groups):
+    def forward(ctx, x, weight, wb, stride, padding, dilation, groups):
         ext = ops._load_extension()
-        wb = weight.detach().to(torch.bfloat16) \
-            .contiguous(memory_format=torch.channels_last)
         n, c, h, w = x.shape
         k = wb.shape[0]
         # The custom fwd/bwd-data kernel is numerics-verified and beats
@@ -188,7 +186,7 @@ class _FusedConvFunction(torch.autograd.Function):
                          device=x.device) \
             .contiguous(memory_format=torch.channels_last)
         ext.conv_wrw(x, dy, ws, dw)
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
 class _S2ConvFunction(torch.autograd.Function):
@@ -347,6 +345,28 @@ class FusedConv2d(nn.Conv2d):
                 x.shape[2] % 2 == 0 and x.shape[3] % 2 == 0 and
                 x.is_contiguous(memory_format=torch.channels_last))
 
+    def _cast_weight(self):
+        """bf16 channels_last weight, cached across the accumulation
+        microbatches of one optimizer cycle.
+
+        autocast re-casts module weights on EVERY forward (its cast
+        cache lives only inside one autocast region), which measured
+        ~4% of flagship kernel time as small elementwise launches.
+        Weights only change between optimizer cycles, so the cast is
+        keyed on (engine cycle serial, weight._version) — the serial
+        covers the fused optimizers (raw-kernel updates do not bump
+        _version), _version covers stock optimizers without an engine.
+        """
+        from adaptdl_amd.torch import _engine
+        key = (_engine.cycle_serial(), self.weight._version)
+        cached = getattr(self, "_wb_cache", None)
+        if cached is not None and cached[1] == key:
+            return cached[0]
+        wb = self.weight.detach().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        self._wb_cache = (wb, key)
+        return wb
+
     def forward(self, x):
         if torch.is_autocast_enabled() and x.is_cuda and \
                 x.dtype != torch.bfloat16 and \
@@ -354,9 +374,10 @@ class FusedConv2d(nn.Conv2d):
             x = x.to(torch.bfloat16)
         if self._wrw_ok(x) and (self.weight.requires_grad or
                                 x.requires_grad):
-            return _FusedConvFunction.apply(x, self.weight, self.stride,
-                                            self.padding, self.dilation,
-                                            self.groups)
+            return _FusedConvFunction.apply(x, self.weight,
+                                            self._cast_weight(),
+                                            self.stride, self.padding,
+                                            self.dilation, self.groups)
         if x.requires_grad and self._s2_ok(x):
             return _S2ConvFunction.apply(x, self.weight, self.stride,
                                          self.padding, self.dilation,
