@@ -360,7 +360,8 @@ class FusedConv2d(nn.Conv2d):
         from adaptdl_amd.torch import _engine
         key = (_engine.cycle_serial(), self.weight._version)
         cached = getattr(self, "_wb_cache", None)
-        if cached is not None and cached[1] == key:
+        if cached is not None and cached[1] == key \
+                and os.getenv("ADAPTDL_NO_WB_CACHE") != "1":
             return cached[0]
         wb = self.weight.detach().to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)

@@ -138,3 +138,28 @@ def _run_full_adp_step():
 def test_channels_last_full_adp_step():
     """Two SGD steps via the full ADP stack: NCHW vs channels_last equal."""
     _run_full_adp_step()
+
+
+def test_weight_cast_cache_invalidation():
+    """FusedConv2d caches its bf16 weight cast per optimizer cycle;
+    the cache must refresh on a cycle-serial bump (fused optimizers)
+    and on an in-place weight mutation (stock optimizers)."""
+    from adaptdl_amd.torch import _engine
+    from adaptdl_amd.torch.layers import FusedConv2d
+
+    conv = FusedConv2d(8, 8, 3, padding=1, bias=False)
+    wb1 = conv._cast_weight()
+    assert conv._cast_weight() is wb1  # same cycle: cached
+
+    _engine._cycle_serial += 1         # new optimizer cycle
+    wb2 = conv._cast_weight()
+    assert wb2 is not wb1
+
+    with torch.no_grad():
+        conv.weight.add_(1.0)          # stock-optimizer-style update
+    wb3 = conv._cast_weight()
+    assert wb3 is not wb2
+    assert torch.allclose(wb3.float(),
+                          conv.weight.detach().to(torch.bfloat16)
+                          .float().contiguous(
+                              memory_format=torch.channels_last))
