@@ -14,6 +14,7 @@ LOG = logging.getLogger(__name__)
 
 _EXIT_FLAG = False
 _INSTALLED = False
+_RESCALE_SEEN = False  # SIGUSR2: an in-place rescale directive exists
 
 
 def _handler(signum, frame):
@@ -25,13 +26,47 @@ def _handler(signum, frame):
     _EXIT_FLAG = True
 
 
+def _usr2_handler(signum, frame):
+    global _RESCALE_SEEN
+    _RESCALE_SEEN = True
+
+
 def install_signal_handlers():
     global _INSTALLED
     if _INSTALLED or threading.current_thread() is not threading.main_thread():
         return
     signal.signal(signal.SIGTERM, _handler)
     signal.signal(signal.SIGINT, _handler)
+    signal.signal(signal.SIGUSR2, _usr2_handler)
     _INSTALLED = True
+
+
+def get_rescale_request():
+    """(version, directive) of a pending in-place rescale, or (0, None).
+
+    The directive file is written by the controller before SIGUSR2, so
+    reading it after the flag is race-free.  Versions already applied
+    in this process (tracked by torch._rejoin) read as not-pending.
+    """
+    if not _RESCALE_SEEN:
+        return 0, None
+    import json
+    import os
+    from adaptdl_amd import env
+    from adaptdl_amd.torch import _rejoin
+    root = env.checkpoint_path()
+    if root is None:
+        return 0, None
+    path = os.path.join(root, "rescale-inplace.json")
+    try:
+        with open(path) as f:
+            directive = json.load(f)
+    except (OSError, ValueError):
+        return 0, None
+    version = int(directive.get("version", 0))
+    if version <= _rejoin.applied_version():
+        return 0, None
+    return version, directive
 
 
 def get_exit_flag():

@@ -34,9 +34,16 @@ from adaptdl_amd.torch.epoch import current_epoch
 from adaptdl_amd.torch._metrics import (
     profile_step_start, profile_step_commit,
     set_batch_size, get_goodput_fn, get_progress)
-from adaptdl_amd._signal import get_exit_flag
+from adaptdl_amd._signal import get_exit_flag, get_rescale_request
+from adaptdl_amd.torch import _rejoin
 
 LOG = logging.getLogger(__name__)
+
+
+def _control_reduce(a, b):
+    """Elementwise reduce of the per-iteration control payload:
+    (exit requested anywhere, rescale version seen by EVERYONE)."""
+    return (a[0] or b[0], min(a[1], b[1]))
 
 
 class ElasticSampler(Sampler):
@@ -101,6 +108,10 @@ class AdaptiveDataLoaderHelper(object):
     _position = collections.Counter()
     _training = None
     _current = None
+
+    #: Set by loaders whose __iter__ can re-partition mid-pass (the
+    #: in-place rescale raise is only safe there).
+    _supports_inplace = False
 
     def __init__(self, batch_size=1):
         self.batch_size = batch_size
@@ -249,11 +260,24 @@ class AdaptiveDataLoaderHelper(object):
         rescale/preemption signal was agreed on by all replicas, saves a
         checkpoint and exits with code 143.
         """
-        if self.future_exit is not None and self.future_exit.result():
-            adaptdl_amd.checkpoint.save_all_states()
-            exit(143)
+        if self.future_exit is not None:
+            do_exit, rescale_ver = self.future_exit.result()
+            self.future_exit = None
+            if do_exit:
+                adaptdl_amd.checkpoint.save_all_states()
+                exit(143)
+            if rescale_ver > _rejoin.applied_version() \
+                    and self._supports_inplace and self.training \
+                    and self._accum_count == 0:
+                # Every replica agreed on the directive and is at an
+                # optimizer-cycle boundary: hand control to __iter__,
+                # which performs the in-place rejoin and re-partitions
+                # the current pass (leavers never return from it).
+                _, directive = get_rescale_request()
+                if directive is not None:
+                    raise _rejoin.InplaceRescale(directive)
         self.future_exit = adaptdl_amd.collective.allreduce_async(
-            get_exit_flag(), lambda a, b: a or b)
+            (get_exit_flag(), get_rescale_request()[0]), _control_reduce)
         profile_step_start(self.current_local_bsz)
         yield
         if commit:
@@ -380,26 +404,43 @@ class AdaptiveDataLoader(DataLoader, AdaptiveDataLoaderMixin):
         """
         epoch = current_epoch()
         adaptive = self._elastic.max_batch_size is not None
-        replicas = adaptdl_amd.env.num_replicas()
+        self._elastic._supports_inplace = True
         with self._elastic.context():
             if self._elastic.skipdone():
                 return
             while True:
+                # Re-read per pass: an in-place rescale changes the
+                # replica count without restarting this generator.
+                replicas = adaptdl_amd.env.num_replicas()
+                self.sampler.num_replicas = replicas
+                self.sampler.rank = adaptdl_amd.env.replica_rank()
                 self.sampler.set_epoch(
                     epoch, index=self._elastic.current_index)
                 atomic_bsz = self._elastic._sync_local_bsz()
                 self.batch_sampler.batch_size = atomic_bsz
                 stop = False
-                for idx, batch in enumerate(super().__iter__()):
-                    # Skip profiling the first batch of each pass (loader
-                    # worker startup would pollute the perf model).
-                    with self._elastic.profile(self.training and idx >= 1):
-                        yield batch
-                        self._elastic.current_index += \
-                            replicas * self.batch_sampler.batch_size
-                        if adaptive and self._epoch_target_reached(epoch):
-                            stop = True
-                            break
+                try:
+                    inner = enumerate(super().__iter__())
+                    for idx, batch in inner:
+                        # Skip profiling the first batch of each pass
+                        # (loader worker startup would pollute the perf
+                        # model).
+                        with self._elastic.profile(self.training
+                                                   and idx >= 1):
+                            yield batch
+                            self._elastic.current_index += \
+                                replicas * self.batch_sampler.batch_size
+                            if adaptive and \
+                                    self._epoch_target_reached(epoch):
+                                stop = True
+                                break
+                except _rejoin.InplaceRescale as req:
+                    # Survivors return with the world resized; leavers
+                    # exit(143) inside.  The current pass then restarts
+                    # from the exact global sample index with the new
+                    # partitioning (same math as a restart resume).
+                    _rejoin.perform(req.directive)
+                    continue
                 # Round current_index up to a multiple of the dataset size
                 # (ends the data pass even if it stopped mid-way).
                 self._elastic.current_index += \
