@@ -41,6 +41,9 @@ try:  # optional Prometheus metrics (reference controller.py:35-41)
         "failed": Counter("adaptdl_jobs_failed", "Jobs failed"),
         "preemptions": Counter("adaptdl_job_preemptions",
                                "Graceful checkpoint-restart preemptions"),
+        "inplace_rescales": Counter(
+            "adaptdl_job_inplace_rescales",
+            "In-place (no-restart) scale-downs started"),
         "replicas": Gauge("adaptdl_running_replicas",
                           "Currently running replica processes"),
     }
@@ -95,7 +98,8 @@ class JobSpec(object):
 
     def __init__(self, argv, name, job_dir, min_replicas=0, max_replicas=8,
                  gpus_per_replica=1, env=None, workdir=None,
-                 restart_limit=3, preemptible=True):
+                 restart_limit=3, preemptible=True,
+                 inplace_scaledown=False):
         # Admission validation (reference validator.py:70-101 enforces
         # these via a k8s webhook; locally they are constructor checks).
         if not argv:
@@ -115,6 +119,12 @@ class JobSpec(object):
         self.workdir = workdir
         self.restart_limit = restart_limit
         self.preemptible = preemptible
+        # Scale-downs rejoin in place (SIGUSR2 directive; survivors keep
+        # all state in memory, leavers exit) instead of checkpoint-
+        # restarting the whole group.  Requires the worker to train via
+        # AdaptiveDataLoader (BPTT/eval-only phases fall back to the
+        # restart path via the controller's escalation timeout).
+        self.inplace_scaledown = inplace_scaledown
         self._frozen = False
 
     def freeze(self):
@@ -151,6 +161,8 @@ class _Job(object):
         self.failures = 0
         self.completion = None    # set when SUCCEEDED/FAILED
         self.warm_dir = None      # RAM-backed rescale checkpoint root
+        self.inplace = None       # pending in-place scale-down state
+        self.inplace_version = 0
 
     @property
     def num_replicas(self):
@@ -336,11 +348,18 @@ class LocalController(object):
                     job.target_allocation = None
                     self._start_group(job)
             elif job.state == RUNNING:
-                if job.target_allocation is not None and \
+                if job.inplace is not None:
+                    self._check_inplace(job)
+                elif job.target_allocation is not None and \
                         sorted(job.target_allocation) != \
                         sorted(job.allocation):
-                    job.state = STOPPING
-                    self._signal_group(job, signal.SIGTERM)
+                    n_target = len(job.target_allocation)
+                    if job.spec.inplace_scaledown and \
+                            0 < n_target < len(job.allocation):
+                        self._start_inplace(job, n_target)
+                    else:
+                        job.state = STOPPING
+                        self._signal_group(job, signal.SIGTERM)
                 else:
                     self._check_group(job)
             elif job.state == STOPPING:
@@ -423,6 +442,73 @@ class LocalController(object):
         job.state = RUNNING
         LOG.info("job %s group %d started with %d replicas (gpus=%s)",
                  spec.name, job.num_restarts, n, gpus)
+
+    # ---- in-place scale-down (north star: in-memory rejoin) ----------
+
+    INPLACE_TIMEOUT = 60.0
+
+    def _start_inplace(self, job, world):
+        """Write the rescale directive and signal the group; survivors
+        rejoin at the next optimizer-cycle boundary, leavers exit(143).
+        State never leaves the survivors' memory."""
+        import json
+        job.inplace_version += 1
+        directive = {"version": job.inplace_version, "world": world,
+                     "master_port": _free_port()}
+        path = os.path.join(job.spec.job_dir, "rescale-inplace.json")
+        tmp = path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(directive, f)
+        os.replace(tmp, path)
+        self._signal_group(job, signal.SIGUSR2)
+        job.inplace = {"world": world, "since": time.time()}
+        _metric("inplace_rescales")
+        LOG.info("job %s: in-place scale-down %d -> %d (directive v%d)",
+                 job.spec.name, len(job.allocation), world,
+                 job.inplace_version)
+
+    def _check_inplace(self, job):
+        info = job.inplace
+        world = info["world"]
+        codes = [p.poll() for p in job.procs]
+        survivors, leavers = codes[:world], codes[world:]
+        if any(c is not None for c in survivors):
+            # A survivor died mid-rescale: abandon the in-place path and
+            # let the normal crash/preemption machinery take over.
+            LOG.warning("job %s: survivor exited during in-place rescale "
+                        "(codes=%s); falling back", job.spec.name, codes)
+            job.inplace = None
+            self._check_group(job)
+            return
+        if all(c in (0, GRACEFUL_EXIT) for c in leavers):
+            # Leavers gone, survivors running: adopt the new size with
+            # NO restart (num_restarts unchanged, same processes).
+            for p in job.procs[world:]:
+                try:
+                    p._adaptdl_log.close()
+                except Exception:  # noqa: BLE001
+                    pass
+            if METRICS is not None:
+                METRICS["replicas"].dec(len(job.procs) - world)
+            job.procs = job.procs[:world]
+            job.gpus = job.gpus[:world * job.spec.gpus_per_replica]
+            job.allocation = ["local"] * world
+            job.target_allocation = None
+            job.inplace = None
+            self.supervisor.set_endpoints(job.spec.name, job.num_restarts,
+                                          ["127.0.0.1"] * world)
+            LOG.info("job %s: in-place scale-down complete (%d replicas, "
+                     "restarts still %d)", job.spec.name, world,
+                     job.num_restarts)
+            return
+        if time.time() - info["since"] > self.INPLACE_TIMEOUT:
+            # Workers never reached a safe point (BPTT loop, eval
+            # phase, hung): escalate to the checkpoint-restart path.
+            LOG.warning("job %s: in-place rescale timed out; escalating "
+                        "to SIGTERM restart", job.spec.name)
+            job.inplace = None
+            job.state = STOPPING
+            self._signal_group(job, signal.SIGTERM)
 
     def _signal_group(self, job, sig):
         for p in job.procs:

@@ -67,6 +67,19 @@ class GradientNoiseScale(object):
         self._engine = GradSyncEngine(
             [g["params"] for g in optimizer.param_groups], owner=self,
             process_group=process_group)
+        from adaptdl_amd.torch import _rejoin
+        _rejoin.register_gns(self)
+
+    def _rebind_world(self, world):
+        """Adopt a new replica count after an in-place rescale (the
+        torch process group was re-created by torch._rejoin.perform;
+        gradient/optimizer state is untouched — it is replicated)."""
+        self._num_replicas = world
+        self._engine._world = world
+        self._engine._pg = None  # the fresh default group
+        # accum_scale is re-derived from the dataloader on the next
+        # forward (ADP.forward -> set_accum_scale), which also resets
+        # the accumulation cycle if it changed.
 
     # ---- state accessors (same API as the reference) ---------------------
 

@@ -122,7 +122,12 @@ def main():
                            interval=3600)
     spec = JobSpec([sys.executable, script], name="elastic-resnet50",
                    job_dir=job_dir, min_replicas=1, max_replicas=peak,
-                   gpus_per_replica=gpr)
+                   gpus_per_replica=gpr,
+                   # Scale-downs (the N -> N/2 phase) rejoin in place:
+                   # survivors keep model/optimizer state in HBM, no
+                   # checkpoint-restart (scale-ups still respawn via the
+                   # warm RAM checkpoint).
+                   inplace_scaledown=True)
     ctrl.submit(spec)
     try:
         for target in phases:

@@ -309,3 +309,64 @@ def test_external_sigterm_preemption(tmp_path, controller):
     assert max(t["restarts"] for t in trace) >= 1
     final = json.load(open(os.path.join(job_dir, "final.json")))
     assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
+
+
+def test_inplace_scaledown_no_restart(tmp_path, controller):
+    """In-place (no-restart) scale-down: 2 -> 1 replicas via the SIGUSR2
+    directive.  The surviving process must keep training WITHOUT a
+    checkpoint-restart (num_restarts stays 0, ADAPTDL_NUM_RESTARTS never
+    changes, every epoch runs exactly once) — model/optimizer state
+    never leaves the survivor's memory (north star: in-HBM rejoin)."""
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="inplace-job",
+                   job_dir=job_dir, min_replicas=3, max_replicas=3,
+                   gpus_per_replica=0, inplace_scaledown=True)
+    controller.submit(spec)
+
+    deadline = time.time() + 90
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("inplace-job")
+        time.sleep(0.1)
+    with controller._lock:
+        survivor_pid = controller._jobs["inplace-job"].procs[0].pid
+
+    # 3 -> 2: the two survivors must re-rendezvous with each other.
+    controller.rescale("inplace-job", 2)
+    while controller.status("inplace-job")["replicas"] != 2:
+        assert time.time() < deadline, controller.status("inplace-job")
+        assert controller.status("inplace-job")["restarts"] == 0
+        time.sleep(0.1)
+    # let it train a bit at 2 before shrinking again
+    n_lines = len(open(trace_path).readlines())
+    while len(open(trace_path).readlines()) < n_lines + 2:
+        assert time.time() < deadline
+        time.sleep(0.1)
+
+    # 2 -> 1: down to a solo survivor.
+    controller.rescale("inplace-job", 1)
+    while controller.status("inplace-job")["replicas"] != 1:
+        assert time.time() < deadline, controller.status("inplace-job")
+        assert controller.status("inplace-job")["restarts"] == 0
+        time.sleep(0.1)
+    with controller._lock:
+        assert controller._jobs["inplace-job"].procs[0].pid == survivor_pid
+
+    state = controller.wait("inplace-job", timeout=180)
+    assert state == "Succeeded", controller.status("inplace-job")
+    assert controller.status("inplace-job")["restarts"] == 0
+
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(30))
+    # every epoch ran exactly once: the in-place path replays nothing
+    assert len(trace) == 30
+    assert {t["replicas"] for t in trace} == {3, 2, 1}
+    assert all(t["restarts"] == 0 for t in trace)
+    # no checkpoint was written anywhere (neither disk nor warm root)
+    assert not any(n.startswith("checkpoint-")
+                   for n in os.listdir(job_dir))
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
