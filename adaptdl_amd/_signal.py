@@ -39,6 +39,19 @@ def install_signal_handlers():
     signal.signal(signal.SIGINT, _handler)
     signal.signal(signal.SIGUSR2, _usr2_handler)
     _INSTALLED = True
+    # Readiness marker: the controller only sends SIGUSR2 (whose default
+    # disposition TERMINATES a process) to workers that have installed
+    # the handler — a directive racing process startup must not kill the
+    # worker it is addressed to.
+    try:
+        import os
+        from adaptdl_amd import env
+        root = env.checkpoint_path()
+        if root is not None and os.path.isdir(root):
+            open(os.path.join(root, ".sigusr2-ready-{}".format(
+                env.replica_rank())), "w").close()
+    except OSError:
+        pass
 
 
 def get_rescale_request():

@@ -393,6 +393,13 @@ class LocalController(object):
         master_port = _free_port()
         if job.warm_dir is None:
             job.warm_dir = self._make_warm_dir(spec.name)
+        # Stale SIGUSR2-readiness markers from the previous group.
+        for f in os.listdir(spec.job_dir):
+            if f.startswith(".sigusr2-ready-"):
+                try:
+                    os.unlink(os.path.join(spec.job_dir, f))
+                except OSError:
+                    pass
         gpus = []
         if spec.gpus_per_replica > 0:
             busy = self._assigned_gpus()
@@ -460,16 +467,40 @@ class LocalController(object):
         with open(tmp, "w") as f:
             json.dump(directive, f)
         os.replace(tmp, path)
-        self._signal_group(job, signal.SIGUSR2)
-        job.inplace = {"world": world, "since": time.time()}
+        job.inplace = {"world": world, "since": time.time(),
+                       "last_signal": 0.0}
+        self._signal_inplace_ready(job)
         _metric("inplace_rescales")
         LOG.info("job %s: in-place scale-down %d -> %d (directive v%d)",
                  job.spec.name, len(job.allocation), world,
                  job.inplace_version)
 
+    def _signal_inplace_ready(self, job):
+        """SIGUSR2 only the workers that installed the handler (marker
+        file) — the default disposition of SIGUSR2 terminates, so a
+        directive racing worker startup must not kill the group.
+        Idempotent; re-sent ~1/s while the directive is pending."""
+        info = job.inplace
+        now = time.time()
+        if now - info["last_signal"] < 1.0:
+            return
+        info["last_signal"] = now
+        for rank, p in enumerate(job.procs):
+            if p.poll() is not None:
+                continue
+            marker = os.path.join(job.spec.job_dir,
+                                  ".sigusr2-ready-{}".format(rank))
+            if not os.path.exists(marker):
+                continue
+            try:
+                os.killpg(p.pid, signal.SIGUSR2)
+            except (ProcessLookupError, PermissionError):
+                pass
+
     def _check_inplace(self, job):
         info = job.inplace
         world = info["world"]
+        self._signal_inplace_ready(job)
         codes = [p.poll() for p in job.procs]
         survivors, leavers = codes[:world], codes[world:]
         if any(c is not None for c in survivors):

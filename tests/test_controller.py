@@ -370,3 +370,41 @@ def test_inplace_scaledown_no_restart(tmp_path, controller):
                    for n in os.listdir(job_dir))
     final = json.load(open(os.path.join(job_dir, "final.json")))
     assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
+
+
+def test_inplace_rescale_escalates_on_timeout(tmp_path, controller, monkeypatch):
+    """Workers that never reach a safe point (no AdaptiveDataLoader
+    iteration) cannot act on the in-place directive; the controller must
+    escalate to the SIGTERM checkpoint-restart path after its timeout."""
+    monkeypatch.setattr(LocalController, "INPLACE_TIMEOUT", 2.0)
+    script = tmp_path / "sleeper.py"
+    # Installs the SIGUSR2/SIGTERM handlers (via init_process_group)
+    # but never iterates an AdaptiveDataLoader, so the directive is
+    # seen but never acted on; on the escalation SIGTERM it exits 143.
+    script.write_text(
+        "import sys, time\n"
+        "sys.path.insert(0, {!r})\n"
+        "import adaptdl_amd.torch as adl\n"
+        "from adaptdl_amd._signal import get_exit_flag\n"
+        "adl.init_process_group('gloo')\n"
+        "while not get_exit_flag():\n"
+        "    time.sleep(0.05)\n"
+        "sys.exit(143)\n".format(REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="stuck-job",
+                   job_dir=job_dir, min_replicas=2, max_replicas=2,
+                   gpus_per_replica=0, inplace_scaledown=True)
+    controller.submit(spec)
+    deadline = time.time() + 60
+    while controller.status("stuck-job")["state"] != "Running":
+        assert time.time() < deadline, controller.status("stuck-job")
+        time.sleep(0.1)
+    controller.rescale("stuck-job", 1)
+    # Escalation: directive times out -> SIGTERM restart at 1 replica.
+    while not (controller.status("stuck-job")["replicas"] == 1
+               and controller.status("stuck-job")["state"] == "Running"):
+        assert time.time() < deadline, controller.status("stuck-job")
+        time.sleep(0.2)
+    assert controller.status("stuck-job")["restarts"] >= 1
+    controller.rescale("stuck-job", 0)  # cleanup is shutdown's job
