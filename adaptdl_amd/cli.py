@@ -105,7 +105,9 @@ class _AdminServer(object):
                                                       1),
                             env=body.get("env") or {},
                             workdir=body.get("workdir"),
-                            preemptible=body.get("preemptible", True))
+                            preemptible=body.get("preemptible", True),
+                            inplace_scaledown=body.get(
+                                "inplace_scaledown", False))
                         srv.controller.submit(spec)
                         self._reply(200, {"name": name,
                                           "job_dir": job_dir})
@@ -175,7 +177,8 @@ def cmd_run(args, command):
     spec = JobSpec(command, name=name, job_dir=job_dir,
                    min_replicas=args.min_replicas,
                    max_replicas=args.max_replicas,
-                   gpus_per_replica=args.gpus_per_replica)
+                   gpus_per_replica=args.gpus_per_replica,
+                   inplace_scaledown=args.inplace_scaledown)
     controller.submit(spec)
     print("job {} -> {}".format(name, job_dir))
 
@@ -212,6 +215,7 @@ def cmd_submit(args, command):
         "min_replicas": args.min_replicas,
         "max_replicas": args.max_replicas,
         "gpus_per_replica": args.gpus_per_replica,
+        "inplace_scaledown": args.inplace_scaledown,
         "job_dir": args.job_dir,
         "workdir": os.getcwd()})
     print("submitted {} (job_dir {})".format(out["name"],
@@ -246,6 +250,35 @@ def cmd_rescale(args):
 def cmd_stop(args):
     _request(args.url + "/shutdown", "POST", {})
     print("daemon stopping")
+
+
+def cmd_cp(args):
+    """Copy files out of (or into) a job's directory.
+
+    Counterpart of the reference's ``adaptdl cp`` (which proxies a PVC
+    through a copy pod, cli/adaptdl_cli/proxy.py); locally the job dir
+    is a directory on this node, so this resolves it from the daemon
+    and copies with shutil.  ``NAME:PATH`` addresses inside the job
+    dir; a bare path is local.
+    """
+    import shutil
+
+    def resolve(spec):
+        if ":" in spec:
+            name, rel = spec.split(":", 1)
+            st = _request("{}/jobs/{}".format(args.url, name))
+            return os.path.join(st["job_dir"], rel.lstrip("/"))
+        return spec
+
+    src_path = resolve(args.src)
+    dst_path = resolve(args.dst)
+    if os.path.isdir(src_path):
+        dst_dir = (os.path.join(dst_path, os.path.basename(src_path))
+                   if os.path.isdir(dst_path) else dst_path)
+        shutil.copytree(src_path, dst_dir, dirs_exist_ok=True)
+    else:
+        shutil.copy2(src_path, dst_path)
+    print("copied {} -> {}".format(src_path, dst_path))
 
 
 def cmd_tensorboard(args):
@@ -292,6 +325,9 @@ def main(argv=None):
         p.add_argument("--max-replicas", type=int, default=8)
         p.add_argument("--gpus-per-replica", type=int,
                        default=1 if _has_gpu() else 0)
+        p.add_argument("--inplace-scaledown", action="store_true",
+                       help="scale-downs rejoin in place (no restart; "
+                            "survivors keep state in memory)")
 
     p = sub.add_parser("daemon", help="run the persistent controller")
     p.add_argument("--bind", default="127.0.0.1:8077")
@@ -326,6 +362,12 @@ def main(argv=None):
     p = sub.add_parser("stop", help="shut the daemon down")
     p.add_argument("--url", default=DEFAULT_URL)
 
+    p = sub.add_parser("cp", help="copy files from/to a job directory "
+                                  "(NAME:PATH addresses inside it)")
+    p.add_argument("src")
+    p.add_argument("dst")
+    p.add_argument("--url", default=DEFAULT_URL)
+
     p = sub.add_parser("tensorboard",
                        help="launch TensorBoard on a job's directory")
     p.add_argument("name", nargs="?")
@@ -343,6 +385,8 @@ def main(argv=None):
         cmd_submit(args, command)
     elif args.cmd == "ls":
         cmd_ls(args)
+    elif args.cmd == "cp":
+        cmd_cp(args)
     elif args.cmd == "logs":
         cmd_logs(args)
     elif args.cmd == "rescale":

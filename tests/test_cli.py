@@ -149,3 +149,49 @@ def test_trace_gaps_union_math(tmp_path):
         trace_gaps.load_intervals([str(csv_path)]), tail=0.5)
     assert out["kernels"] == 1
     assert abs(out["idle_pct"] - 80.0) < 1e-9
+
+
+def test_cli_cp(tmp_path):
+    """`adaptdl-amd cp NAME:path dst` copies out of a job directory
+    resolved from the daemon (reference `adaptdl cp` parity)."""
+    port = _free_port()
+    env = dict(os.environ, PYTHONPATH=REPO)
+    daemon = subprocess.Popen(
+        [sys.executable, "-m", "adaptdl_amd.cli", "daemon",
+         "--bind", "127.0.0.1:{}".format(port),
+         "--state-dir", str(tmp_path / "state"),
+         "--num-gpus", "0", "--interval", "3600"],
+        env=env, cwd=str(tmp_path),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    url = "http://127.0.0.1:{}".format(port)
+    try:
+        deadline = time.time() + 30
+        while True:
+            try:
+                _req(url + "/jobs")
+                break
+            except Exception:
+                assert time.time() < deadline
+                time.sleep(0.2)
+        script = tmp_path / "writer.py"
+        script.write_text(
+            "import os\n"
+            "p = os.environ['ADAPTDL_CHECKPOINT_PATH']\n"
+            "open(os.path.join(p, 'result.txt'), 'w').write('payload')\n")
+        _req(url + "/jobs", "POST", {
+            "argv": [sys.executable, str(script)], "name": "writer",
+            "min_replicas": 1, "max_replicas": 1, "gpus_per_replica": 0})
+        deadline = time.time() + 30
+        while _req(url + "/jobs/writer")["state"] != "Succeeded":
+            assert time.time() < deadline, _req(url + "/jobs/writer")
+            time.sleep(0.2)
+        dst = tmp_path / "out.txt"
+        out = subprocess.run(
+            [sys.executable, "-m", "adaptdl_amd.cli", "cp",
+             "writer:result.txt", str(dst), "--url", url],
+            env=env, capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, out.stderr
+        assert dst.read_text() == "payload"
+    finally:
+        daemon.terminate()
+        daemon.wait(timeout=10)
