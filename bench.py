@@ -198,6 +198,10 @@ class TransformerWT2(Workload):
     """Transformer LM with adaptive BPTT batches (WikiText-2 shape)."""
 
     name = "transformer-wt2"
+    # Tiny per-kernel times make this the most host-launch-bound
+    # workload - exactly where hipGraph replay pays.  clip_grad_norm_
+    # is capture-safe (tensor-only clip coefficient, no host sync).
+    supports_hipgraph = True
     defaults = dict(init_batch=20, max_batch=1280, bounds=(16, 256))
     vocab = 33278          # WikiText-2 vocabulary size
     bptt = 35
@@ -242,6 +246,9 @@ class BertBase(Workload):
     """BERT MLM pretraining step: FusedAdam + AdamScale, bf16."""
 
     name = "bert-base"
+    # FusedAdam's device-resident preconditioner scalars make the
+    # Adam-preconditioned GNS capture-safe (r2).
+    supports_hipgraph = True
     defaults = dict(init_batch=32, max_batch=1024, bounds=(8, 256))
     seq_len = 128
     pool_size = 4096
