@@ -227,8 +227,7 @@ class TransformerWT2(Workload):
         self.loader = loader
         return self.adp, self.optim, loader
 
-    def fwd_bwd(self, batch):
-        text, target = batch
+    def fwd_bwd(self, text, target):
         self.optim.zero_grad()
         with self._autocast():
             out = self.adp(text)
@@ -402,10 +401,11 @@ def main():
 
     def train_step(batch):
         batch = workload.prep(batch)  # H2D etc., outside any capture
+        parts = batch if isinstance(batch, (tuple, list)) else (batch,)
         if graph_stepper is not None:
-            graph_stepper.microbatch(batch)
+            graph_stepper.microbatch(*parts)
         else:
-            workload.fwd_bwd(batch)
+            workload.fwd_bwd(*parts)
         optim.step()
 
     def barrier_sync():
