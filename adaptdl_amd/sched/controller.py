@@ -67,6 +67,7 @@ RUNNING = "Running"
 STOPPING = "Stopping"
 SUCCEEDED = "Succeeded"
 FAILED = "Failed"
+STOPPED = "Stopped"  # cancelled by the user / a trial scheduler
 
 
 def _free_port():
@@ -163,6 +164,7 @@ class _Job(object):
         self.warm_dir = None      # RAM-backed rescale checkpoint root
         self.inplace = None       # pending in-place scale-down state
         self.inplace_version = 0
+        self.cancelled = False    # cancel() requested -> final Stopped
 
     @property
     def num_replicas(self):
@@ -225,7 +227,7 @@ class LocalController(object):
         while True:
             with self._lock:
                 job = self._jobs[name]
-                if job.state in (SUCCEEDED, FAILED):
+                if job.state in (SUCCEEDED, FAILED, STOPPED):
                     return job.state
             if deadline is not None and time.time() > deadline:
                 raise TimeoutError("job {} still {}".format(
@@ -246,6 +248,26 @@ class LocalController(object):
             if job.state == RUNNING:
                 job.state = STOPPING
                 self._signal_group(job, signal.SIGTERM)
+        self._wake.set()
+
+    def cancel(self, name):
+        """Stop a job permanently (final state: Stopped).
+
+        Used by trial schedulers (tune.run_trials early stopping) and
+        operators; the group gets the graceful SIGTERM-checkpoint
+        treatment but is not restarted."""
+        with self._lock:
+            job = self._jobs[name]
+            if job.state in (SUCCEEDED, FAILED, STOPPED):
+                return
+            job.cancelled = True
+            if job.procs:
+                job.state = STOPPING
+                self._signal_group(job, signal.SIGTERM)
+            else:
+                job.state = STOPPED
+                job.completion = time.time()
+                self._drop_warm_dir(job)
         self._wake.set()
 
     def reallocate(self):
@@ -578,7 +600,13 @@ class LocalController(object):
         job.procs = []
         job.gpus = []
         self.supervisor.clear_job(job.spec.name)
-        if all(c == 0 for c in codes):
+        if job.cancelled:
+            job.state = STOPPED
+            job.completion = time.time()
+            self._drop_warm_dir(job)
+            LOG.info("job %s stopped (cancelled; codes=%s)",
+                     job.spec.name, codes)
+        elif all(c == 0 for c in codes):
             job.state = SUCCEEDED
             job.completion = time.time()
             self._drop_warm_dir(job)

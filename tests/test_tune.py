@@ -67,3 +67,39 @@ def test_run_trials_concurrently(tmp_path):
     # the higher learning rate should fit the linear teacher better
     assert results["lr-0.05"].result["final_loss"] < \
         results["lr-0.001"].result["final_loss"]
+
+
+def test_median_stopper_cancels_weak_trial(tmp_path):
+    """Early stopping (the Tune-scheduler role): a trial whose reported
+    metric trails the median of its peers is cancelled (state Stopped)
+    while the strong trial completes normally."""
+    import textwrap
+    from adaptdl_amd.tune import MedianStopper
+
+    worker = textwrap.dedent("""
+        import json, os, sys, time
+        acc = float(sys.argv[1])
+        share = os.environ["ADAPTDL_SHARE_PATH"]
+        for i in range(60):
+            with open(os.path.join(share, "metrics.jsonl"), "a") as f:
+                f.write(json.dumps({"acc": acc + i * 0.001}) + "\\n")
+            time.sleep(0.1)
+        with open(os.path.join(share, "result.json"), "w") as f:
+            json.dump({"final_acc": acc}, f)
+    """)
+    script = tmp_path / "trial.py"
+    script.write_text(worker)
+    trials = [
+        Trial(name="strong", argv=[sys.executable, str(script), "0.9"],
+              max_replicas=1, gpus_per_replica=0),
+        Trial(name="weak", argv=[sys.executable, str(script), "0.1"],
+              max_replicas=1, gpus_per_replica=0),
+        Trial(name="mid", argv=[sys.executable, str(script), "0.5"],
+              max_replicas=1, gpus_per_replica=0),
+    ]
+    results = run_trials(trials, trial_dir=str(tmp_path / "trials"),
+                         num_gpus=0, interval=3600, timeout=120,
+                         stopper=MedianStopper("acc", grace=3))
+    assert results["weak"].state == "Stopped"
+    assert results["strong"].state == "Succeeded"
+    assert results["strong"].result == {"final_acc": 0.9}
