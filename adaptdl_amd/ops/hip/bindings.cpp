@@ -28,6 +28,13 @@ void launch_sqsum_diff_update_bf16(const unsigned short*, unsigned short*,
                                    long, double*, hipStream_t);
 void launch_sqsum_avg_bf16(const unsigned short*, const unsigned short*,
                            long, double*, hipStream_t);
+void launch_sqsum_fp16(const unsigned short*, long, double*, hipStream_t);
+void launch_scale_sqsum_fp16(unsigned short*, long, float, double*,
+                             hipStream_t);
+void launch_sqsum_diff_update_fp16(const unsigned short*, unsigned short*,
+                                   long, double*, hipStream_t);
+void launch_sqsum_avg_fp16(const unsigned short*, const unsigned short*,
+                           long, double*, hipStream_t);
 void launch_fused_sgd(float*, const float*, float*, long, float, float,
                       float, float, int, int, hipStream_t);
 void launch_fused_adamw(float*, const float*, float*, float*, long, float,
@@ -98,21 +105,30 @@ hipStream_t stream() {
 }
 
 // The statistics entry points accept float32 (master-grad training)
-// and bfloat16 (true-bf16-parameter models) buckets.
+// plus bfloat16/float16 (true-low-precision-parameter models) buckets.
 bool is_bf16(const torch::Tensor& t) {
     return t.scalar_type() == torch::kBFloat16;
+}
+
+bool is_fp16(const torch::Tensor& t) {
+    return t.scalar_type() == torch::kHalf;
 }
 
 void check_stat_in(const torch::Tensor& t, const char* name) {
     TORCH_CHECK(t.is_cuda(), name, " must be a GPU tensor");
     TORCH_CHECK(t.scalar_type() == torch::kFloat32 ||
-                t.scalar_type() == torch::kBFloat16,
-                name, " must be float32 or bfloat16");
+                t.scalar_type() == torch::kBFloat16 ||
+                t.scalar_type() == torch::kHalf,
+                name, " must be float32, bfloat16, or float16");
     TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
 unsigned short* bf16_ptr(torch::Tensor& t) {
     return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+
+unsigned short* fp16_ptr(torch::Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr<at::Half>());
 }
 
 void sqsum(torch::Tensor x, torch::Tensor out) {
@@ -121,6 +137,9 @@ void sqsum(torch::Tensor x, torch::Tensor out) {
     if (n == 0) return;
     if (is_bf16(x))
         launch_sqsum_bf16(bf16_ptr(x), n, out.data_ptr<double>(),
+                          stream());
+    else if (is_fp16(x))
+        launch_sqsum_fp16(fp16_ptr(x), n, out.data_ptr<double>(),
                           stream());
     else
         launch_sqsum(x.data_ptr<float>(), n, out.data_ptr<double>(),
@@ -133,6 +152,9 @@ void scale_and_sqsum(torch::Tensor x, double scale, torch::Tensor out) {
     if (n == 0) return;
     if (is_bf16(x))
         launch_scale_sqsum_bf16(bf16_ptr(x), n, (float)scale,
+                                out.data_ptr<double>(), stream());
+    else if (is_fp16(x))
+        launch_scale_sqsum_fp16(fp16_ptr(x), n, (float)scale,
                                 out.data_ptr<double>(), stream());
     else
         launch_scale_sqsum(x.data_ptr<float>(), n, (float)scale,
@@ -151,6 +173,9 @@ void sqsum_diff_update(torch::Tensor cur, torch::Tensor prev,
     if (is_bf16(cur))
         launch_sqsum_diff_update_bf16(bf16_ptr(cur), bf16_ptr(prev), n,
                                       out.data_ptr<double>(), stream());
+    else if (is_fp16(cur))
+        launch_sqsum_diff_update_fp16(fp16_ptr(cur), fp16_ptr(prev), n,
+                                      out.data_ptr<double>(), stream());
     else
         launch_sqsum_diff_update(cur.data_ptr<float>(),
                                  prev.data_ptr<float>(), n,
@@ -167,6 +192,9 @@ void sqsum_avg(torch::Tensor cur, torch::Tensor prev, torch::Tensor out) {
     if (n == 0) return;
     if (is_bf16(cur))
         launch_sqsum_avg_bf16(bf16_ptr(cur), bf16_ptr(prev), n,
+                              out.data_ptr<double>(), stream());
+    else if (is_fp16(cur))
+        launch_sqsum_avg_fp16(fp16_ptr(cur), fp16_ptr(prev), n,
                               out.data_ptr<double>(), stream());
     else
         launch_sqsum_avg(cur.data_ptr<float>(), prev.data_ptr<float>(), n,

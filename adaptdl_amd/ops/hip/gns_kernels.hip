@@ -328,6 +328,125 @@ extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_avg_bf16(
     block_reduce_atomic(acc, out);
 }
 
+// ---- fp16 variants: identical structure to the bf16 set, with
+// __half conversions (fp16-parameter models).
+
+#include <hip/hip_fp16.h>
+
+__device__ __forceinline__ float hf2f(unsigned short h) {
+    __half v = *reinterpret_cast<const __half*>(&h);
+    return __half2float(v);
+}
+
+__device__ __forceinline__ unsigned short f2hf(float f) {
+    __half h = __float2half_rn(f);
+    return *reinterpret_cast<unsigned short*>(&h);
+}
+
+#define DEF_SQSUM_16(SUFFIX, TO_F, FROM_F) \
+extern "C" __global__ __launch_bounds__(BLOCK) void k_sqsum_##SUFFIX( \
+        const unsigned short* __restrict__ x, long n, \
+        double* __restrict__ out) { \
+    double acc = 0.0; \
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x; \
+    const long stride = (long)gridDim.x * BLOCK; \
+    long head = BF16_HEAD(x); \
+    if (head > n) head = n; \
+    for (long j = i; j < head; j += stride) { \
+        double v = (double)TO_F(x[j]); acc += v * v; } \
+    const us16x8* xv = (const us16x8*)(x + head); \
+    const long nv = (n - head) / 8; \
+    for (long j = i; j < nv; j += stride) { \
+        us16x8 v = xv[j]; \
+        _Pragma("unroll") for (int k = 0; k < 8; ++k) { \
+            double q = (double)TO_F(v[k]); acc += q * q; } } \
+    for (long j = head + nv * 8 + i; j < n; j += stride) { \
+        double v = (double)TO_F(x[j]); acc += v * v; } \
+    block_reduce_atomic(acc, out); \
+} \
+extern "C" __global__ __launch_bounds__(BLOCK) \
+void k_scale_sqsum_##SUFFIX( \
+        unsigned short* __restrict__ x, long n, float scale, \
+        double* __restrict__ out) { \
+    double acc = 0.0; \
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x; \
+    const long stride = (long)gridDim.x * BLOCK; \
+    long head = BF16_HEAD(x); \
+    if (head > n) head = n; \
+    for (long j = i; j < head; j += stride) { \
+        float v = TO_F(x[j]) * scale; x[j] = FROM_F(v); \
+        acc += (double)v * v; } \
+    us16x8* xv = (us16x8*)(x + head); \
+    const long nv = (n - head) / 8; \
+    for (long j = i; j < nv; j += stride) { \
+        us16x8 v = xv[j]; \
+        _Pragma("unroll") for (int k = 0; k < 8; ++k) { \
+            float q = TO_F(v[k]) * scale; v[k] = FROM_F(q); \
+            acc += (double)q * q; } \
+        xv[j] = v; } \
+    for (long j = head + nv * 8 + i; j < n; j += stride) { \
+        float v = TO_F(x[j]) * scale; x[j] = FROM_F(v); \
+        acc += (double)v * v; } \
+    block_reduce_atomic(acc, out); \
+} \
+extern "C" __global__ __launch_bounds__(BLOCK) \
+void k_sqsum_diff_update_##SUFFIX( \
+        const unsigned short* __restrict__ cur, \
+        unsigned short* __restrict__ prev, long n, \
+        double* __restrict__ out) { \
+    double acc = 0.0; \
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x; \
+    const long stride = (long)gridDim.x * BLOCK; \
+    long head = BF16_HEAD(cur); \
+    if (head > n) head = n; \
+    for (long j = i; j < head; j += stride) { \
+        unsigned short c = cur[j]; \
+        double d = (double)TO_F(c) - (double)TO_F(prev[j]); \
+        prev[j] = c; acc += d * d; } \
+    const us16x8* cv = (const us16x8*)(cur + head); \
+    us16x8* pv = (us16x8*)(prev + head); \
+    const long nv = (n - head) / 8; \
+    for (long j = i; j < nv; j += stride) { \
+        us16x8 c = cv[j]; us16x8 p = pv[j]; \
+        _Pragma("unroll") for (int k = 0; k < 8; ++k) { \
+            double d = (double)TO_F(c[k]) - (double)TO_F(p[k]); \
+            acc += d * d; } \
+        pv[j] = c; } \
+    for (long j = head + nv * 8 + i; j < n; j += stride) { \
+        unsigned short c = cur[j]; \
+        double d = (double)TO_F(c) - (double)TO_F(prev[j]); \
+        prev[j] = c; acc += d * d; } \
+    block_reduce_atomic(acc, out); \
+} \
+extern "C" __global__ __launch_bounds__(BLOCK) \
+void k_sqsum_avg_##SUFFIX( \
+        const unsigned short* __restrict__ cur, \
+        const unsigned short* __restrict__ prev, long n, \
+        double* __restrict__ out) { \
+    double acc = 0.0; \
+    long i = (long)blockIdx.x * BLOCK + threadIdx.x; \
+    const long stride = (long)gridDim.x * BLOCK; \
+    long head = BF16_HEAD(cur); \
+    if (head > n) head = n; \
+    for (long j = i; j < head; j += stride) { \
+        double a = 0.5 * ((double)TO_F(cur[j]) + (double)TO_F(prev[j])); \
+        acc += a * a; } \
+    const us16x8* cv = (const us16x8*)(cur + head); \
+    const us16x8* pv = (const us16x8*)(prev + head); \
+    const long nv = (n - head) / 8; \
+    for (long j = i; j < nv; j += stride) { \
+        us16x8 c = cv[j]; us16x8 p = pv[j]; \
+        _Pragma("unroll") for (int k = 0; k < 8; ++k) { \
+            double a = 0.5 * ((double)TO_F(c[k]) + (double)TO_F(p[k])); \
+            acc += a * a; } } \
+    for (long j = head + nv * 8 + i; j < n; j += stride) { \
+        double a = 0.5 * ((double)TO_F(cur[j]) + (double)TO_F(prev[j])); \
+        acc += a * a; } \
+    block_reduce_atomic(acc, out); \
+}
+
+DEF_SQSUM_16(fp16, hf2f, f2hf)
+
 // Shared body for the Adam-preconditioned sum-of-squares:
 //   acc = sum((g / pinv)^2), pinv = sqrt(v) * inv_corr_sqrt + eps
 // float4-vectorized when g and v are co-aligned (bucket-flat segments
@@ -532,6 +651,33 @@ extern "C" void launch_sqsum_avg_bf16(const unsigned short* cur,
                                       const unsigned short* prev, long n,
                                       double* out, hipStream_t s) {
     hipLaunchKernelGGL(k_sqsum_avg_bf16, grid_for(n), dim3(BLOCK), 0, s,
+                       cur, prev, n, out);
+}
+
+extern "C" void launch_sqsum_fp16(const unsigned short* x, long n,
+                                  double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_fp16, grid_for(n), dim3(BLOCK), 0, s, x, n,
+                       out);
+}
+
+extern "C" void launch_scale_sqsum_fp16(unsigned short* x, long n,
+                                        float scale, double* out,
+                                        hipStream_t s) {
+    hipLaunchKernelGGL(k_scale_sqsum_fp16, grid_for(n), dim3(BLOCK), 0, s,
+                       x, n, scale, out);
+}
+
+extern "C" void launch_sqsum_diff_update_fp16(const unsigned short* cur,
+                                              unsigned short* prev, long n,
+                                              double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_diff_update_fp16, grid_for(n), dim3(BLOCK),
+                       0, s, cur, prev, n, out);
+}
+
+extern "C" void launch_sqsum_avg_fp16(const unsigned short* cur,
+                                      const unsigned short* prev, long n,
+                                      double* out, hipStream_t s) {
+    hipLaunchKernelGGL(k_sqsum_avg_fp16, grid_for(n), dim3(BLOCK), 0, s,
                        cur, prev, n, out);
 }
 

