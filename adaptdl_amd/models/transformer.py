@@ -58,4 +58,7 @@ class TransformerLM(nn.Module):
         if src_mask is None:
             src_mask = self.causal_mask(src.size(0), src.device)
         x = self.pos(self.embed(src) * math.sqrt(self.d_model))
-        return self.decoder(self.encoder(x, src_mask))
+        # is_causal=True skips nn.Transformer's mask auto-detection,
+        # whose device->host compare would abort hipGraph stream capture
+        # (hipErrorStreamCaptureUnsupported).
+        return self.decoder(self.encoder(x, src_mask, is_causal=True))
