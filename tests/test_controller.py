@@ -408,3 +408,59 @@ def test_inplace_rescale_escalates_on_timeout(tmp_path, controller, monkeypatch)
         time.sleep(0.2)
     assert controller.status("stuck-job")["restarts"] >= 1
     controller.rescale("stuck-job", 0)  # cleanup is shutdown's job
+
+
+def test_mixed_inplace_down_then_respawn_up(tmp_path, controller):
+    """Interplay of the two rescale mechanisms: 2 -> 1 in place (no
+    restart), then 1 -> 2 via the warm-checkpoint respawn path; state
+    must flow survivor-memory -> warm RAM checkpoint -> new group, with
+    every epoch still running exactly once and nothing on disk."""
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="mixed-job",
+                   job_dir=job_dir, min_replicas=1, max_replicas=2,
+                   gpus_per_replica=0, inplace_scaledown=True)
+    controller.submit(spec)
+    deadline = time.time() + 90
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("mixed-job")
+        time.sleep(0.1)
+
+    # needs 2 replicas first (min_replicas=1 may start at 1)
+    if controller.status("mixed-job")["replicas"] != 2:
+        controller.rescale("mixed-job", 2)
+        while controller.status("mixed-job")["replicas"] != 2:
+            assert time.time() < deadline, controller.status("mixed-job")
+            time.sleep(0.1)
+    base_restarts = controller.status("mixed-job")["restarts"]
+
+    controller.rescale("mixed-job", 1)   # in place
+    while controller.status("mixed-job")["replicas"] != 1:
+        assert time.time() < deadline, controller.status("mixed-job")
+        time.sleep(0.1)
+    assert controller.status("mixed-job")["restarts"] == base_restarts
+
+    n_lines = len(open(trace_path).readlines())
+    while len(open(trace_path).readlines()) < n_lines + 1:
+        assert time.time() < deadline
+        time.sleep(0.1)
+
+    controller.rescale("mixed-job", 2)   # respawn via warm checkpoint
+    while not (controller.status("mixed-job")["replicas"] == 2 and
+               controller.status("mixed-job")["state"] == "Running"):
+        assert time.time() < deadline, controller.status("mixed-job")
+        time.sleep(0.1)
+    assert controller.status("mixed-job")["restarts"] == base_restarts + 1
+
+    state = controller.wait("mixed-job", timeout=180)
+    assert state == "Succeeded", controller.status("mixed-job")
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(30))
+    assert len(trace) == 30  # no epoch replayed
+    assert not any(n.startswith("checkpoint-")
+                   for n in os.listdir(job_dir))
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1

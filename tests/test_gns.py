@@ -225,9 +225,14 @@ def test_adam_precond_stats_one_launch_per_bucket(tmp_ckpt_env):
     collective.teardown()
 
 
-def test_bf16_parameter_model_trains(tmp_ckpt_env):
-    """True-bf16-parameter model through the engine: statistics run on
-    bf16 buckets; fused optimizers refuse clearly."""
+import pytest as _pytest
+
+
+@_pytest.mark.parametrize("lp_dtype", [torch.bfloat16, torch.float16])
+def test_bf16_parameter_model_trains(tmp_ckpt_env, lp_dtype):
+    """True-low-precision-parameter model through the engine:
+    statistics run on bf16/fp16 buckets; fused
+    optimizers refuse clearly."""
     import pytest
     import adaptdl_amd.torch as adl
 
@@ -235,12 +240,14 @@ def test_bf16_parameter_model_trains(tmp_ckpt_env):
         collective.initialize(master_addr="127.0.0.1")
     torch.manual_seed(0)
     model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.ReLU(),
-                                torch.nn.Linear(16, 4)).to(torch.bfloat16)
+                                torch.nn.Linear(16, 4)).to(lp_dtype)
     optim = torch.optim.SGD(model.parameters(), lr=0.05)
-    adp = adl.AdaptiveDataParallel(model, optim, name="bf16-model")
-    assert all(b.flat.dtype == torch.bfloat16
+    adp = adl.AdaptiveDataParallel(
+        model, optim,
+        name="lp-model-{}".format(str(lp_dtype).split(".")[-1]))
+    assert all(b.flat.dtype == lp_dtype
                for b in adp.gns.engine.buckets)
-    xs = torch.randn(64, 8).to(torch.bfloat16)
+    xs = torch.randn(64, 8).to(lp_dtype)
     ys = torch.randint(0, 4, (64,))
     loader = adl.AdaptiveDataLoader(
         torch.utils.data.TensorDataset(xs, ys), batch_size=16)
@@ -257,7 +264,7 @@ def test_bf16_parameter_model_trains(tmp_ckpt_env):
         assert torch.isfinite(p).all()
 
     # Fused optimizers state their fp32 requirement up front.
-    model2 = torch.nn.Linear(4, 2).to(torch.bfloat16)
+    model2 = torch.nn.Linear(4, 2).to(lp_dtype)
     optim2 = adl.FusedSGD(model2.parameters(), lr=0.1)
     with pytest.raises(ValueError, match="float32"):
         adl.AdaptiveDataParallel(model2, optim2, name="bf16-fused")
