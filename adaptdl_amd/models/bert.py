@@ -10,6 +10,7 @@ workload uses base dimensions.
 
 import torch
 import torch.nn as nn
+import torch.nn.functional as F
 
 
 class BertConfig(object):
@@ -34,6 +35,55 @@ class BertConfig(object):
                    intermediate=256, max_len=128)
 
 
+class _SelfAttention(nn.Module):
+    """Multi-head self-attention on F.scaled_dot_product_attention.
+
+    nn.TransformerEncoder's training-mode attention decomposes into
+    eager matmul/softmax/dropout kernels; SDPA keeps the whole
+    softmax(QK^T)V in one fused (flash-class) kernel on ROCm.
+    """
+
+    def __init__(self, hidden, heads, dropout):
+        super().__init__()
+        assert hidden % heads == 0
+        self.heads = heads
+        self.head_dim = hidden // heads
+        self.qkv = nn.Linear(hidden, 3 * hidden)
+        self.out = nn.Linear(hidden, hidden)
+        self.dropout = dropout
+
+    def forward(self, x, keep_mask=None):
+        b, s, h = x.shape
+        qkv = self.qkv(x).view(b, s, 3, self.heads, self.head_dim)
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+        y = F.scaled_dot_product_attention(
+            q, k, v, attn_mask=keep_mask,
+            dropout_p=self.dropout if self.training else 0.0)
+        y = y.transpose(1, 2).reshape(b, s, h)
+        return self.out(y)
+
+
+class _EncoderBlock(nn.Module):
+    """Post-LN Transformer block (BERT layout: residual -> LayerNorm),
+    GELU MLP — same structure nn.TransformerEncoderLayer(norm_first=
+    False, activation="gelu") computes."""
+
+    def __init__(self, config):
+        super().__init__()
+        self.attn = _SelfAttention(config.hidden, config.heads,
+                                   config.dropout)
+        self.ln1 = nn.LayerNorm(config.hidden)
+        self.fc1 = nn.Linear(config.hidden, config.intermediate)
+        self.fc2 = nn.Linear(config.intermediate, config.hidden)
+        self.ln2 = nn.LayerNorm(config.hidden)
+        self.drop = nn.Dropout(config.dropout)
+
+    def forward(self, x, keep_mask=None):
+        x = self.ln1(x + self.drop(self.attn(x, keep_mask)))
+        mlp = self.fc2(self.drop(F.gelu(self.fc1(x))))
+        return self.ln2(x + self.drop(mlp))
+
+
 class BertModel(nn.Module):
     def __init__(self, config):
         super().__init__()
@@ -43,10 +93,8 @@ class BertModel(nn.Module):
         self.seg = nn.Embedding(config.type_vocab, config.hidden)
         self.norm = nn.LayerNorm(config.hidden)
         self.drop = nn.Dropout(config.dropout)
-        layer = nn.TransformerEncoderLayer(
-            config.hidden, config.heads, config.intermediate,
-            config.dropout, activation="gelu", batch_first=True)
-        self.encoder = nn.TransformerEncoder(layer, config.layers)
+        self.blocks = nn.ModuleList(
+            _EncoderBlock(config) for _ in range(config.layers))
 
     def forward(self, input_ids, token_type_ids=None, attention_mask=None):
         b, s = input_ids.shape
@@ -55,10 +103,13 @@ class BertModel(nn.Module):
         if token_type_ids is not None:
             x = x + self.seg(token_type_ids)
         x = self.drop(self.norm(x))
-        pad_mask = None
+        keep_mask = None
         if attention_mask is not None:
-            pad_mask = attention_mask == 0
-        return self.encoder(x, src_key_padding_mask=pad_mask)
+            # SDPA bool mask semantics: True = may attend.
+            keep_mask = (attention_mask != 0)[:, None, None, :]
+        for block in self.blocks:
+            x = block(x, keep_mask)
+        return x
 
 
 class BertForPreTraining(nn.Module):
