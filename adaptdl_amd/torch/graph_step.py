@@ -194,6 +194,13 @@ class GraphedStepper(object):
                              "optimizers (FusedAdam/FusedAdamW); stock "
                              "torch.optim.Adam bakes per-step bias "
                              "correction into captured kernels")
+        if getattr(adp.gns.engine, "_skip_unused", False):
+            # Usage-conditional collectives contradict a static capture:
+            # a replayed cycle always re-issues the kernels of its
+            # capture, regardless of which parameters produced
+            # gradients this time.
+            raise ValueError("GraphedStepper is incompatible with "
+                             "ADAPTDL_SKIP_UNUSED_BUCKETS=1")
         self._adp = adp
         self._gns = adp.gns
         self._engine = adp.gns.engine
