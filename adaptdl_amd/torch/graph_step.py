@@ -1,12 +1,12 @@
 """hipGraph capture of the steady-state training microbatch cycle.
 
-EXPERIMENTAL, env-gated with ``ADAPTDL_HIPGRAPH=1`` (ROADMAP item 5):
-a trace-union analysis of the flagship bench measured ~4.3% of step
-time as GPU idle gaps between the many small kernel launches of
-forward+backward (earlier-session scratch trace; re-measured by the
-round-2 A/B in tools/ab_round2.sh).  Capturing the pinned
-steady-state microbatch into a hipGraph replays the whole kernel sequence
-with one launch, closing those gaps.
+Hardware-validated in round 2: the flagship bench's steady-state GPU
+idle drops 19% -> 1.7% under replay (profiles/r2_gaps_*), worth
+~0.6-0.9 ms of the 43 ms step; engaged by default for single-GPU bench
+runs of the conv workloads (`maybe_graphed_stepper(default_on=...)`,
+ADAPTDL_HIPGRAPH=0/1 overrides).  Attention stacks measured worse or
+unstable under capture (SDPA loses its flash backend; see
+profiles/README.md r2 passes l/m) and keep the eager path.
 
 Scope and shape of the capture
 ------------------------------
