@@ -123,8 +123,9 @@ class JobSpec(object):
         # Scale-downs rejoin in place (SIGUSR2 directive; survivors keep
         # all state in memory, leavers exit) instead of checkpoint-
         # restarting the whole group.  Requires the worker to train via
-        # AdaptiveDataLoader (BPTT/eval-only phases fall back to the
-        # restart path via the controller's escalation timeout).
+        # AdaptiveDataLoader or AdaptiveBPTTIterator (eval-only phases
+        # fall back to the restart path via the controller's escalation
+        # timeout).
         self.inplace_scaledown = inplace_scaledown
         self._frozen = False
 

@@ -76,52 +76,70 @@ class AdaptiveBPTTIterator(AdaptiveDataLoaderMixin):
             rows / (self.bptt_len * adaptdl_amd.env.num_replicas()))
 
     def __iter__(self):
-        num_replicas = adaptdl_amd.env.num_replicas()
-        rank = adaptdl_amd.env.replica_rank()
+        from adaptdl_amd.torch import _rejoin
+        self._elastic._supports_inplace = True
         with self._elastic.context():
             if self._elastic.skipdone():
                 return
-            atomic_bsz = self._elastic._sync_local_bsz()
-            # Reshape the corpus to the ATOMIC (per-replica) width; data
-            # parallelism comes from ranks striding disjoint bptt-sized
-            # time segments of the same reshaped matrix (reference
-            # iterator.py:60-97).
-            bsz = atomic_bsz
-            n = len(self.data)
-            pad = int(math.ceil(n / bsz) * bsz - n)
-            data = torch.cat([self.data,
-                              self.data.new_full((pad,), self.pad_token)])
-            # (bsz, rows) -> (rows, bsz): column b is a contiguous slice
-            # of the corpus, the classic word-LM "batchify".
-            data = data.view(bsz, -1).t().contiguous()
-            if self.device is not None:
-                data = data.to(self.device)
-            end = data.size(0)
+            while True:
+                # Re-read per attempt: an in-place rescale resumes this
+                # generator at a new replica count; _recompute_start
+                # remaps the corpus position exactly as a restart would.
+                num_replicas = adaptdl_amd.env.num_replicas()
+                rank = adaptdl_amd.env.replica_rank()
+                atomic_bsz = self._elastic._sync_local_bsz()
+                # Reshape the corpus to the ATOMIC (per-replica) width;
+                # data parallelism comes from ranks striding disjoint
+                # bptt-sized time segments of the same reshaped matrix
+                # (reference iterator.py:60-97).
+                bsz = atomic_bsz
+                n = len(self.data)
+                pad = int(math.ceil(n / bsz) * bsz - n)
+                data = torch.cat([
+                    self.data,
+                    self.data.new_full((pad,), self.pad_token)])
+                # (bsz, rows) -> (rows, bsz): column b is a contiguous
+                # slice of the corpus, the classic word-LM "batchify".
+                data = data.view(bsz, -1).t().contiguous()
+                if self.device is not None:
+                    data = data.to(self.device)
+                end = data.size(0)
 
-            self._elastic.current_index = self._recompute_start(
-                self._elastic.current_index, self._elastic.end_index, end)
-            self._elastic.end_index = end
+                self._elastic.current_index = self._recompute_start(
+                    self._elastic.current_index, self._elastic.end_index,
+                    end)
+                self._elastic.end_index = end
 
-            start = self._elastic.current_index + self.bptt_len * rank
-            step = self.bptt_len * num_replicas
-            highest_start = self._elastic.current_index + \
-                self.bptt_len * (num_replicas - 1)
-            # Cap iterations at the count of the most-starved rank so
-            # every replica enters profile()/collectives equally often.
-            min_steps = max(math.ceil((end - 1 - highest_start) / step), 0)
+                start = self._elastic.current_index + self.bptt_len * rank
+                step = self.bptt_len * num_replicas
+                highest_start = self._elastic.current_index + \
+                    self.bptt_len * (num_replicas - 1)
+                # Cap iterations at the count of the most-starved rank so
+                # every replica enters profile()/collectives equally
+                # often.
+                min_steps = max(
+                    math.ceil((end - 1 - highest_start) / step), 0)
 
-            iterations = 0
-            for i in range(start, end, step):
-                iterations += 1
-                if iterations > min_steps:
-                    break
-                with self._elastic.profile(self.training and i > 0):
-                    seq_len = min(self.bptt_len, end - i - 1)
-                    assert seq_len > 0
-                    text = data[i:i + seq_len]
-                    target = data[i + 1:i + 1 + seq_len]
-                    if self.batch_first:
-                        text = text.t().contiguous()
-                        target = target.t().contiguous()
-                    yield text, target
-                    self._elastic.current_index += step
+                iterations = 0
+                try:
+                    for i in range(start, end, step):
+                        iterations += 1
+                        if iterations > min_steps:
+                            break
+                        with self._elastic.profile(self.training
+                                                   and i > 0):
+                            seq_len = min(self.bptt_len, end - i - 1)
+                            assert seq_len > 0
+                            text = data[i:i + seq_len]
+                            target = data[i + 1:i + 1 + seq_len]
+                            if self.batch_first:
+                                text = text.t().contiguous()
+                                target = target.t().contiguous()
+                            yield text, target
+                            self._elastic.current_index += step
+                except _rejoin.InplaceRescale as req:
+                    # Leavers exit inside; survivors resume the pass at
+                    # the remapped corpus position with the new world.
+                    _rejoin.perform(req.directive)
+                    continue
+                return

@@ -464,3 +464,78 @@ def test_mixed_inplace_down_then_respawn_up(tmp_path, controller):
                    for n in os.listdir(job_dir))
     final = json.load(open(os.path.join(job_dir, "final.json")))
     assert abs(final[0] - 3.0) < 0.1 and abs(final[1] - 4.0) < 0.1
+
+
+BPTT_WORKER = textwrap.dedent("""
+    import json, os, sys
+    sys.path.insert(0, "@@REPO@@")
+    import torch
+    torch.set_num_threads(1)
+    import torch.nn.functional as F
+    import adaptdl_amd.env as env
+    import adaptdl_amd.torch as adl
+
+    adl.init_process_group("gloo")
+    torch.manual_seed(3)
+    vocab = 40
+    corpus = torch.randint(0, vocab, (3000,))
+
+    class LM(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = torch.nn.Embedding(vocab, 12)
+            self.fc = torch.nn.Linear(12, vocab)
+
+        def forward(self, x):
+            return self.fc(self.emb(x))
+
+    model = LM()
+    optim = torch.optim.SGD(model.parameters(), lr=0.05)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    it = adl.AdaptiveBPTTIterator(corpus, batch_size=4, bptt_len=10)
+    trace = os.path.join(env.checkpoint_path(), "trace.jsonl")
+    for epoch in adl.remaining_epochs_until(25):
+        for text, target in it:
+            optim.zero_grad()
+            loss = F.cross_entropy(adp(text).view(-1, vocab),
+                                   target.reshape(-1))
+            loss.backward()
+            optim.step()
+        if env.replica_rank() == 0:
+            with open(trace, "a") as f:
+                f.write(json.dumps(dict(
+                    epoch=epoch, replicas=env.num_replicas(),
+                    restarts=env.num_restarts())) + "\\n")
+        import time as _t
+        _t.sleep(0.05)
+""")
+
+
+def test_inplace_scaledown_bptt(tmp_path, controller):
+    """The BPTT iterator also supports in-place scale-downs: 2 -> 1 with
+    the corpus position remapped in place (no restart, no replay)."""
+    script = tmp_path / "bptt_worker.py"
+    script.write_text(BPTT_WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="bptt-job",
+                   job_dir=job_dir, min_replicas=2, max_replicas=2,
+                   gpus_per_replica=0, inplace_scaledown=True)
+    controller.submit(spec)
+    deadline = time.time() + 90
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("bptt-job")
+        time.sleep(0.1)
+    controller.rescale("bptt-job", 1)
+    while controller.status("bptt-job")["replicas"] != 1:
+        assert time.time() < deadline, controller.status("bptt-job")
+        assert controller.status("bptt-job")["restarts"] == 0
+        time.sleep(0.1)
+    state = controller.wait("bptt-job", timeout=120)
+    assert state == "Succeeded", controller.status("bptt-job")
+    assert controller.status("bptt-job")["restarts"] == 0
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(25))
+    assert len(trace) == 25
+    assert {t["replicas"] for t in trace} == {2, 1}
