@@ -9,7 +9,10 @@ Covers the reference's example-exercised paths on CPU/gloo:
 """
 
 import numpy as np
+import pytest
 import torch
+
+import adaptdl_amd.torch
 
 import adaptdl_amd.collective
 import adaptdl_amd.env
@@ -146,3 +149,94 @@ def test_cifar_example_script_cpu_smoke(tmp_path):
         cwd=repo, capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
     assert "epoch 0" in out.stdout
+
+
+@pytest.mark.gpu
+def test_bert_sdpa_encoder_gpu(tmp_ckpt_env):
+    """BERT-mini on GPU: the SDPA encoder under bf16 autocast through
+    ADP + FusedAdam (the config-4 bench path), vs an fp32 eager
+    reference of the same forward."""
+    import torch.nn.functional as F
+    from adaptdl_amd.models import BertConfig, BertForMaskedLM
+
+    if not adaptdl_amd.collective.initialized():
+        adaptdl_amd.collective.initialize("127.0.0.1")
+    dev = torch.device("cuda")
+    torch.manual_seed(0)
+    config = BertConfig.mini()
+    model = BertForMaskedLM(config).to(dev)
+    optim = adaptdl_amd.torch.FusedAdam(model.parameters(), lr=1e-4)
+    adp = adaptdl_amd.torch.AdaptiveDataParallel(model, optim,
+                                                 name="bert-gpu")
+    x = torch.randint(10, config.vocab_size, (8, 32), device=dev)
+    y = torch.full((8, 32), -100, device=dev)
+    y[:, ::5] = x[:, ::5]
+
+    model.eval()  # disable dropout for the numerics comparison
+    with torch.no_grad():
+        ref = model.bert(x).float()
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            got = model.bert(x).float()
+    assert torch.isfinite(got).all()
+    # bf16 SDPA vs fp32 decomposed path: loose agreement.
+    assert (got - ref).abs().max().item() < 0.1, \
+        (got - ref).abs().max().item()
+
+    model.train()
+    dataset = torch.utils.data.TensorDataset(torch.arange(32))
+    loader = adaptdl_amd.torch.AdaptiveDataLoader(dataset, batch_size=8)
+    for _epoch in adaptdl_amd.torch.remaining_epochs_until(1):
+        for (idx,) in loader:
+            optim.zero_grad()
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                logits = adp(x)
+                loss = F.cross_entropy(
+                    logits.view(-1, logits.size(-1)), y.view(-1),
+                    ignore_index=-100)
+            loss.backward()
+            optim.step()
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
+
+
+@pytest.mark.gpu
+def test_transformer_lm_gpu(tmp_ckpt_env):
+    """TransformerLM on GPU (is_causal fast path): bf16 fwd/bwd + one
+    fused-SGD cycle, finite outputs, and causal masking actually
+    enforced (token t must not see t+1)."""
+    import torch.nn.functional as F
+    from adaptdl_amd.models import TransformerLM
+
+    if not adaptdl_amd.collective.initialized():
+        adaptdl_amd.collective.initialize("127.0.0.1")
+    dev = torch.device("cuda")
+    torch.manual_seed(1)
+    vocab = 100
+    model = TransformerLM(vocab).to(dev).eval()
+    a = torch.randint(0, vocab, (12, 4), device=dev)
+    b = a.clone()
+    b[-1] = (b[-1] + 1) % vocab  # change only the LAST position
+    with torch.no_grad():
+        ya = model(a)
+        yb = model(b)
+    # causal: outputs before the changed position are identical
+    assert torch.allclose(ya[:-1], yb[:-1], atol=1e-5)
+
+    model.train()
+    optim = adaptdl_amd.torch.FusedSGD(model.parameters(), lr=0.1)
+    adp = adaptdl_amd.torch.AdaptiveDataParallel(model, optim,
+                                                 name="tr-gpu")
+    dataset = torch.utils.data.TensorDataset(torch.arange(16))
+    loader = adaptdl_amd.torch.AdaptiveDataLoader(dataset, batch_size=4)
+    tgt = torch.randint(0, vocab, (12, 4), device=dev)
+    for _epoch in adaptdl_amd.torch.remaining_epochs_until(1):
+        for _ in loader:
+            optim.zero_grad()
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                out = adp(a)
+                loss = F.cross_entropy(out.view(-1, vocab),
+                                       tgt.reshape(-1))
+            loss.backward()
+            optim.step()
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
