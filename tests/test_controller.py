@@ -539,3 +539,86 @@ def test_inplace_scaledown_bptt(tmp_path, controller):
     assert sorted(set(t["epoch"] for t in trace)) == list(range(25))
     assert len(trace) == 25
     assert {t["replicas"] for t in trace} == {2, 1}
+
+
+ACCUM_WORKER = textwrap.dedent("""
+    import json, os, sys
+    sys.path.insert(0, "@@REPO@@")
+    import torch
+    torch.set_num_threads(1)
+    import torch.nn.functional as F
+    import adaptdl_amd.env as env
+    import adaptdl_amd.torch as adl
+    from adaptdl_amd.torch.data import AdaptiveDataLoaderHelper
+
+    # Force a 3-microbatch accumulation cycle so the in-place rescale
+    # must defer to the cycle boundary.
+    def fake_sync(self):
+        self._state.current_local_bsz = 4
+        self._state.accumulation_steps = 2
+        return 4
+
+    AdaptiveDataLoaderHelper._sync_local_bsz = fake_sync
+
+    adl.init_process_group("gloo")
+    torch.manual_seed(11)
+    xs = torch.randn(96, 6)
+    ys = (xs @ torch.tensor([[1.0], [2.0], [0.5], [-1.0], [0.3], [2.5]]))
+    model = torch.nn.Linear(6, 1, bias=False)
+    optim = torch.optim.SGD(model.parameters(), lr=0.02)
+    adp = adl.AdaptiveDataParallel(model, optim)
+    loader = adl.AdaptiveDataLoader(
+        torch.utils.data.TensorDataset(xs, ys), batch_size=8)
+
+    trace = os.path.join(env.checkpoint_path(), "trace.jsonl")
+    for epoch in adl.remaining_epochs_until(25):
+        for x, y in loader:
+            optim.zero_grad()
+            ((adp(x) - y) ** 2).mean().backward()
+            optim.step()
+        if env.replica_rank() == 0:
+            with open(trace, "a") as f:
+                f.write(json.dumps(dict(
+                    epoch=epoch, replicas=env.num_replicas(),
+                    restarts=env.num_restarts(),
+                    accum=adp.gns.accum_count)) + "\\n")
+        import time as _t
+        _t.sleep(0.05)
+    if env.replica_rank() == 0:
+        with open(os.path.join(env.checkpoint_path(), "final.json"),
+                  "w") as f:
+            json.dump(model.weight.detach().reshape(-1).tolist(), f)
+""")
+
+
+def test_inplace_scaledown_with_accumulation(tmp_path, controller):
+    """In-place rescale under gradient accumulation: the rejoin must
+    land on an optimizer-cycle boundary (no partial gradients lost),
+    with training completing at 0 restarts and finite learned weights."""
+    script = tmp_path / "accum_worker.py"
+    script.write_text(ACCUM_WORKER.replace("@@REPO@@", REPO))
+    job_dir = str(tmp_path / "job")
+    os.makedirs(job_dir)
+    spec = JobSpec([sys.executable, str(script)], name="accum-job",
+                   job_dir=job_dir, min_replicas=2, max_replicas=2,
+                   gpus_per_replica=0, inplace_scaledown=True)
+    controller.submit(spec)
+    deadline = time.time() + 90
+    trace_path = os.path.join(job_dir, "trace.jsonl")
+    while not os.path.exists(trace_path):
+        assert time.time() < deadline, controller.status("accum-job")
+        time.sleep(0.1)
+    controller.rescale("accum-job", 1)
+    while controller.status("accum-job")["replicas"] != 1:
+        assert time.time() < deadline, controller.status("accum-job")
+        assert controller.status("accum-job")["restarts"] == 0
+        time.sleep(0.1)
+    state = controller.wait("accum-job", timeout=120)
+    assert state == "Succeeded", controller.status("accum-job")
+    assert controller.status("accum-job")["restarts"] == 0
+    trace = [json.loads(line) for line in open(trace_path)]
+    assert sorted(set(t["epoch"] for t in trace)) == list(range(25))
+    assert len(trace) == 25
+    assert {t["replicas"] for t in trace} == {2, 1}
+    final = json.load(open(os.path.join(job_dir, "final.json")))
+    assert all(abs(v) < 10 for v in final)
