@@ -73,6 +73,11 @@ class _AdminServer(object):
                 try:
                     if parts == ["jobs"]:
                         self._reply(200, srv.controller.jobs())
+                    elif parts == ["cluster"]:
+                        self._reply(200, {
+                            "num_gpus": srv.controller.num_gpus,
+                            "desired_nodes":
+                                srv.controller.desired_nodes()})
                     elif len(parts) == 2 and parts[0] == "jobs":
                         self._reply(200, srv.controller.status(parts[1]))
                     elif len(parts) == 3 and parts[0] == "jobs" and \

@@ -93,6 +93,7 @@ class LocalAllocator(object):
             dict job-key -> list of node names (one per replica).
         """
         if not jobs:
+            self.desired_nodes = min(1, len(self._nodes))
             return {}
         t0 = time.time()
         allocations, desired_nodes = self._policy.optimize(
@@ -100,4 +101,9 @@ class LocalAllocator(object):
         LOG.info("allocator cycle: %d jobs in %.2fs -> %s (desired nodes "
                  "%d)", len(jobs), time.time() - t0,
                  {k: len(v) for k, v in allocations.items()}, desired_nodes)
+        # Cluster-expander signal (reference cluster_expander.py creates
+        # placeholder pods so the k8s autoscaler grows the cluster; the
+        # local analog is an observable target an external provisioner
+        # can consume via the controller status/metrics).
+        self.desired_nodes = desired_nodes
         return allocations

@@ -46,6 +46,10 @@ try:  # optional Prometheus metrics (reference controller.py:35-41)
             "In-place (no-restart) scale-downs started"),
         "replicas": Gauge("adaptdl_running_replicas",
                           "Currently running replica processes"),
+        "desired_nodes": Gauge(
+            "adaptdl_desired_nodes",
+            "Pollux's desired node count (cluster-expander signal for "
+            "an external provisioner)"),
     }
 except ImportError:  # pragma: no cover - prometheus_client is optional
     METRICS = None
@@ -222,6 +226,12 @@ class LocalController(object):
         with self._lock:
             return {name: self.status(name) for name in self._jobs}
 
+    def desired_nodes(self):
+        """Pollux's desired node count from the last allocator cycle
+        (the reference cluster-expander's output, exposed for external
+        provisioners; also a Prometheus gauge)."""
+        return getattr(self.allocator, "desired_nodes", 1)
+
     def wait(self, name, timeout=None):
         """Block until the job completes; returns its final state."""
         deadline = None if timeout is None else time.time() + timeout
@@ -334,6 +344,9 @@ class LocalController(object):
                            "pods": 1})
             base[name] = list(job.allocation)
         allocations = self.allocator.optimize(jobs_info, base)
+        if METRICS is not None:
+            METRICS["desired_nodes"].set(
+                getattr(self.allocator, "desired_nodes", 1))
         with self._lock:
             for name, alloc in allocations.items():
                 job = self._jobs.get(name)

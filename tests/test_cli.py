@@ -195,3 +195,33 @@ def test_cli_cp(tmp_path):
     finally:
         daemon.terminate()
         daemon.wait(timeout=10)
+
+
+def test_daemon_cluster_endpoint(tmp_path):
+    """GET /cluster exposes the Pollux desired-node count (the
+    cluster-expander signal for external provisioners)."""
+    port = _free_port()
+    env = dict(os.environ, PYTHONPATH=REPO)
+    daemon = subprocess.Popen(
+        [sys.executable, "-m", "adaptdl_amd.cli", "daemon",
+         "--bind", "127.0.0.1:{}".format(port),
+         "--state-dir", str(tmp_path / "state"),
+         "--num-gpus", "4", "--interval", "3600"],
+        env=env, cwd=str(tmp_path),
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    url = "http://127.0.0.1:{}".format(port)
+    try:
+        deadline = time.time() + 30
+        while True:
+            try:
+                out = _req(url + "/cluster")
+                break
+            except Exception:
+                assert time.time() < deadline
+                time.sleep(0.2)
+        assert out["num_gpus"] == 4
+        assert isinstance(out["desired_nodes"], int)
+        assert out["desired_nodes"] >= 0
+    finally:
+        daemon.terminate()
+        daemon.wait(timeout=10)
