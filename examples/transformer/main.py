@@ -62,6 +62,10 @@ def main():
                                   bptt_len=args.bptt,
                                   max_batch_size=args.bs * 64,
                                   local_bsz_bounds=(16, 256),
+                                  # Extension over the reference: lets
+                                  # a single replica scale its batch
+                                  # via accumulation (BASELINE config 3)
+                                  gradient_accumulation=True,
                                   device=device)
     for epoch in adl.remaining_epochs_until(args.epochs):
         total_loss, total_tok = 0.0, 0
