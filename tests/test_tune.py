@@ -103,3 +103,35 @@ def test_median_stopper_cancels_weak_trial(tmp_path):
     assert results["weak"].state == "Stopped"
     assert results["strong"].state == "Succeeded"
     assert results["strong"].result == {"final_acc": 0.9}
+
+
+def test_median_stopper_rules():
+    """MedianStopper unit semantics: grace period, median tie-keeping,
+    and min-mode inversion."""
+    from adaptdl_amd.tune import MedianStopper
+
+    s = MedianStopper("acc", grace=2)
+    hist = {
+        "a": [{"acc": 0.9}, {"acc": 0.95}],
+        "b": [{"acc": 0.5}, {"acc": 0.55}],
+        "c": [{"acc": 0.7}, {"acc": 0.75}],
+    }
+    assert not s.should_stop("a", hist)
+    assert s.should_stop("b", hist)       # below median of bests
+    assert not s.should_stop("c", hist)   # the median itself is kept
+    # grace: one report only -> never stopped
+    hist["d"] = [{"acc": 0.01}]
+    assert not s.should_stop("d", hist)
+    # missing metric entries don't count toward grace
+    hist["e"] = [{"loss": 1.0}, {"loss": 0.9}]
+    assert not s.should_stop("e", hist)
+
+    smin = MedianStopper("loss", mode="min", grace=1)
+    hist2 = {"x": [{"loss": 0.2}], "y": [{"loss": 0.9}],
+             "z": [{"loss": 0.5}]}
+    assert smin.should_stop("y", hist2)
+    assert not smin.should_stop("x", hist2)
+
+    import pytest
+    with pytest.raises(ValueError):
+        MedianStopper("m", mode="best")
