@@ -220,6 +220,9 @@ class LocalController(object):
             return {"state": job.state, "replicas": job.num_replicas,
                     "restarts": job.num_restarts,
                     "allocation": list(job.allocation),
+                    "inplace_rescale_pending":
+                        None if job.inplace is None
+                        else job.inplace["world"],
                     "job_dir": job.spec.job_dir}
 
     def jobs(self):
