@@ -39,13 +39,15 @@ LOG = logging.getLogger(__name__)
 
 class Trial(object):
     def __init__(self, name, argv, env=None, min_replicas=0,
-                 max_replicas=8, gpus_per_replica=1):
+                 max_replicas=8, gpus_per_replica=1,
+                 inplace_scaledown=False):
         self.name = name
         self.argv = list(argv)
         self.env = dict(env or {})
         self.min_replicas = min_replicas
         self.max_replicas = max_replicas
         self.gpus_per_replica = gpus_per_replica
+        self.inplace_scaledown = inplace_scaledown
 
 
 class MedianStopper(object):
@@ -138,7 +140,8 @@ def run_trials(trials, trial_dir=".adaptdl/trials", num_gpus=None,
                 trial.argv, name=trial.name, job_dir=job_dir,
                 min_replicas=trial.min_replicas,
                 max_replicas=trial.max_replicas,
-                gpus_per_replica=trial.gpus_per_replica, env=env))
+                gpus_per_replica=trial.gpus_per_replica, env=env,
+                inplace_scaledown=trial.inplace_scaledown))
         deadline = None if timeout is None else time.time() + timeout
         results = {}
         pending = {t.name for t in trials}
