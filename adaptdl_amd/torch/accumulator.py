@@ -61,13 +61,14 @@ class Accumulator(collections.abc.MutableMapping):
             self._state.sync()
             from adaptdl_amd.torch.data import current_dataloader
             if current_dataloader() is None:
-                # Inside dataloader iterations code is not replayed, so only
-                # record history outside of them.
+                # Inside dataloader iterations code is not replayed, so
+                # only record history outside of them.
                 results_list.append(copy.deepcopy(self._state.results))
             self._synchronized = self._state.results
         try:
             yield self
         finally:
+            # Back to accumulation mode.
             self._synchronized = None
 
     def update(self, *args, **kwargs):
@@ -133,6 +134,10 @@ class Accumulator(collections.abc.MutableMapping):
 
 
 class _Value(object):
+    """Accumulation-mode proxy: ``accum[k] += v`` reads this out of
+    __getitem__, applies +/-, and hands it back to __setitem__, which
+    records the delta."""
+
     __slots__ = ["accum", "key", "update"]
 
     def __init__(self, accum, key):
@@ -140,17 +145,18 @@ class _Value(object):
         self.key = key
         self.update = 0
 
-    def __add__(self, update):
-        if isinstance(update, _Value):
-            raise TypeError("invalid update type: {}".format(type(update)))
-        self.update += update
+    def _apply(self, delta):
+        if isinstance(delta, _Value):
+            raise TypeError(
+                "invalid update type: {}".format(type(delta)))
+        self.update += delta
         return self
 
-    def __sub__(self, update):
-        if isinstance(update, _Value):
-            raise TypeError("invalid update type: {}".format(type(update)))
-        self.update -= update
-        return self
+    def __add__(self, delta):
+        return self._apply(delta)
+
+    def __sub__(self, delta):
+        return self._apply(-delta)
 
 
 def _dict_iadd(a, b):

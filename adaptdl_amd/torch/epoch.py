@@ -24,20 +24,21 @@ def remaining_epochs_until(epoch):
     Raises:
         RuntimeError: If invoked before a previous epoch loop has ended.
     """
-    if current_epoch() is not None:
+    state = _epoch_state()
+    if state.current_epoch is not None:
         raise RuntimeError("overlapping epoch loops detected")
-    if finished_epochs() < epoch:
-        LOG.info("starting at epoch %s", finished_epochs())
-    else:
-        LOG.info("skipping all epochs up to %s", epoch)
-    while finished_epochs() < epoch:
-        _epoch_state().current_epoch = finished_epochs()
+    LOG.info("starting at epoch %s" if state.finished_epochs < epoch
+             else "skipping all epochs up to %s",
+             state.finished_epochs if state.finished_epochs < epoch
+             else epoch)
+    while state.finished_epochs < epoch:
+        state.current_epoch = state.finished_epochs
         try:
-            yield current_epoch()
+            yield state.current_epoch
         finally:
             # Catches breaks and exceptions escaping the epoch body too.
-            _epoch_state().finished_epochs += 1
-            _epoch_state().current_epoch = None
+            state.finished_epochs += 1
+            state.current_epoch = None
 
 
 def current_epoch():
@@ -51,6 +52,10 @@ def finished_epochs():
 
 
 class _EpochState(adaptdl_amd.checkpoint.State):
+    """Checkpointed epoch counter; current_epoch is deliberately NOT
+    persisted (a restart resumes AT the interrupted epoch via the
+    dataloader's position replay, not inside it)."""
+
     def __init__(self):
         super().__init__(".adaptdl-epoch")
         self.finished_epochs = 0
