@@ -227,6 +227,11 @@ def cmd_submit(args, command):
                                              out["job_dir"]))
 
 
+def cmd_status(args):
+    print(json.dumps(_request("{}/jobs/{}".format(args.url, args.name)),
+                     indent=2))
+
+
 def cmd_ls(args):
     jobs = _request(args.url + "/jobs")
     fmt = "{:<24} {:<10} {:>8} {:>8}"
@@ -354,6 +359,10 @@ def main(argv=None):
     p = sub.add_parser("ls", help="list daemon jobs")
     p.add_argument("--url", default=DEFAULT_URL)
 
+    p = sub.add_parser("status", help="print one job's status JSON")
+    p.add_argument("name")
+    p.add_argument("--url", default=DEFAULT_URL)
+
     p = sub.add_parser("logs", help="print job logs")
     p.add_argument("name")
     p.add_argument("--rank", type=int, default=None)
@@ -390,6 +399,8 @@ def main(argv=None):
         cmd_submit(args, command)
     elif args.cmd == "ls":
         cmd_ls(args)
+    elif args.cmd == "status":
+        cmd_status(args)
     elif args.cmd == "cp":
         cmd_cp(args)
     elif args.cmd == "logs":

@@ -60,8 +60,10 @@ Safety rules (all checked per cycle; violation falls back to eager):
   read the bias-correction scalars from device memory) but refused for
   stock torch.optim.Adam (per-step scalars would be baked into the
   capture),
+- ``ADAPTDL_SKIP_UNUSED_BUCKETS`` is refused (usage-conditional
+  collectives contradict a static capture),
 - any capture error permanently disables the stepper for the run (the
-  current microbatch is re-run eagerly).
+  current microbatch is re-run eagerly after a clean cycle reset).
 
 Reference note: the reference has no equivalent (it delegates launch
 scheduling to torch DDP); this is an MI355X-native addition.
